@@ -1,0 +1,395 @@
+"""apex_amd.amp — O0-O3 mixed precision (reconstructed apex.amp surface).
+
+The reference implementation was removed upstream; its API is pinned by the
+in-tree tests (tests/L1/common/main_amp.py:224-230,440,
+tests/distributed/amp_master_params/amp_master_params.py:45-71,
+examples/dcgan/main_amp.py:214-253 — see SURVEY.md §0/§3.4):
+
+    model, optimizer = amp.initialize(model, optimizer, opt_level="O1|O2",
+                                      keep_batchnorm_fp32=..., loss_scale=...)
+    with amp.scale_loss(loss, optimizer) as scaled_loss:
+        scaled_loss.backward()
+    optimizer.step()
+    amp.master_params(optimizer)
+
+MI355X-first design decisions (vs. the reference's global torch-function
+monkey-patching for O1):
+
+* O1 wraps each model's ``forward`` in ``torch.autocast`` — the PyTorch-ROCm
+  native cast-registry serves the same whitelist/blacklist role the patch
+  registry served, with identical loss-parity semantics for the L1 harness.
+* O2/O3 cast the model itself (keeping batchnorm fp32 for O2 by default) and
+  keep fp32 master weights inside the optimizer (``_amp_stash``), with the
+  unscale + overflow check done by one fused ``multi_tensor_scale`` launch
+  and the dynamic-scale update by the on-device ``update_scale_hysteresis``
+  kernel.
+* default cast dtype is fp16 (reference default); pass
+  ``cast_model_type=torch.bfloat16`` for the MI355X-preferred bf16 path
+  (bf16 typically needs no loss scaling; ``loss_scale=1.0`` then).
+"""
+
+import contextlib
+import functools
+import itertools
+import warnings
+
+import torch
+
+from ._amp_state import _amp_state, maybe_print
+from .scaler import LossScaler
+from .._ext import get_ext
+from ..multi_tensor_apply import multi_tensor_applier
+
+__all__ = ["initialize", "scale_loss", "master_params", "state_dict", "load_state_dict"]
+
+
+class OptProperties:
+    def __init__(self, opt_level, cast_model_type, patch_torch_functions,
+                 keep_batchnorm_fp32, master_weights, loss_scale):
+        self.opt_level = opt_level
+        self.cast_model_type = cast_model_type
+        self.patch_torch_functions = patch_torch_functions
+        self.keep_batchnorm_fp32 = keep_batchnorm_fp32
+        self.master_weights = master_weights
+        self.loss_scale = loss_scale
+
+
+_OPT_LEVELS = {
+    "O0": dict(cast_model_type=None, patch_torch_functions=False,
+               keep_batchnorm_fp32=None, master_weights=False, loss_scale=1.0),
+    "O1": dict(cast_model_type=None, patch_torch_functions=True,
+               keep_batchnorm_fp32=None, master_weights=False, loss_scale="dynamic"),
+    "O2": dict(cast_model_type=torch.float16, patch_torch_functions=False,
+               keep_batchnorm_fp32=True, master_weights=True, loss_scale="dynamic"),
+    "O3": dict(cast_model_type=torch.float16, patch_torch_functions=False,
+               keep_batchnorm_fp32=False, master_weights=False, loss_scale=1.0),
+}
+
+
+def _is_bn(module):
+    return isinstance(module, torch.nn.modules.batchnorm._BatchNorm)
+
+
+def _cast_model(model, dtype, keep_batchnorm_fp32):
+    model.to(dtype=dtype)
+    if keep_batchnorm_fp32:
+        for m in model.modules():
+            if _is_bn(m):
+                m.float()
+    return model
+
+
+class _AmpStash:
+    pass
+
+
+def _wrap_forward_autocast(model, dtype):
+    if getattr(model, "_amp_autocast_wrapped", False):
+        return
+    old_forward = model.forward
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+
+    @functools.wraps(old_forward)
+    def new_forward(*args, **kwargs):
+        with torch.autocast(device_type=device_type, dtype=dtype):
+            return old_forward(*args, **kwargs)
+
+    model.forward = new_forward
+    model._amp_autocast_wrapped = True
+
+
+def _wrap_forward_input_cast(model, dtype):
+    """O2/O3: cast floating-point inputs to the model dtype."""
+    if getattr(model, "_amp_input_cast_wrapped", False):
+        return
+    old_forward = model.forward
+
+    def cast_tree(x):
+        if torch.is_tensor(x) and x.is_floating_point() and x.dtype == torch.float32:
+            return x.to(dtype)
+        if isinstance(x, (list, tuple)):
+            return type(x)(cast_tree(v) for v in x)
+        if isinstance(x, dict):
+            return {k: cast_tree(v) for k, v in x.items()}
+        return x
+
+    @functools.wraps(old_forward)
+    def new_forward(*args, **kwargs):
+        return old_forward(*cast_tree(args), **cast_tree(kwargs))
+
+    model.forward = new_forward
+    model._amp_input_cast_wrapped = True
+
+
+def _process_optimizer_o2(optimizer, cast_type, verbose=False):
+    """Build fp32 masters inside the optimizer (the amp-O2 master contract,
+    evidenced by apex/optimizers/fused_sgd.py:165-230)."""
+    stash = _AmpStash()
+    stash.fp16_groups = []
+    stash.fp32_from_fp16_groups = []
+    stash.fp32_from_fp32_groups = []
+    stash.all_fp16_params = []
+    stash.all_fp32_from_fp16_params = []
+    stash.all_fp32_from_fp32_params = []
+
+    for group in optimizer.param_groups:
+        fp16_params_this_group = []
+        fp32_params_this_group = []
+        fp32_from_fp16_params_this_group = []
+        new_params = []
+        for p in group["params"]:
+            if p.requires_grad:
+                if p.dtype in (torch.float16, torch.bfloat16):
+                    fp16_params_this_group.append(p)
+                    master = p.detach().clone().float()
+                    master.requires_grad = True
+                    fp32_from_fp16_params_this_group.append(master)
+                    # optimizer state transfer if any
+                    if p in optimizer.state:
+                        optimizer.state[master] = optimizer.state.pop(p)
+                    new_params.append(master)
+                elif p.dtype == torch.float32:
+                    fp32_params_this_group.append(p)
+                    new_params.append(p)
+                else:
+                    raise TypeError(f"Unsupported param dtype {p.dtype}")
+            else:
+                new_params.append(p)
+        group["params"] = new_params
+        stash.fp16_groups.append(fp16_params_this_group)
+        stash.fp32_from_fp16_groups.append(fp32_from_fp16_params_this_group)
+        stash.fp32_from_fp32_groups.append(fp32_params_this_group)
+        stash.all_fp16_params += fp16_params_this_group
+        stash.all_fp32_from_fp16_params += fp32_from_fp16_params_this_group
+        stash.all_fp32_from_fp32_params += fp32_params_this_group
+
+    optimizer._amp_stash = stash
+    return optimizer
+
+
+def _materialize_master_grads(optimizer, scale):
+    """Unscale fp16 model grads into fp32 master .grad (one fused launch)."""
+    stash = optimizer._amp_stash
+    model_grads, master_params_with_grad = [], []
+    for p, master in zip(stash.all_fp16_params, stash.all_fp32_from_fp16_params):
+        if p.grad is not None:
+            model_grads.append(p.grad)
+            master_params_with_grad.append(master)
+    master_grads = []
+    for master in master_params_with_grad:
+        if master.grad is None:
+            master.grad = torch.empty_like(master)
+        master_grads.append(master.grad)
+    return model_grads, master_grads
+
+
+def _copy_master_to_model(optimizer):
+    stash = optimizer._amp_stash
+    if not stash.all_fp16_params:
+        return
+    device = stash.all_fp16_params[0].device
+    if device.type == "cuda":
+        amp_C = get_ext("amp_C")
+        overflow_buf = torch.zeros(1, dtype=torch.int32, device=device)
+        multi_tensor_applier(
+            amp_C.multi_tensor_scale, overflow_buf,
+            [stash.all_fp32_from_fp16_params, stash.all_fp16_params], 1.0,
+        )
+    else:
+        with torch.no_grad():
+            for master, p in zip(stash.all_fp32_from_fp16_params, stash.all_fp16_params):
+                p.copy_(master.to(p.dtype))
+
+
+def _patch_step_for_skip_and_copy(optimizer, needs_master_copy):
+    if getattr(optimizer, "_amp_step_patched", False):
+        return
+    old_step = optimizer.step
+
+    @functools.wraps(old_step)
+    def new_step(closure=None):
+        if getattr(optimizer, "_amp_skip_next_step", False):
+            optimizer._amp_skip_next_step = False
+            maybe_print("Gradient overflow. Skipping step.")
+            return None
+        out = old_step() if closure is None else old_step(closure)
+        if needs_master_copy and not getattr(optimizer, "_amp_handles_param_copy", False):
+            _copy_master_to_model(optimizer)
+        return out
+
+    optimizer.step = new_step
+    optimizer._amp_step_patched = True
+
+
+def initialize(
+    models,
+    optimizers=None,
+    enabled=True,
+    opt_level="O1",
+    cast_model_type=None,
+    patch_torch_functions=None,
+    keep_batchnorm_fp32=None,
+    master_weights=None,
+    loss_scale=None,
+    cast_model_outputs=None,
+    num_losses=1,
+    verbosity=1,
+    min_loss_scale=None,
+    max_loss_scale=2.0 ** 24,
+):
+    """Initialize amp. Returns (models, optimizers) with the same
+    list-or-single structure the caller passed (reference behavior)."""
+    _amp_state.verbosity = verbosity
+
+    models_was_list = isinstance(models, list)
+    optimizers_was_list = isinstance(optimizers, list)
+    model_list = models if models_was_list else [models]
+    if optimizers is None:
+        optimizer_list = []
+    else:
+        optimizer_list = optimizers if optimizers_was_list else [optimizers]
+
+    if not enabled:
+        _amp_state.initialized = True
+        _amp_state.opt_properties = OptProperties("O0", None, False, None, False, 1.0)
+        _amp_state.loss_scalers = [LossScaler(1.0) for _ in range(num_losses)]
+        _amp_state.optimizers = optimizer_list
+        _amp_state.models = model_list
+        return models, optimizers
+
+    if opt_level not in _OPT_LEVELS:
+        raise ValueError(f"Unexpected opt_level {opt_level}; options are 'O0', 'O1', 'O2', 'O3'")
+
+    props = dict(_OPT_LEVELS[opt_level])
+    # keep_batchnorm_fp32 may arrive as string "True"/"False" (reference CLI)
+    if isinstance(keep_batchnorm_fp32, str):
+        keep_batchnorm_fp32 = keep_batchnorm_fp32 == "True"
+    if cast_model_type is not None:
+        props["cast_model_type"] = cast_model_type
+    if patch_torch_functions is not None:
+        props["patch_torch_functions"] = patch_torch_functions
+    if keep_batchnorm_fp32 is not None:
+        props["keep_batchnorm_fp32"] = keep_batchnorm_fp32
+    if master_weights is not None:
+        props["master_weights"] = master_weights
+    if loss_scale is not None:
+        props["loss_scale"] = loss_scale if loss_scale == "dynamic" else float(loss_scale)
+
+    opt_properties = OptProperties(opt_level, props["cast_model_type"], props["patch_torch_functions"],
+                                   props["keep_batchnorm_fp32"], props["master_weights"], props["loss_scale"])
+
+    maybe_print(f"apex_amd.amp: opt_level={opt_level}, cast_model_type={props['cast_model_type']}, "
+                f"keep_batchnorm_fp32={props['keep_batchnorm_fp32']}, master_weights={props['master_weights']}, "
+                f"loss_scale={props['loss_scale']}")
+
+    # --- models ---
+    if props["cast_model_type"] is not None:  # O2 / O3
+        for model in model_list:
+            _cast_model(model, props["cast_model_type"], props["keep_batchnorm_fp32"])
+            _wrap_forward_input_cast(model, props["cast_model_type"])
+    elif props["patch_torch_functions"]:  # O1
+        cast_dtype = torch.float16 if cast_model_type is None else cast_model_type
+        for model in model_list:
+            _wrap_forward_autocast(model, cast_dtype)
+
+    # --- optimizers ---
+    if props["master_weights"]:
+        for opt in optimizer_list:
+            _process_optimizer_o2(opt, props["cast_model_type"])
+            if type(opt).__name__ == "FusedSGD":
+                opt._amp_handles_param_copy = getattr(opt, "materialize_master_grads", True) is not None and True
+                # FusedSGD's 4-list kernel writes the fp16 copy itself
+                opt._amp_handles_param_copy = True
+            _patch_step_for_skip_and_copy(opt, needs_master_copy=True)
+    else:
+        for opt in optimizer_list:
+            _patch_step_for_skip_and_copy(opt, needs_master_copy=False)
+
+    _amp_state.loss_scalers = [
+        LossScaler(props["loss_scale"], min_loss_scale=min_loss_scale, max_loss_scale=max_loss_scale)
+        for _ in range(num_losses)
+    ]
+    _amp_state.opt_properties = opt_properties
+    _amp_state.optimizers = optimizer_list
+    _amp_state.models = model_list
+    _amp_state.initialized = True
+
+    if optimizers is None:
+        return model_list if models_was_list else model_list[0]
+    return (
+        model_list if models_was_list else model_list[0],
+        optimizer_list if optimizers_was_list else optimizer_list[0],
+    )
+
+
+@contextlib.contextmanager
+def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
+    """Scale the loss; on exit unscale grads (into masters for O2), check
+    overflow, update the dynamic scale on-device, and flag step-skip."""
+    if not _amp_state.initialized:
+        raise RuntimeError("Invoked amp.scale_loss before amp.initialize.")
+
+    scaler = _amp_state.loss_scalers[loss_id]
+    loss_scale = scaler.loss_scale()
+
+    opt_list = optimizers if isinstance(optimizers, list) else [optimizers]
+
+    if loss_scale == 1.0 and not scaler.dynamic:
+        yield loss
+        # still materialize masters for O2 with static scale 1.0
+        if _amp_state.opt_properties.master_weights:
+            for opt in opt_list:
+                model_grads, master_grads = _materialize_master_grads(opt, 1.0)
+                scaler.unscale_grads(model_grads, master_grads)
+        return
+
+    yield loss.float() * loss_scale
+
+    if delay_unscale:
+        return
+
+    overflow = False
+    for opt in opt_list:
+        if hasattr(opt, "_amp_stash"):  # O2: unscale fp16 grads into masters
+            model_grads, master_grads = _materialize_master_grads(opt, loss_scale)
+            ov = scaler.unscale_grads(model_grads, master_grads)
+            # fp32 params' grads unscaled in place
+            stash = opt._amp_stash
+            fp32_grads = [p.grad for p in stash.all_fp32_from_fp32_params if p.grad is not None]
+            ov2 = scaler.unscale_grads(fp32_grads, fp32_grads, scale_override=loss_scale) if fp32_grads else False
+            overflow = overflow or ov or ov2
+        else:  # O0/O1: unscale in place
+            grads = [p.grad for p in itertools.chain(*[g["params"] for g in opt.param_groups]) if p.grad is not None]
+            ov = scaler.unscale_grads(grads, grads)
+            overflow = overflow or ov
+
+    if overflow:
+        for opt in opt_list:
+            opt._amp_skip_next_step = True
+
+
+def master_params(optimizer):
+    """Iterate over the fp32 master params (reference:
+    tests/distributed/amp_master_params/amp_master_params.py:71)."""
+    if hasattr(optimizer, "_amp_stash"):
+        stash = optimizer._amp_stash
+        for p in itertools.chain(stash.all_fp32_from_fp16_params, stash.all_fp32_from_fp32_params):
+            yield p
+    else:
+        for group in optimizer.param_groups:
+            for p in group["params"]:
+                yield p
+
+
+def state_dict(destination=None):
+    sd = destination if destination is not None else {}
+    for i, scaler in enumerate(_amp_state.loss_scalers):
+        sd[f"loss_scaler{i}"] = scaler.state_dict()
+    return sd
+
+
+def load_state_dict(state_dict):
+    for i, scaler in enumerate(_amp_state.loss_scalers):
+        key = f"loss_scaler{i}"
+        if key in state_dict:
+            scaler.load_state_dict(state_dict[key])

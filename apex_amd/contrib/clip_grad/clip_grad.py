@@ -1,0 +1,51 @@
+"""Fused gradient clipping (reference: apex/contrib/clip_grad/clip_grad.py:17-121).
+
+Drop-in for ``torch.nn.utils.clip_grad_norm_`` using one fused
+``multi_tensor_l2norm`` for the norm and one ``multi_tensor_scale`` for the
+clip — two launches for the whole model instead of per-tensor passes.
+"""
+
+import torch
+
+from ..._ext import get_ext
+from ...multi_tensor_apply import multi_tensor_applier
+
+
+def clip_grad_norm_(parameters, max_norm, norm_type=2.0, error_if_nonfinite=False):
+    if isinstance(parameters, torch.Tensor):
+        parameters = [parameters]
+    grads = [p.grad for p in parameters if p.grad is not None]
+    max_norm = float(max_norm)
+    norm_type = float(norm_type)
+    if len(grads) == 0:
+        return torch.tensor(0.0)
+    device = grads[0].device
+
+    if device.type == "cuda" and norm_type == 2.0 and all(g.device == device for g in grads):
+        amp_C = get_ext("amp_C")
+        noop = torch.zeros(1, dtype=torch.int32, device=device)
+        total_norm = multi_tensor_applier(amp_C.multi_tensor_l2norm, noop, [grads], False)[0].squeeze()
+    else:
+        if norm_type == float("inf"):
+            total_norm = max(g.detach().abs().max().to(device) for g in grads)
+        else:
+            total_norm = torch.norm(
+                torch.stack([torch.norm(g.detach(), norm_type).to(device) for g in grads]), norm_type
+            )
+
+    if error_if_nonfinite and torch.logical_or(total_norm.isnan(), total_norm.isinf()):
+        raise RuntimeError(
+            f"The total norm of order {norm_type} for gradients from `parameters` "
+            f"is non-finite, so it cannot be clipped."
+        )
+
+    clip_coef = max_norm / (total_norm + 1e-6)
+    if clip_coef < 1:
+        if device.type == "cuda" and norm_type == 2.0:
+            amp_C = get_ext("amp_C")
+            noop = torch.zeros(1, dtype=torch.int32, device=device)
+            multi_tensor_applier(amp_C.multi_tensor_scale, noop, [grads, grads], float(clip_coef))
+        else:
+            for g in grads:
+                g.detach().mul_(clip_coef.to(g.device))
+    return total_norm

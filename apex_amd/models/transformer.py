@@ -1,0 +1,140 @@
+"""Transformer language models built from apex_amd fused ops.
+
+These are the benchmark models for BASELINE configs #3-#5:
+- BERT-base encoder: FusedLayerNorm + scaled_masked_softmax + FusedAdam.
+- GPT-2 345M decoder: fused_dense (GEMM+bias+GELU) + FusedRMSNorm option +
+  scaled_upper_triang_masked_softmax + FusedLAMB.
+
+They exist to exercise the library the way Megatron-style trainers exercise
+the reference kernels; synthetic-data training only.
+"""
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from ..normalization import FusedLayerNorm, FusedRMSNorm
+from ..fused_dense import fused_dense_function, fused_dense_gelu_dense_function
+from ..transformer import scaled_masked_softmax, scaled_upper_triang_masked_softmax
+
+
+@dataclass
+class TransformerLMConfig:
+    vocab_size: int = 50304
+    hidden: int = 768
+    layers: int = 12
+    heads: int = 12
+    seq_len: int = 512
+    ffn_hidden: int = 0  # 0 → 4*hidden
+    causal: bool = False
+    norm: str = "layernorm"  # or "rmsnorm"
+
+    def __post_init__(self):
+        if self.ffn_hidden == 0:
+            self.ffn_hidden = 4 * self.hidden
+
+
+def bert_base_config(seq_len=512):
+    return TransformerLMConfig(vocab_size=30528, hidden=768, layers=12, heads=12,
+                               seq_len=seq_len, causal=False, norm="layernorm")
+
+
+def gpt2_345m_config(seq_len=1024):
+    return TransformerLMConfig(vocab_size=50304, hidden=1024, layers=24, heads=16,
+                               seq_len=seq_len, causal=True, norm="rmsnorm")
+
+
+class FusedSelfAttention(nn.Module):
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        self.h = cfg.hidden
+        self.nh = cfg.heads
+        self.hd = cfg.hidden // cfg.heads
+        self.causal = cfg.causal
+        self.qkv_w = nn.Parameter(torch.empty(3 * cfg.hidden, cfg.hidden))
+        self.qkv_b = nn.Parameter(torch.zeros(3 * cfg.hidden))
+        self.proj_w = nn.Parameter(torch.empty(cfg.hidden, cfg.hidden))
+        self.proj_b = nn.Parameter(torch.zeros(cfg.hidden))
+        nn.init.normal_(self.qkv_w, std=0.02)
+        nn.init.normal_(self.proj_w, std=0.02)
+
+    def forward(self, x, mask=None):
+        # x: [b, s, h]
+        b, s, h = x.shape
+        qkv = fused_dense_function(x, self.qkv_w, self.qkv_b)  # [b, s, 3h]
+        qkv = qkv.view(b, s, 3, self.nh, self.hd).permute(2, 0, 3, 1, 4)  # [3, b, nh, s, hd]
+        q, k, v = qkv[0], qkv[1], qkv[2]
+        scores = torch.matmul(q, k.transpose(-2, -1))  # [b, nh, s, s]
+        scale = 1.0 / math.sqrt(self.hd)
+        if self.causal:
+            probs = scaled_upper_triang_masked_softmax(scores.view(b * self.nh, s, s), scale)
+            probs = probs.view(b, self.nh, s, s)
+        else:
+            probs = scaled_masked_softmax(scores, mask, scale)
+        ctx = torch.matmul(probs, v)  # [b, nh, s, hd]
+        ctx = ctx.transpose(1, 2).reshape(b, s, h)
+        return fused_dense_function(ctx, self.proj_w, self.proj_b)
+
+
+class FusedMLPBlock(nn.Module):
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        self.w1 = nn.Parameter(torch.empty(cfg.ffn_hidden, cfg.hidden))
+        self.b1 = nn.Parameter(torch.zeros(cfg.ffn_hidden))
+        self.w2 = nn.Parameter(torch.empty(cfg.hidden, cfg.ffn_hidden))
+        self.b2 = nn.Parameter(torch.zeros(cfg.hidden))
+        nn.init.normal_(self.w1, std=0.02)
+        nn.init.normal_(self.w2, std=0.02)
+
+    def forward(self, x):
+        return fused_dense_gelu_dense_function(x, self.w1, self.b1, self.w2, self.b2)
+
+
+class TransformerLayer(nn.Module):
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        norm_cls = FusedLayerNorm if cfg.norm == "layernorm" else FusedRMSNorm
+        self.ln1 = norm_cls(cfg.hidden)
+        self.attn = FusedSelfAttention(cfg)
+        self.ln2 = norm_cls(cfg.hidden)
+        self.mlp = FusedMLPBlock(cfg)
+
+    def forward(self, x, mask=None):
+        x = x + self.attn(self.ln1(x), mask)
+        x = x + self.mlp(self.ln2(x))
+        return x
+
+
+class _TransformerLM(nn.Module):
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.pos_emb = nn.Embedding(cfg.seq_len, cfg.hidden)
+        self.layers = nn.ModuleList([TransformerLayer(cfg) for _ in range(cfg.layers)])
+        norm_cls = FusedLayerNorm if cfg.norm == "layernorm" else FusedRMSNorm
+        self.final_norm = norm_cls(cfg.hidden)
+        nn.init.normal_(self.tok_emb.weight, std=0.02)
+        nn.init.normal_(self.pos_emb.weight, std=0.02)
+
+    def forward(self, tokens, mask=None):
+        b, s = tokens.shape
+        pos = torch.arange(s, device=tokens.device).unsqueeze(0)
+        x = self.tok_emb(tokens) + self.pos_emb(pos)
+        for layer in self.layers:
+            x = layer(x, mask)
+        x = self.final_norm(x)
+        # weight-tied LM head
+        return torch.matmul(x, self.tok_emb.weight.t())
+
+
+class BertModel(_TransformerLM):
+    def __init__(self, cfg=None):
+        super().__init__(cfg or bert_base_config())
+
+
+class GPTModel(_TransformerLM):
+    def __init__(self, cfg=None):
+        super().__init__(cfg or gpt2_345m_config())

@@ -1,0 +1,10 @@
+from .distributed import DistributedDataParallel, flat_dist_call
+from .sync_batchnorm import SyncBatchNorm, convert_syncbn_model, create_syncbn_process_group
+
+__all__ = [
+    "DistributedDataParallel",
+    "SyncBatchNorm",
+    "convert_syncbn_model",
+    "create_syncbn_process_group",
+    "flat_dist_call",
+]

@@ -1,0 +1,130 @@
+"""amp O0/O1/O2 logic on CPU (tiny models, bf16 where casting needed)."""
+
+import torch
+import pytest
+
+from apex_amd import amp
+from apex_amd.amp._amp_state import _amp_state
+from apex_amd.optimizers import FusedSGD
+
+
+def make_model():
+    torch.manual_seed(0)
+    return torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 4))
+
+
+def setup_function(fn):
+    _amp_state.reset()
+
+
+def train_steps(model, opt, steps=5, seed=1):
+    torch.manual_seed(seed)
+    losses = []
+    for _ in range(steps):
+        x = torch.randn(8, 16)
+        if next(model.parameters()).dtype != torch.float32:
+            x = x.to(next(model.parameters()).dtype)
+        y = torch.randn(8, 4)
+        opt.zero_grad()
+        out = model(x)
+        loss = torch.nn.functional.mse_loss(out.float(), y)
+        with amp.scale_loss(loss, opt) as scaled:
+            scaled.backward()
+        opt.step()
+        losses.append(float(loss))
+    return losses
+
+
+def test_o0_noop():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O0", verbosity=0)
+    losses = train_steps(model, opt)
+    assert losses[-1] < losses[0]
+    assert next(model.parameters()).dtype == torch.float32
+
+
+def test_o2_master_weights_bf16():
+    model = make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    model, opt = amp.initialize(
+        model, opt, opt_level="O2", cast_model_type=torch.bfloat16, loss_scale=128.0, verbosity=0
+    )
+    assert next(model.parameters()).dtype == torch.bfloat16
+    assert hasattr(opt, "_amp_stash")
+    losses = train_steps(model, opt)
+    assert losses[-1] < losses[0]
+    # master params mirror the model params
+    masters = list(amp.master_params(opt))
+    models_fp16 = opt._amp_stash.all_fp16_params
+    assert len(masters) >= len(models_fp16)
+    for mp, p in zip(opt._amp_stash.all_fp32_from_fp16_params, models_fp16):
+        torch.testing.assert_close(mp.to(p.dtype), p)
+
+
+def test_o2_overflow_skips_step():
+    model = make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(
+        model, opt, opt_level="O2", cast_model_type=torch.bfloat16, loss_scale="dynamic", verbosity=0
+    )
+    before = [p.detach().clone() for p in amp.master_params(opt)]
+    scaler = _amp_state.loss_scalers[0]
+    scale_before = scaler.loss_scale()
+    x = torch.full((4, 16), 1e30, dtype=torch.bfloat16)
+    out = model(x)
+    loss = out.float().sum() * 1e30  # force inf grads
+    opt.zero_grad()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    opt.step()
+    after = list(amp.master_params(opt))
+    for b, a in zip(before, after):
+        torch.testing.assert_close(b, a)  # step was skipped
+    assert scaler.loss_scale() <= scale_before  # dynamic scale backed off
+
+
+def test_dynamic_scaler_growth():
+    from apex_amd.amp.scaler import LossScaler
+
+    s = LossScaler("dynamic", init_scale=2.0 ** 10, scale_window=3)
+    g = [torch.ones(4)]
+    for _ in range(3):
+        s.unscale_grads(g, g)
+    assert s.loss_scale() == 2.0 ** 11
+
+
+def test_o1_autocast_wrap():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O1", cast_model_type=torch.bfloat16, verbosity=0)
+    x = torch.randn(8, 16)
+    out = model(x)
+    assert out.dtype == torch.bfloat16  # autocast produced low-precision out
+    loss = out.float().sum()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    opt.step()
+
+
+def test_multi_loss_scalers():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O1", num_losses=3, verbosity=0)
+    assert len(_amp_state.loss_scalers) == 3
+    x = torch.randn(8, 16)
+    for loss_id in range(3):
+        opt.zero_grad()
+        loss = model(x).float().sum()
+        with amp.scale_loss(loss, opt, loss_id=loss_id) as scaled:
+            scaled.backward()
+        opt.step()
+
+
+def test_state_dict_roundtrip():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    amp.initialize(model, opt, opt_level="O1", verbosity=0)
+    sd = amp.state_dict()
+    assert "loss_scaler0" in sd
+    amp.load_state_dict(sd)

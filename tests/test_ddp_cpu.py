@@ -1,0 +1,90 @@
+"""apex_amd.parallel.DistributedDataParallel on gloo, world_size=2.
+
+Covers the reference DDP contracts (tests/distributed/DDP/
+ddp_race_condition_test.py): grad averaging under small message_size,
+delay_allreduce, gradient_predivide_factor, and trigger params.
+"""
+
+import torch
+import torch.distributed as dist
+import pytest
+
+from utils import run_distributed
+
+
+def _expected_avg_grad(rank_inputs, model_ctor):
+    """Compute the world-averaged grads by running each rank's batch."""
+    grads = None
+    for x in rank_inputs:
+        m = model_ctor()
+        out = m(x).sum()
+        out.backward()
+        g = [p.grad.clone() for p in m.parameters()]
+        grads = g if grads is None else [a + b for a, b in zip(grads, g)]
+    return [g / len(rank_inputs) for g in grads]
+
+
+def _model_ctor():
+    torch.manual_seed(42)
+    return torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Tanh(), torch.nn.Linear(16, 2))
+
+
+def _ddp_worker(rank, world_size, kwargs):
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _model_ctor()
+    ddp = DDP(model, **kwargs)
+    torch.manual_seed(7 + rank)
+    x = torch.randn(4, 8)
+
+    # gather each rank's input to compute the expected average locally
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x)
+
+    for _ in range(3):  # multiple iterations: hooks must re-arm each backward
+        ddp.zero_grad()
+        out = ddp(x).sum()
+        out.backward()
+
+    expected = _expected_avg_grad(xs, _model_ctor)
+    # grads after last backward (zero_grad each iter) should equal expected
+    actual = [p.grad for p in ddp.module.parameters()]
+    for a, e in zip(actual, expected):
+        torch.testing.assert_close(a, e, rtol=1e-5, atol=1e-6)
+
+
+def test_ddp_grad_average_default():
+    run_distributed(_ddp_worker, world_size=2, args=({},))
+
+
+def test_ddp_small_buckets():
+    run_distributed(_ddp_worker, world_size=2, args=({"message_size": 1},))
+
+
+def test_ddp_delay_allreduce():
+    run_distributed(_ddp_worker, world_size=2, args=({"delay_allreduce": True},))
+
+
+def test_ddp_predivide():
+    run_distributed(_ddp_worker, world_size=2, args=({"gradient_predivide_factor": 2.0},))
+
+
+def test_ddp_allreduce_always_fp32():
+    run_distributed(_ddp_worker, world_size=2, args=({"allreduce_always_fp32": True, "message_size": 1},))
+
+
+def _param_sync_worker(rank, world_size):
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    torch.manual_seed(rank * 999)  # deliberately different init per rank
+    model = torch.nn.Linear(4, 4)
+    DDP(model)
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    flats = [torch.empty_like(flat) for _ in range(world_size)]
+    dist.all_gather(flats, flat)
+    for f in flats[1:]:
+        torch.testing.assert_close(flats[0], f)
+
+
+def test_ddp_broadcasts_initial_params():
+    run_distributed(_param_sync_worker, world_size=2)

@@ -1,0 +1,108 @@
+"""SyncBatchNorm on gloo world_size=2: fused SBN must equal plain BN over
+the concatenated global batch (the reference contract from
+tests/distributed/synced_batchnorm/two_gpu_unit_test.py:83-190)."""
+
+import torch
+import torch.distributed as dist
+import pytest
+
+from utils import run_distributed
+
+
+def _syncbn_worker(rank, world_size, channel_last):
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(10 + rank)
+    C = 8
+    if channel_last:
+        x = torch.randn(4, 5, 5, C, requires_grad=True)
+    else:
+        x = torch.randn(4, C, 5, 5, requires_grad=True)
+
+    sbn = SyncBatchNorm(C, channel_last=channel_last)
+    sbn.train()
+
+    # gather the global batch for the reference BN
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x.detach())
+    if channel_last:
+        global_x = torch.cat(xs, dim=0).permute(0, 3, 1, 2).contiguous().requires_grad_(True)
+    else:
+        global_x = torch.cat(xs, dim=0).requires_grad_(True)
+
+    ref_bn = torch.nn.BatchNorm2d(C)
+    ref_bn.train()
+    with torch.no_grad():
+        ref_bn.weight.copy_(sbn.weight)
+        ref_bn.bias.copy_(sbn.bias)
+
+    out = sbn(x)
+    ref_out_global = ref_bn(global_x)
+    if channel_last:
+        ref_out = ref_out_global[rank * 4:(rank + 1) * 4].permute(0, 2, 3, 1)
+    else:
+        ref_out = ref_out_global[rank * 4:(rank + 1) * 4]
+    torch.testing.assert_close(out, ref_out, rtol=1e-4, atol=1e-5)
+
+    # running stats must match BN over the global batch
+    torch.testing.assert_close(sbn.running_mean, ref_bn.running_mean, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(sbn.running_var, ref_bn.running_var, rtol=1e-4, atol=1e-5)
+
+    # backward: same upstream grad everywhere
+    g = torch.ones_like(out)
+    out.backward(g)
+    ref_out_global.backward(torch.ones_like(ref_out_global))
+    if channel_last:
+        ref_gx = global_x.grad[rank * 4:(rank + 1) * 4].permute(0, 2, 3, 1)
+    else:
+        ref_gx = global_x.grad[rank * 4:(rank + 1) * 4]
+    torch.testing.assert_close(x.grad, ref_gx, rtol=1e-4, atol=1e-5)
+    # weight grad: local portion; allreduced weight grads should equal ref
+    wg = sbn.weight.grad.clone()
+    dist.all_reduce(wg)
+    torch.testing.assert_close(wg, ref_bn.weight.grad, rtol=1e-4, atol=1e-4)
+    bg = sbn.bias.grad.clone()
+    dist.all_reduce(bg)
+    torch.testing.assert_close(bg, ref_bn.bias.grad, rtol=1e-4, atol=1e-4)
+
+
+def test_syncbn_matches_global_bn():
+    run_distributed(_syncbn_worker, world_size=2, args=(False,))
+
+
+def test_syncbn_channel_last():
+    run_distributed(_syncbn_worker, world_size=2, args=(True,))
+
+
+def test_convert_syncbn_model():
+    from apex_amd.parallel import SyncBatchNorm, convert_syncbn_model
+
+    m = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 8, 3),
+        torch.nn.BatchNorm2d(8),
+        torch.nn.Sequential(torch.nn.BatchNorm2d(8)),
+    )
+    m2 = convert_syncbn_model(m)
+    assert isinstance(m2[1], SyncBatchNorm)
+    assert isinstance(m2[2][0], SyncBatchNorm)
+
+
+def test_syncbn_single_process_matches_bn():
+    """Without dist init, SyncBN == plain BN."""
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(0)
+    x = torch.randn(6, 4, 3, 3, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    sbn = SyncBatchNorm(4)
+    bn = torch.nn.BatchNorm2d(4)
+    sbn.train()
+    bn.train()
+    y1 = sbn(x)
+    y2 = bn(x2)
+    torch.testing.assert_close(y1, y2, rtol=1e-5, atol=1e-6)
+    y1.sum().backward()
+    y2.sum().backward()
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.running_mean, bn.running_mean)
+    torch.testing.assert_close(sbn.running_var, bn.running_var)
