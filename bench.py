@@ -1,0 +1,167 @@
+"""apex_amd flagship benchmark (driver contract — see BASELINE.json).
+
+Measures BOTH halves of the baseline metric on synthetic data /
+random-init weights:
+
+1. FusedAdam step time on 350M fp32 params (reported as
+   ``fusedadam_350m_ms`` inside the JSON line), measured on rank 0.
+2. ResNet-50 AMP training throughput (img/s) — the primary ``value`` —
+   with amp O1 bf16 + FusedSGD (+ SyncBatchNorm when the syncbn extension is
+   built) under apex_amd DDP over RCCL for N>1.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+The driver launches N>1 via torch.distributed.run with one rank per GPU.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def bench_fused_adam_350m(device, steps=20, warmup=5):
+    """Time FusedAdam.step() on ~350M fp32 params split into ~200 tensors."""
+    from apex_amd.optimizers import FusedAdam
+
+    torch.manual_seed(0)
+    n_tensors = 192
+    numel_each = 350_000_000 // n_tensors
+    params = [torch.empty(numel_each, device=device).normal_(0, 0.02).requires_grad_(True)
+              for _ in range(n_tensors)]
+    for p in params:
+        p.grad = torch.empty_like(p).normal_(0, 0.01)
+    opt = FusedAdam(params, lr=1e-3, weight_decay=0.01)
+    for _ in range(warmup):
+        opt.step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        opt.step()
+    torch.cuda.synchronize()
+    ms = (time.perf_counter() - t0) / steps * 1000.0
+    del params, opt
+    torch.cuda.empty_cache()
+    return ms
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=128, help="per-GPU batch size")
+    ap.add_argument("--image-size", type=int, default=224)
+    ap.add_argument("--skip-adam-bench", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group(backend="nccl")
+        torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+    torch.cuda.set_device(device)
+
+    from apex_amd import amp
+    from apex_amd._ext import has_ext
+    from apex_amd.models import resnet50
+    from apex_amd.optimizers import FusedSGD
+    from apex_amd.parallel import DistributedDataParallel as DDP
+    from apex_amd.parallel import convert_syncbn_model
+
+    # --- part 1: FusedAdam 350M step time (rank 0, 1 GPU) ---
+    adam_ms = None
+    if rank == 0 and not args.skip_adam_bench:
+        adam_ms = bench_fused_adam_350m(device, steps=max(10, args.steps // 2), warmup=args.warmup)
+
+    if distributed:
+        dist.barrier()
+
+    # --- part 2: ResNet-50 AMP img/s ---
+    torch.manual_seed(1234)
+    model = resnet50(num_classes=1000)
+    use_syncbn = has_ext("syncbn")
+    if use_syncbn:
+        model = convert_syncbn_model(model)
+    model = model.to(device)
+    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+    model, opt = amp.initialize(model, opt, opt_level="O1",
+                                cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
+    if distributed:
+        model = DDP(model, message_size=16_000_000)
+
+    x = torch.randn(args.batch, 3, args.image_size, args.image_size, device=device)
+    y = torch.randint(0, 1000, (args.batch,), device=device)
+    criterion = torch.nn.CrossEntropyLoss()
+
+    def step():
+        opt.zero_grad()
+        out = model(x)
+        loss = criterion(out.float(), y)
+        with amp.scale_loss(loss, opt) as scaled:
+            scaled.backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if distributed:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    imgs_per_s = world * args.batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "resnet50_amp_imgs_per_s",
+            "value": imgs_per_s,
+            "unit": "img/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "fusedadam_350m_ms": adam_ms,
+            "config": {
+                "model": "resnet50",
+                "global_batch": world * args.batch,
+                "image_size": args.image_size,
+                "amp": "O1-bf16",
+                "optimizer": "FusedSGD(momentum=0.9)",
+                "syncbn": use_syncbn,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,76 @@
+"""Build apex_amd's in-tree HIP extensions for MI355X (gfx950).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Every extension is built into the package directory (apex_amd/_*.so) so the
+snapshot that travels to a GPU box carries the binaries. No JIT cache, no
+site-packages install. Mirrors the reference's per-extension registry
+(setup.py:165-845) but with plain HIP sources.
+"""
+
+import os
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+COMMON_FLAGS = ["-O3", "-std=c++17"]
+HIP_FLAGS = ["-O3", "-std=c++17"]
+
+
+def hip_ext(name, sources, extra_hip_flags=None, libraries=None, extra_link=None):
+    return CUDAExtension(
+        name=name,
+        sources=sources,
+        extra_compile_args={
+            "cxx": COMMON_FLAGS,
+            "nvcc": HIP_FLAGS + (extra_hip_flags or []),
+        },
+        libraries=libraries or [],
+        extra_link_args=extra_link or [],
+    )
+
+
+ext_modules = [
+    CppExtension(
+        name="apex_amd._apex_C",
+        sources=["csrc/flatten_unflatten.cpp"],
+        extra_compile_args={"cxx": COMMON_FLAGS},
+    ),
+    hip_ext(
+        "apex_amd._amp_C",
+        [
+            "csrc/amp_C_frontend.cpp",
+            "csrc/multi_tensor_elementwise.hip",
+            "csrc/multi_tensor_l2norm.hip",
+            "csrc/multi_tensor_sgd.hip",
+            "csrc/multi_tensor_adam.hip",
+            "csrc/multi_tensor_opt.hip",
+            "csrc/multi_tensor_lamb.hip",
+            "csrc/update_scale_hysteresis.hip",
+        ],
+    ),
+]
+
+setup(
+    name="apex_amd",
+    version="0.1.0",
+    description="MI355X-native mixed-precision and fused-kernel training library",
+    packages=[
+        "apex_amd",
+        "apex_amd.amp",
+        "apex_amd.contrib",
+        "apex_amd.contrib.clip_grad",
+        "apex_amd.fused_dense",
+        "apex_amd.mlp",
+        "apex_amd.models",
+        "apex_amd.multi_tensor_apply",
+        "apex_amd.normalization",
+        "apex_amd.optimizers",
+        "apex_amd.parallel",
+        "apex_amd.transformer",
+    ],
+    ext_modules=ext_modules,
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)
