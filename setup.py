@@ -51,6 +51,8 @@ ext_modules = [
             "csrc/update_scale_hysteresis.hip",
         ],
     ),
+    hip_ext("apex_amd._fused_norm", ["csrc/fused_norm.hip"]),
+    hip_ext("apex_amd._syncbn", ["csrc/syncbn.hip"]),
 ]
 
 setup(
