@@ -1,0 +1,139 @@
+// apex_amd._fused_dense — GEMM+bias(+GELU) with hipBLASLt epilogue fusion,
+// plus the fused weight-gradient accumulation GEMM.
+// Reference surface: csrc/fused_dense.cpp:161-167 (linear_bias_forward/
+// backward, linear_gelu_linear_forward/backward) and
+// csrc/megatron/fused_weight_gradient_dense.cpp:11-13 (wgrad_gemm_accum_*).
+#include "lt_gemm.h"
+
+#include <vector>
+
+namespace {
+
+at::Tensor flat2d(const at::Tensor& t) {
+  return t.contiguous().reshape({-1, t.size(-1)});
+}
+
+}  // namespace
+
+at::Tensor linear_bias_forward(at::Tensor input, at::Tensor weight, at::Tensor bias) {
+  auto x = flat2d(input);
+  auto w = weight.contiguous();
+  auto b = bias.contiguous();
+  auto out = at::empty({x.size(0), w.size(0)}, x.options());
+  lt_linear(x, w, out, &b, HIPBLASLT_EPILOGUE_BIAS, nullptr);
+  auto sizes = input.sizes().vec();
+  sizes.back() = w.size(0);
+  return out.reshape(sizes);
+}
+
+at::Tensor linear_forward(at::Tensor input, at::Tensor weight) {
+  auto x = flat2d(input);
+  auto w = weight.contiguous();
+  auto out = at::empty({x.size(0), w.size(0)}, x.options());
+  lt_linear(x, w, out, nullptr, HIPBLASLT_EPILOGUE_DEFAULT, nullptr);
+  auto sizes = input.sizes().vec();
+  sizes.back() = w.size(0);
+  return out.reshape(sizes);
+}
+
+std::vector<at::Tensor> linear_bias_backward(at::Tensor input, at::Tensor weight,
+                                             at::Tensor grad_output) {
+  auto x = flat2d(input);
+  auto w = weight.contiguous();
+  auto dy = flat2d(grad_output);
+  auto dx = at::empty_like(x);
+  auto dw = at::empty_like(w);
+  auto db = at::empty({w.size(0)}, w.options());
+  // dgrad: dX = dY @ W
+  lt_linear_dgrad(dy, w, dx, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr);
+  // wgrad + fused bias grad: dW = dY^T @ X, db = colsum(dY)
+  lt_linear_wgrad(x, dy, dw, HIPBLASLT_EPILOGUE_BGRADB, &db, 0.f);
+  return {dx.reshape(input.sizes()), dw, db};
+}
+
+std::vector<at::Tensor> linear_backward(at::Tensor input, at::Tensor weight,
+                                        at::Tensor grad_output) {
+  auto x = flat2d(input);
+  auto w = weight.contiguous();
+  auto dy = flat2d(grad_output);
+  auto dx = at::empty_like(x);
+  auto dw = at::empty_like(w);
+  lt_linear_dgrad(dy, w, dx, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr);
+  lt_linear_wgrad(x, dy, dw, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, 0.f);
+  return {dx.reshape(input.sizes()), dw};
+}
+
+std::vector<at::Tensor> linear_gelu_linear_forward(at::Tensor input, at::Tensor weight1,
+                                                   at::Tensor bias1, at::Tensor weight2,
+                                                   at::Tensor bias2) {
+  auto x = flat2d(input);
+  auto w1 = weight1.contiguous();
+  auto b1 = bias1.contiguous();
+  auto w2 = weight2.contiguous();
+  auto b2 = bias2.contiguous();
+  const long m = x.size(0), n1 = w1.size(0), n2 = w2.size(0);
+  auto output1 = at::empty({m, n1}, x.options());   // GELU(X@W1^T + b1)
+  auto gelu_in = at::empty({m, n1}, x.options());   // pre-GELU (post-bias) aux
+  auto output2 = at::empty({m, n2}, x.options());
+  lt_linear(x, w1, output1, &b1, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, &gelu_in);
+  lt_linear(output1, w2, output2, &b2, HIPBLASLT_EPILOGUE_BIAS, nullptr);
+  auto sizes = input.sizes().vec();
+  sizes.back() = n2;
+  return {output1, output2.reshape(sizes), gelu_in};
+}
+
+std::vector<at::Tensor> linear_gelu_linear_backward(at::Tensor input, at::Tensor gelu_in,
+                                                    at::Tensor output1, at::Tensor weight1,
+                                                    at::Tensor weight2, at::Tensor grad_output) {
+  auto x = flat2d(input);
+  auto o1 = flat2d(output1);
+  auto gi = flat2d(gelu_in);
+  auto w1 = weight1.contiguous();
+  auto w2 = weight2.contiguous();
+  auto dy2 = flat2d(grad_output);
+  const long m = x.size(0), n1 = w1.size(0);
+
+  auto dw2 = at::empty_like(w2);
+  auto db2 = at::empty({w2.size(0)}, w2.options());
+  lt_linear_wgrad(o1, dy2, dw2, HIPBLASLT_EPILOGUE_BGRADB, &db2, 0.f);
+
+  // d_gelu = dGELU(dY2 @ W2, gelu_in), db1 fused
+  auto d_gelu = at::empty({m, n1}, x.options());
+  auto db1 = at::empty({n1}, w1.options());
+  lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DGELU_BGRAD, &gi, &db1);
+
+  auto dw1 = at::empty_like(w1);
+  lt_linear_wgrad(x, d_gelu, dw1, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, 0.f);
+  auto dx = at::empty_like(x);
+  lt_linear_dgrad(d_gelu, w1, dx, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr);
+  return {dx.reshape(input.sizes()), dw1, db1, dw2, db2};
+}
+
+// main_grad[n,k] += grad_output[m,n]^T @ input[m,k] — fp32 accumulate,
+// beta=1 so the result lands in the persistent main-grad buffer.
+void wgrad_gemm_accum_fp32(at::Tensor input, at::Tensor grad_output, at::Tensor main_grad) {
+  auto x = flat2d(input);
+  auto dy = flat2d(grad_output);
+  TORCH_CHECK(main_grad.scalar_type() == at::ScalarType::Float, "main_grad must be fp32");
+  lt_linear_wgrad(x, dy, main_grad, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, 1.f);
+}
+
+void wgrad_gemm_accum_fp16(at::Tensor input, at::Tensor grad_output, at::Tensor main_grad) {
+  auto x = flat2d(input);
+  auto dy = flat2d(grad_output);
+  TORCH_CHECK(main_grad.scalar_type() == at::ScalarType::Half ||
+                  main_grad.scalar_type() == at::ScalarType::BFloat16,
+              "main_grad must be fp16/bf16");
+  lt_linear_wgrad(x, dy, main_grad, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, 1.f);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("linear_bias_forward", &linear_bias_forward);
+  m.def("linear_forward", &linear_forward);
+  m.def("linear_bias_backward", &linear_bias_backward);
+  m.def("linear_backward", &linear_backward);
+  m.def("linear_gelu_linear_forward", &linear_gelu_linear_forward);
+  m.def("linear_gelu_linear_backward", &linear_gelu_linear_backward);
+  m.def("wgrad_gemm_accum_fp32", &wgrad_gemm_accum_fp32);
+  m.def("wgrad_gemm_accum_fp16", &wgrad_gemm_accum_fp16);
+}

@@ -53,6 +53,10 @@ ext_modules = [
     ),
     hip_ext("apex_amd._fused_norm", ["csrc/fused_norm.hip"]),
     hip_ext("apex_amd._syncbn", ["csrc/syncbn.hip"]),
+    hip_ext("apex_amd._softmax", ["csrc/softmax.hip"]),
+    hip_ext("apex_amd._rope", ["csrc/rope.hip"]),
+    hip_ext("apex_amd._fused_dense", ["csrc/fused_dense.hip"], libraries=["hipblaslt"]),
+    hip_ext("apex_amd._mlp", ["csrc/mlp.hip"], libraries=["hipblaslt"]),
 ]
 
 setup(

@@ -1,0 +1,253 @@
+"""GPU numerics for the softmax family, RoPE, fused_dense/mlp, and wgrad
+vs fp32 torch references."""
+
+import math
+
+import torch
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+TOL16 = dict(rtol=1e-3, atol=1e-3)
+TOLBF = dict(rtol=1.6e-2, atol=1.6e-2)
+TOL32 = dict(rtol=1e-5, atol=1e-5)
+
+
+def tol_for(dtype):
+    return {torch.float32: TOL32, torch.float16: TOL16, torch.bfloat16: TOLBF}[dtype]
+
+
+# ---------------- softmax ----------------
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16, torch.float32])
+@pytest.mark.parametrize("sk", [128, 511, 2048])
+def test_scaled_softmax(dtype, sk):
+    from apex_amd.transformer import scaled_softmax
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 4, 32, sk, device="cuda", dtype=dtype, requires_grad=True)
+    xr = x.detach().float().clone().requires_grad_(True)
+    scale = 0.7
+    y = scaled_softmax(x, scale)
+    y_ref = torch.softmax(xr * scale, dim=-1)
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    torch.testing.assert_close(x.grad.float(), xr.grad, **tol_for(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_scaled_masked_softmax(dtype):
+    from apex_amd.transformer import scaled_masked_softmax
+
+    torch.manual_seed(1)
+    b, np_, sq, sk = 2, 4, 33, 257
+    x = torch.randn(b, np_, sq, sk, device="cuda", dtype=dtype, requires_grad=True)
+    mask = torch.randint(0, 2, (b, 1, sq, sk), device="cuda", dtype=torch.bool)
+    mask[..., 0] = False
+    xr = x.detach().float().clone().requires_grad_(True)
+    scale = 1.0 / math.sqrt(64)
+    y = scaled_masked_softmax(x, mask, scale)
+    y_ref = torch.softmax((xr * scale).masked_fill(mask, -10000.0), dim=-1)
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    torch.testing.assert_close(x.grad.float(), xr.grad, **tol_for(dtype))
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_scaled_upper_triang_masked_softmax(dtype):
+    from apex_amd.transformer import scaled_upper_triang_masked_softmax
+
+    torch.manual_seed(2)
+    ab, sq = 8, 129
+    x = torch.randn(ab, sq, sq, device="cuda", dtype=dtype, requires_grad=True)
+    xr = x.detach().float().clone().requires_grad_(True)
+    scale = 0.5
+    y = scaled_upper_triang_masked_softmax(x, scale)
+    mask = torch.triu(torch.ones(sq, sq, device="cuda", dtype=torch.bool), diagonal=1)
+    y_ref = torch.softmax((xr * scale).masked_fill(mask, float("-inf")), dim=-1)
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    torch.testing.assert_close(x.grad.float(), xr.grad, **tol_for(dtype))
+
+
+def test_generic_scaled_masked_softmax_large_sk():
+    from apex_amd.transformer import generic_scaled_masked_softmax
+
+    torch.manual_seed(3)
+    # beyond the reference's 16K warp-kernel ceiling
+    x = torch.randn(1, 2, 4, 20000, device="cuda", dtype=torch.bfloat16)
+    mask = torch.zeros(1, 1, 4, 20000, device="cuda", dtype=torch.bool)
+    y = generic_scaled_masked_softmax(x, mask, 1.0)
+    y_ref = torch.softmax(x.float(), dim=-1)
+    torch.testing.assert_close(y.float(), y_ref, **TOLBF)
+
+
+# ---------------- rope ----------------
+def _make_freqs(s, d2, device="cuda"):
+    inv_freq = 1.0 / (10000 ** (torch.arange(0, d2, 2, device=device).float() / d2))
+    t = torch.arange(s, device=device).float()
+    freqs = torch.einsum("s,f->sf", t, inv_freq)
+    return torch.cat([freqs, freqs], dim=-1).view(s, 1, 1, d2)
+
+
+def _rotate_half(x):
+    x1, x2 = torch.chunk(x, 2, dim=-1)
+    return torch.cat((-x2, x1), dim=-1)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("rot_frac", [1.0, 0.5])
+def test_rope_sbhd_gpu(dtype, rot_frac):
+    from apex_amd.transformer import fused_apply_rotary_pos_emb
+
+    torch.manual_seed(4)
+    s, b, h, d = 33, 2, 4, 64
+    d2 = int(d * rot_frac)
+    t = torch.randn(s, b, h, d, device="cuda", dtype=dtype, requires_grad=True)
+    freqs = _make_freqs(s, d2)
+    y = fused_apply_rotary_pos_emb(t, freqs)
+    tr = t.detach().float()
+    t_rot, t_pass = tr[..., :d2], tr[..., d2:]
+    ref = torch.cat([t_rot * freqs.cos() + _rotate_half(t_rot) * freqs.sin(), t_pass], dim=-1)
+    torch.testing.assert_close(y.float(), ref, **tol_for(dtype))
+    # backward: orthogonality — rope_bwd(rope_fwd(g)) recovers magnitude
+    g = torch.randn_like(y)
+    y.backward(g)
+    t2 = tr.clone().requires_grad_(True)
+    t_rot2, t_pass2 = t2[..., :d2], t2[..., d2:]
+    ref2 = torch.cat([t_rot2 * freqs.cos() + _rotate_half(t_rot2) * freqs.sin(), t_pass2], -1)
+    ref2.backward(g.float())
+    torch.testing.assert_close(t.grad.float(), t2.grad, **tol_for(dtype))
+
+
+def test_rope_cached_gpu():
+    from apex_amd.transformer import fused_apply_rotary_pos_emb, fused_apply_rotary_pos_emb_cached
+
+    torch.manual_seed(5)
+    s, b, h, d = 16, 2, 2, 32
+    t = torch.randn(s, b, h, d, device="cuda")
+    freqs = _make_freqs(s, d)
+    y1 = fused_apply_rotary_pos_emb(t, freqs)
+    y2 = fused_apply_rotary_pos_emb_cached(t, freqs.cos(), freqs.sin())
+    torch.testing.assert_close(y1, y2, rtol=1e-5, atol=1e-5)
+
+
+def test_rope_thd_gpu():
+    from apex_amd.transformer import fused_apply_rotary_pos_emb, fused_apply_rotary_pos_emb_thd
+
+    torch.manual_seed(6)
+    h, d = 4, 32
+    seqlens = [5, 11, 3]
+    cu = torch.tensor([0, 5, 16, 19], dtype=torch.int32, device="cuda")
+    total = sum(seqlens)
+    t = torch.randn(total, h, d, device="cuda")
+    freqs = _make_freqs(max(seqlens), d)
+    y = fused_apply_rotary_pos_emb_thd(t, cu, freqs)
+    # reference: per-sequence sbhd with b=1
+    outs = []
+    for i, L in enumerate(seqlens):
+        seg = t[int(cu[i]):int(cu[i + 1])].unsqueeze(1)
+        outs.append(fused_apply_rotary_pos_emb(seg, freqs[:L]).squeeze(1))
+    ref = torch.cat(outs, 0)
+    torch.testing.assert_close(y, ref, rtol=1e-5, atol=1e-5)
+
+
+# ---------------- fused_dense / mlp / wgrad ----------------
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16, torch.float32])
+def test_fused_dense_gpu(dtype):
+    from apex_amd.fused_dense import FusedDense
+
+    torch.manual_seed(7)
+    fd = FusedDense(128, 96).cuda().to(dtype)
+    x = torch.randn(64, 128, device="cuda", dtype=dtype, requires_grad=True)
+    xr = x.detach().float().clone().requires_grad_(True)
+    wr = fd.weight.detach().float().clone().requires_grad_(True)
+    br = fd.bias.detach().float().clone().requires_grad_(True)
+    y = fd(x)
+    y_ref = torch.nn.functional.linear(xr, wr, br)
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    torch.testing.assert_close(x.grad.float(), xr.grad, **tol_for(dtype))
+    wtol = {k: v * 8 for k, v in tol_for(dtype).items()}
+    torch.testing.assert_close(fd.weight.grad.float(), wr.grad, **wtol)
+    torch.testing.assert_close(fd.bias.grad.float(), br.grad, **wtol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_fused_dense_gelu_dense_gpu(dtype):
+    from apex_amd.fused_dense import FusedDenseGeluDense
+
+    torch.manual_seed(8)
+    m = FusedDenseGeluDense(64, 256, 48).cuda().to(dtype)
+    x = torch.randn(32, 64, device="cuda", dtype=dtype, requires_grad=True)
+    xr = x.detach().float().clone().requires_grad_(True)
+    w1 = m.weight1.detach().float().clone().requires_grad_(True)
+    b1 = m.bias1.detach().float().clone().requires_grad_(True)
+    w2 = m.weight2.detach().float().clone().requires_grad_(True)
+    b2 = m.bias2.detach().float().clone().requires_grad_(True)
+    y = m(x)
+    y_ref = torch.nn.functional.linear(
+        torch.nn.functional.gelu(torch.nn.functional.linear(xr, w1, b1)), w2, b2
+    )
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    wtol = {k: v * 16 for k, v in tol_for(dtype).items()}
+    torch.testing.assert_close(x.grad.float(), xr.grad, **wtol)
+    torch.testing.assert_close(m.weight1.grad.float(), w1.grad, **wtol)
+    torch.testing.assert_close(m.bias1.grad.float(), b1.grad, **wtol)
+    torch.testing.assert_close(m.weight2.grad.float(), w2.grad, **wtol)
+    torch.testing.assert_close(m.bias2.grad.float(), b2.grad, **wtol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
+def test_mlp_gpu(dtype):
+    from apex_amd.mlp import MLP
+
+    torch.manual_seed(9)
+    sizes = [80, 128, 96, 32]
+    mlp = MLP(sizes, activation="relu").cuda().to(dtype)
+    x = torch.randn(64, 80, device="cuda", dtype=dtype, requires_grad=True)
+    # fp32 reference
+    layers = []
+    for i in range(mlp.num_layers):
+        lin = torch.nn.Linear(sizes[i], sizes[i + 1]).cuda()
+        with torch.no_grad():
+            lin.weight.copy_(mlp.weights[i].float())
+            lin.bias.copy_(mlp.biases[i].float())
+        layers += [lin, torch.nn.ReLU()]
+    ref = torch.nn.Sequential(*layers)
+    xr = x.detach().float().clone().requires_grad_(True)
+    y = mlp(x)
+    y_ref = ref(xr)
+    torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g.float())
+    wtol = {k: v * 8 for k, v in tol_for(dtype).items()}
+    torch.testing.assert_close(x.grad.float(), xr.grad, **wtol)
+    for i in range(mlp.num_layers):
+        torch.testing.assert_close(mlp.weights[i].grad.float(), ref[2 * i].weight.grad, **wtol)
+        torch.testing.assert_close(mlp.biases[i].grad.float(), ref[2 * i].bias.grad, **wtol)
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_wgrad_gemm_accum_fp32_gpu(dtype):
+    from apex_amd.transformer import wgrad_gemm_accum_fp32
+
+    torch.manual_seed(10)
+    x = torch.randn(256, 64, device="cuda", dtype=dtype)
+    dy = torch.randn(256, 96, device="cuda", dtype=dtype)
+    main = torch.randn(96, 64, device="cuda", dtype=torch.float32)
+    expected = main + (dy.float().t() @ x.float())
+    wgrad_gemm_accum_fp32(x, dy, main)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(main, expected, rtol=1e-2, atol=1e-2)
