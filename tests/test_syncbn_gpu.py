@@ -71,8 +71,10 @@ def test_syncbn_module_matches_bn_gpu(dtype):
     y.backward(g.to(y.dtype))
     y_ref.backward(g)
     torch.testing.assert_close(x.grad.float(), x2.grad, **tol)
-    torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, rtol=1e-3, atol=1e-3)
-    torch.testing.assert_close(sbn.bias.grad, bn.bias.grad, rtol=1e-3, atol=1e-3)
+    # weight/bias grads accumulate 1568 low-precision products per channel
+    wtol = dict(rtol=1e-3, atol=1e-3) if dtype == torch.float32 else dict(rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, **wtol)
+    torch.testing.assert_close(sbn.bias.grad, bn.bias.grad, **wtol)
     torch.testing.assert_close(sbn.running_mean, bn.running_mean, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(sbn.running_var, bn.running_var, rtol=1e-4, atol=1e-5)
 
