@@ -57,6 +57,9 @@ ext_modules = [
     hip_ext("apex_amd._rope", ["csrc/rope.hip"]),
     hip_ext("apex_amd._fused_dense", ["csrc/fused_dense.hip"], libraries=["hipblaslt"]),
     hip_ext("apex_amd._mlp", ["csrc/mlp.hip"], libraries=["hipblaslt"]),
+    hip_ext("apex_amd._xentropy", ["csrc/xentropy.hip"]),
+    hip_ext("apex_amd._focal_loss", ["csrc/focal_loss.hip"]),
+    hip_ext("apex_amd._index_mul_2d", ["csrc/index_mul_2d.hip"]),
 ]
 
 setup(
@@ -68,6 +71,9 @@ setup(
         "apex_amd.amp",
         "apex_amd.contrib",
         "apex_amd.contrib.clip_grad",
+        "apex_amd.contrib.xentropy",
+        "apex_amd.contrib.focal_loss",
+        "apex_amd.contrib.index_mul_2d",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",
