@@ -1,0 +1,355 @@
+"""DistributedFusedAdam — ZeRO-2 sharded Adam/AdamW over RCCL/xGMI.
+
+API and behavior parity with the reference
+``apex.contrib.optimizers.DistributedFusedAdam``
+(apex/contrib/optimizers/distributed_fused_adam.py:270-3488): params are
+flattened into fixed-capacity buckets; gradients are reduce-scattered as
+backward produces them (overlapped on a side stream), optimizer state is
+sharded per rank, and updated param shards are all-gathered back. Key
+differences from a line-by-line port, for MI355X:
+
+* bucket capacity defaults to 64 MB — xGMI ring collectives are per-link
+  bound (7 p2p links x ~153 GB/s), so buckets must be big enough to amortize
+  per-collective latency but small enough to overlap with backward.
+* model params are re-bound as views into the flat param bucket, so the
+  trailing all_gather writes directly into the working weights (no separate
+  copy pass).
+* the local shard step is one ``multi_tensor_adam`` launch per bucket group
+  (the same gfx950 kernel as FusedAdam).
+* on gloo (CPU CI) reduce_scatter/all_gather_into_tensor fall back to
+  all_reduce / all_gather-list — the sharding logic is identical.
+"""
+
+import math
+from collections import defaultdict
+
+import torch
+import torch.distributed as dist
+
+from ..._ext import get_ext
+from ...multi_tensor_apply import multi_tensor_applier
+
+
+def _backend_supports_rs(group):
+    try:
+        return dist.get_backend(group) == "nccl"
+    except Exception:
+        return False
+
+
+class _Bucket:
+    def __init__(self, numel, dtype, grad_dtype, device, world_size):
+        # pad to a multiple of world_size
+        self.numel_unpadded = numel
+        self.numel = ((numel + world_size - 1) // world_size) * world_size
+        self.shard_size = self.numel // world_size
+        self.params = []          # (param, offset)
+        self.filled = 0
+        self.param_data = torch.zeros(self.numel, dtype=dtype, device=device)
+        self.grad_data = torch.zeros(self.numel, dtype=grad_dtype, device=device)
+        # sharded fp32 state (created lazily once param values are final)
+        self.master_shard = None
+        self.exp_avg = None
+        self.exp_avg_sq = None
+        self.grad_shard = None
+        self.ready_params = set()
+        self.sync_work = None
+        self.synced = False
+
+
+class DistributedFusedAdam(torch.optim.Optimizer):
+    """ZeRO-2 Adam. Supported reference knobs: lr, bias_correction, betas,
+    eps, weight_decay, adam_w_mode, bucket_cap_mb, overlap_grad_sync,
+    grad_sync_dtype, process_group, set_grad_none, average_grad_sync.
+    """
+
+    def __init__(
+        self,
+        params,
+        lr=1e-3,
+        bias_correction=True,
+        betas=(0.9, 0.999),
+        eps=1e-8,
+        adam_w_mode=True,
+        weight_decay=0.0,
+        amsgrad=False,
+        bucket_cap_mb=64,
+        overlap_grad_sync=True,
+        average_grad_sync=True,
+        grad_sync_dtype=None,
+        process_group=None,
+        set_grad_none=True,
+    ):
+        if amsgrad:
+            raise RuntimeError("DistributedFusedAdam does not support AMSGrad")
+        defaults = dict(lr=lr, bias_correction=bias_correction, betas=betas, eps=eps,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+        self.adam_w_mode = 1 if adam_w_mode else 0
+        self.set_grad_none = set_grad_none
+        self.overlap_grad_sync = overlap_grad_sync
+        self.average_grad_sync = average_grad_sync
+        self.process_group = process_group
+        self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
+        self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
+        self._step = 0
+
+        device = self.param_groups[0]["params"][0].device
+        self.device = device
+        self._use_stream = device.type == "cuda"
+        self._comm_stream = torch.cuda.Stream() if self._use_stream else None
+        self._noop = torch.zeros(1, dtype=torch.int32, device=device)
+
+        self._build_buckets(grad_sync_dtype)
+        self._register_hooks()
+
+    # ---------- setup ----------
+    def _build_buckets(self, grad_sync_dtype):
+        self.buckets = []
+        self.param_to_bucket = {}
+        for group in self.param_groups:
+            by_dtype = defaultdict(list)
+            for p in group["params"]:
+                if p.requires_grad:
+                    by_dtype[p.dtype].append(p)
+            for dtype, plist in by_dtype.items():
+                gdtype = grad_sync_dtype or dtype
+                cap_elems = max(self.bucket_cap // dtype.itemsize, self.world_size)
+                cur = []
+                cur_numel = 0
+                for p in plist:
+                    if cur and cur_numel + p.numel() > cap_elems:
+                        self._finalize_bucket(cur, dtype, gdtype, group)
+                        cur, cur_numel = [], 0
+                    cur.append(p)
+                    cur_numel += p.numel()
+                if cur:
+                    self._finalize_bucket(cur, dtype, gdtype, group)
+
+    def _finalize_bucket(self, plist, dtype, grad_dtype, group):
+        numel = sum(p.numel() for p in plist)
+        b = _Bucket(numel, dtype, grad_dtype, self.device, self.world_size)
+        b.group = group
+        offset = 0
+        for p in plist:
+            n = p.numel()
+            b.param_data[offset:offset + n].copy_(p.detach().reshape(-1))
+            # re-bind the param as a view into the bucket so all_gather of the
+            # updated shards lands directly in the working weights
+            p.data = b.param_data[offset:offset + n].view_as(p)
+            b.params.append((p, offset))
+            self.param_to_bucket[p] = (b, offset)
+            offset += n
+        lo = self.rank * b.shard_size
+        hi = lo + b.shard_size
+        b.master_shard = b.param_data[lo:hi].float().clone()
+        b.exp_avg = torch.zeros_like(b.master_shard)
+        b.exp_avg_sq = torch.zeros_like(b.master_shard)
+        b.grad_shard = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
+        self.buckets.append(b)
+
+    def _register_hooks(self):
+        self._hook_handles = []
+        for b in self.buckets:
+            for p, offset in b.params:
+                handle = p.register_post_accumulate_grad_hook(self._make_hook(p))
+                self._hook_handles.append(handle)
+
+    def _make_hook(self, p):
+        def hook(param):
+            self._grad_copy(p)
+            if self.overlap_grad_sync:
+                b, _ = self.param_to_bucket[p]
+                if len(b.ready_params) == len(b.params) and not b.synced and b.sync_work is None:
+                    self._start_bucket_grad_sync(b)
+
+        return hook
+
+    def _grad_copy(self, p):
+        b, offset = self.param_to_bucket[p]
+        if p.grad is not None:
+            n = p.numel()
+            b.grad_data[offset:offset + n].add_(p.grad.detach().reshape(-1).to(b.grad_data.dtype))
+            p.grad = None
+        b.ready_params.add(p)
+
+    # ---------- grad sync ----------
+    def _start_bucket_grad_sync(self, b):
+        if self.world_size == 1:
+            b.grad_shard.copy_(b.grad_data.float())
+            b.synced = True
+            return
+        if self._use_stream:
+            self._comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._comm_stream):
+                self._issue_grad_collective(b)
+        else:
+            self._issue_grad_collective(b)
+
+    def _issue_grad_collective(self, b):
+        if self.average_grad_sync:
+            b.grad_data.div_(self.world_size)
+        if _backend_supports_rs(self.process_group):
+            shard = b.grad_data[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+            b.sync_work = dist.reduce_scatter_tensor(
+                shard, b.grad_data, group=self.process_group, async_op=True
+            )
+            b._sync_shard = shard
+        else:
+            b.sync_work = dist.all_reduce(b.grad_data, group=self.process_group, async_op=True)
+            b._sync_shard = b.grad_data[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+
+    def _finish_bucket_grad_sync(self, b):
+        if b.synced:
+            return
+        if b.sync_work is None:
+            self._start_bucket_grad_sync(b)
+        if b.sync_work is not None:
+            b.sync_work.wait()
+            if self._use_stream:
+                torch.cuda.current_stream().wait_stream(self._comm_stream)
+            b.grad_shard.copy_(b._sync_shard.float())
+            b.sync_work = None
+        b.synced = True
+
+    def grad_sync(self):
+        """Finish all outstanding gradient reductions."""
+        for b in self.buckets:
+            self._finish_bucket_grad_sync(b)
+
+    # ---------- norms / clipping ----------
+    def grad_norm(self):
+        """Global L2 norm over the sharded (already reduced) gradients."""
+        self.grad_sync()
+        local_sq = sum(float(b.grad_shard.pow(2).sum()) for b in self.buckets)
+        t = torch.tensor([local_sq], device=self.device)
+        if self.world_size > 1:
+            dist.all_reduce(t, group=self.process_group)
+        return t.sqrt().squeeze()
+
+    def clip_grad_norm(self, max_norm):
+        norm = self.grad_norm()
+        clip = max_norm / (float(norm) + 1e-6)
+        if clip < 1.0:
+            for b in self.buckets:
+                b.grad_shard.mul_(clip)
+        return norm
+
+    # ---------- step ----------
+    def zero_grad(self, set_to_none=True):
+        super().zero_grad(set_to_none=True)
+
+    @torch.no_grad()
+    def step(self, closure=None, grad_scaler=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        self.grad_sync()
+        self._step += 1
+
+        if grad_scaler is not None:
+            inv_scale = float(grad_scaler._get_scale_async().double().reciprocal())
+            for b in self.buckets:
+                b.grad_shard.mul_(inv_scale)
+
+        for b in self.buckets:
+            group = b.group
+            beta1, beta2 = group["betas"]
+            bias_correction = 1 if group["bias_correction"] else 0
+            if self.device.type == "cuda":
+                amp_C = get_ext("amp_C")
+                multi_tensor_applier(
+                    amp_C.multi_tensor_adam, self._noop,
+                    [[b.grad_shard], [b.master_shard], [b.exp_avg], [b.exp_avg_sq]],
+                    group["lr"], beta1, beta2, group["eps"], self._step,
+                    self.adam_w_mode, bias_correction, group["weight_decay"],
+                )
+            else:
+                self._adam_ref(group, bias_correction, beta1, beta2, b)
+
+        # param sync: shard -> param bucket, then all_gather
+        for b in self.buckets:
+            lo = self.rank * b.shard_size
+            shard = b.param_data[lo:lo + b.shard_size]
+            shard.copy_(b.master_shard.to(b.param_data.dtype))
+            if self.world_size > 1:
+                if _backend_supports_rs(self.process_group):
+                    dist.all_gather_into_tensor(b.param_data, shard, group=self.process_group)
+                else:
+                    chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                    dist.all_gather(chunks, shard, group=self.process_group)
+                    for r, c in enumerate(chunks):
+                        b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
+            # reset for the next iteration
+            b.grad_data.zero_()
+            b.ready_params.clear()
+            b.synced = False
+            b.sync_work = None
+        return loss
+
+    def _adam_ref(self, group, bias_correction, beta1, beta2, b):
+        step = self._step
+        bc1 = 1 - beta1 ** step if bias_correction else 1.0
+        bc2 = 1 - beta2 ** step if bias_correction else 1.0
+        lr, wd, eps = group["lr"], group["weight_decay"], group["eps"]
+        g = b.grad_shard
+        if self.adam_w_mode == 0 and wd != 0:
+            g = g + wd * b.master_shard
+        b.exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+        b.exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        update = (b.exp_avg / bc1) / ((b.exp_avg_sq / bc2).sqrt() + eps)
+        if self.adam_w_mode == 1 and wd != 0:
+            update = update + wd * b.master_shard
+        b.master_shard.add_(update, alpha=-lr)
+
+    # ---------- checkpoint (shard-local, v2-style) ----------
+    def state_dict(self, gather_on_root=False):
+        if gather_on_root:
+            raise NotImplementedError("gather_on_root checkpointing lands in a later round")
+        return {
+            "step": self._step,
+            "world_size": self.world_size,
+            "param_groups": [
+                {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
+            ],
+            "buckets": [
+                {
+                    "master_shard": b.master_shard,
+                    "exp_avg": b.exp_avg,
+                    "exp_avg_sq": b.exp_avg_sq,
+                }
+                for b in self.buckets
+            ],
+        }
+
+    def load_state_dict(self, sd):
+        if "buckets" not in sd:
+            raise ValueError("expected a DistributedFusedAdam shard-local state dict")
+        assert sd["world_size"] == self.world_size, (
+            "world size changed; resharding load is not supported yet"
+        )
+        self._step = sd["step"]
+        for g, gsd in zip(self.param_groups, sd["param_groups"]):
+            g.update(gsd)
+        for b, bsd in zip(self.buckets, sd["buckets"]):
+            b.master_shard.copy_(bsd["master_shard"])
+            b.exp_avg.copy_(bsd["exp_avg"])
+            b.exp_avg_sq.copy_(bsd["exp_avg_sq"])
+            lo = self.rank * b.shard_size
+            b.param_data[lo:lo + b.shard_size].copy_(b.master_shard.to(b.param_data.dtype))
+        # rebroadcast params
+        if self.world_size > 1:
+            for b in self.buckets:
+                shard = b.param_data[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+                if _backend_supports_rs(self.process_group):
+                    dist.all_gather_into_tensor(b.param_data, shard.contiguous(),
+                                                group=self.process_group)
+                else:
+                    chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                    dist.all_gather(chunks, shard.contiguous(), group=self.process_group)
+                    for r, c in enumerate(chunks):
+                        b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)

@@ -1,0 +1,147 @@
+"""DistributedFusedLAMB — ZeRO-style sharded LAMB over RCCL/xGMI.
+
+API parity with the reference ``apex.contrib.optimizers.DistributedFusedLAMB``
+(apex/contrib/optimizers/distributed_fused_lamb.py:26-1333): bucketed
+reduce-scatter of gradients, device-side global grad norm with optional
+clipping (``clip_after_ar`` semantics collapse to post-reduction clipping
+here), Adam-style stage-1 update, per-tensor trust-ratio apply, and chunked
+all_gather of updated params. The reference's blocks/chunks/shards layout is
+replaced by the same flat-bucket sharding used by DistributedFusedAdam; the
+per-tensor norms are computed as shard-local partial sums all-reduced as one
+[n_tensors, 2] vector (one collective per bucket group instead of per
+tensor).
+"""
+
+import torch
+import torch.distributed as dist
+
+from .distributed_fused_adam import DistributedFusedAdam, _backend_supports_rs
+
+
+class DistributedFusedLAMB(DistributedFusedAdam):
+    def __init__(
+        self,
+        params,
+        lr=1e-3,
+        bias_correction=True,
+        betas=(0.9, 0.999),
+        eps=1e-6,
+        weight_decay=0.01,
+        max_grad_norm=1.0,
+        adam_w_mode=True,
+        grad_averaging=True,
+        use_nvlamb=False,
+        clip_after_ar=True,
+        **kwargs,
+    ):
+        super().__init__(
+            params, lr=lr, bias_correction=bias_correction, betas=betas, eps=eps,
+            adam_w_mode=adam_w_mode, weight_decay=weight_decay, **kwargs,
+        )
+        self.max_grad_norm = max_grad_norm
+        self.grad_averaging = grad_averaging
+        self.use_nvlamb = use_nvlamb
+        self.clip_after_ar = clip_after_ar
+        self._global_scale = 1.0
+
+    def set_global_scale(self, global_scale):
+        """External loss-scaler hookup (reference :1222)."""
+        self._global_scale = float(global_scale)
+
+    def complete_reductions(self):
+        """Reference :1235 — finish all grad reductions."""
+        self.grad_sync()
+
+    @torch.no_grad()
+    def step(self, closure=None, grad_scaler=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        self.grad_sync()
+        self._step += 1
+
+        inv_scale = 1.0 / self._global_scale
+        if grad_scaler is not None:
+            inv_scale *= float(grad_scaler._get_scale_async().double().reciprocal())
+        if inv_scale != 1.0:
+            for b in self.buckets:
+                b.grad_shard.mul_(inv_scale)
+
+        gnorm = float(self.grad_norm_from_shards())
+        clip = gnorm / self.max_grad_norm if (self.max_grad_norm > 0 and gnorm > self.max_grad_norm) else 1.0
+
+        for b in self.buckets:
+            group = b.group
+            beta1, beta2 = group["betas"]
+            step = self._step
+            bc1 = 1 - beta1 ** step if group["bias_correction"] else 1.0
+            bc2 = 1 - beta2 ** step if group["bias_correction"] else 1.0
+            beta3 = 1 - beta1 if self.grad_averaging else 1.0
+            wd, eps, lr = group["weight_decay"], group["eps"], group["lr"]
+
+            # stage 1: Adam-style update written over grad_shard
+            g = b.grad_shard.div_(clip)
+            if self.adam_w_mode == 0 and wd != 0:
+                g = g.add_(b.master_shard, alpha=wd)
+            b.exp_avg.mul_(beta1).add_(g, alpha=beta3)
+            b.exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+            update = (b.exp_avg / bc1) / ((b.exp_avg_sq / bc2).sqrt() + eps)
+            if self.adam_w_mode == 1 and wd != 0:
+                update = update + wd * b.master_shard
+            b.grad_shard.copy_(update)
+
+            # per-tensor partial norms over this shard
+            lo = self.rank * b.shard_size
+            hi = lo + b.shard_size
+            partials = torch.zeros(len(b.params), 2, device=self.device)
+            for i, (p, offset) in enumerate(b.params):
+                s0, s1 = max(offset, lo), min(offset + p.numel(), hi)
+                if s0 >= s1:
+                    continue
+                seg = slice(s0 - lo, s1 - lo)
+                partials[i, 0] = b.master_shard[seg].pow(2).sum()
+                partials[i, 1] = update[seg].pow(2).sum()
+            if self.world_size > 1:
+                dist.all_reduce(partials, group=self.process_group)
+            norms = partials.sqrt()
+
+            # stage 2: trust-ratio apply on this shard
+            for i, (p, offset) in enumerate(b.params):
+                s0, s1 = max(offset, lo), min(offset + p.numel(), hi)
+                if s0 >= s1:
+                    continue
+                seg = slice(s0 - lo, s1 - lo)
+                pn, un = float(norms[i, 0]), float(norms[i, 1])
+                if (self.use_nvlamb or wd != 0) and pn != 0 and un != 0:
+                    ratio = lr * pn / un
+                else:
+                    ratio = lr
+                b.master_shard[seg].add_(update[seg], alpha=-ratio)
+
+        # param sync
+        for b in self.buckets:
+            lo = self.rank * b.shard_size
+            shard = b.param_data[lo:lo + b.shard_size]
+            shard.copy_(b.master_shard.to(b.param_data.dtype))
+            if self.world_size > 1:
+                if _backend_supports_rs(self.process_group):
+                    dist.all_gather_into_tensor(b.param_data, shard, group=self.process_group)
+                else:
+                    chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                    dist.all_gather(chunks, shard, group=self.process_group)
+                    for r, c in enumerate(chunks):
+                        b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
+            b.grad_data.zero_()
+            b.ready_params.clear()
+            b.synced = False
+            b.sync_work = None
+        return loss
+
+    def grad_norm_from_shards(self):
+        local_sq = sum(float(b.grad_shard.pow(2).sum()) for b in self.buckets)
+        t = torch.tensor([local_sq], device=self.device)
+        if self.world_size > 1:
+            dist.all_reduce(t, group=self.process_group)
+        return t.sqrt().squeeze()

@@ -74,6 +74,7 @@ setup(
         "apex_amd.contrib.xentropy",
         "apex_amd.contrib.focal_loss",
         "apex_amd.contrib.index_mul_2d",
+        "apex_amd.contrib.optimizers",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",

@@ -1,0 +1,160 @@
+"""DistributedFusedAdam / DistributedFusedLAMB on gloo world_size=2 —
+step-by-step parity with non-sharded references (pattern of the reference
+apex/contrib/test/optimizers/test_dist_adam.py)."""
+
+import torch
+import torch.distributed as dist
+import pytest
+
+from utils import run_distributed
+
+
+def _make_model(seed=123):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.Tanh(), torch.nn.Linear(64, 8))
+
+
+def _set_rank_grads(model, rank, world, it):
+    """Deterministic per-rank grads; returns the world-mean grads."""
+    mean_grads = []
+    for i, p in enumerate(model.parameters()):
+        torch.manual_seed(1000 * it + i)
+        base = torch.randn_like(p)
+        # rank-dependent component that averages to zero across ranks
+        delta = torch.randn_like(p)
+        sign = 1.0 if rank == 0 else -1.0
+        p.grad = base + sign * delta
+        mean_grads.append(base.clone())
+    return mean_grads
+
+
+def _dist_adam_worker(rank, world_size, adam_w, overlap):
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    ref_params = [p.detach().clone().requires_grad_(True) for p in model.parameters()]
+    opt = DistributedFusedAdam(
+        model.parameters(), lr=1e-3, weight_decay=0.01, adam_w_mode=adam_w,
+        bucket_cap_mb=1, overlap_grad_sync=overlap,
+    )
+    ref_cls = torch.optim.AdamW if adam_w else torch.optim.Adam
+    ref_opt = ref_cls(ref_params, lr=1e-3, weight_decay=0.01)
+
+    for it in range(8):
+        mean_grads = _set_rank_grads(model, rank, world_size, it)
+        # hooks don't fire without backward; feed grads through the copy path
+        for p in model.parameters():
+            opt._grad_copy(p)
+        opt.step()
+        for p, g in zip(ref_params, mean_grads):
+            p.grad = g
+        ref_opt.step()
+        for p, rp in zip(model.parameters(), ref_params):
+            torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.parametrize("adam_w", [True, False])
+def test_dist_adam_matches_adamw(adam_w):
+    run_distributed(_dist_adam_worker, world_size=2, args=(adam_w, False))
+
+
+def _overlap_worker(rank, world_size):
+        from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+        model = _make_model()
+        ref_model = _make_model()
+        opt = DistributedFusedAdam(model.parameters(), lr=1e-2, weight_decay=0.0,
+                                   bucket_cap_mb=1, overlap_grad_sync=True)
+        ref_opt = torch.optim.AdamW(ref_model.parameters(), lr=1e-2, weight_decay=0.0)
+        for it in range(4):
+            torch.manual_seed(10 + rank + it * world_size)
+            x = torch.randn(4, 32)
+            # reference: average grads over both ranks' inputs
+            xs = [torch.empty_like(x) for _ in range(world_size)]
+            dist.all_gather(xs, x)
+            model(x).pow(2).mean().backward()
+            opt.step()
+            ref_opt.zero_grad()
+            loss = sum(ref_model(xi).pow(2).mean() for xi in xs) / world_size
+            loss.backward()
+            ref_opt.step()
+            for p, rp in zip(model.parameters(), ref_model.parameters()):
+                torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_dist_adam_overlap_via_backward():
+    run_distributed(_overlap_worker, world_size=2)
+
+
+def _sd_worker(rank, world_size):
+        from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+        model = _make_model()
+        opt = DistributedFusedAdam(model.parameters(), lr=1e-3, bucket_cap_mb=1)
+        for it in range(3):
+            _set_rank_grads(model, rank, world_size, it)
+            for p in model.parameters():
+                opt._grad_copy(p)
+            opt.step()
+        sd = opt.state_dict()
+
+        model2 = _make_model()
+        opt2 = DistributedFusedAdam(model2.parameters(), lr=1e-3, bucket_cap_mb=1)
+        opt2.load_state_dict(sd)
+        for p, q in zip(model.parameters(), model2.parameters()):
+            torch.testing.assert_close(p.detach(), q.detach())
+        # both continue identically
+        for it in range(3, 5):
+            for m, o in ((model, opt), (model2, opt2)):
+                _set_rank_grads(m, rank, world_size, it)
+                for p in m.parameters():
+                    o._grad_copy(p)
+                o.step()
+        for p, q in zip(model.parameters(), model2.parameters()):
+            torch.testing.assert_close(p.detach(), q.detach())
+
+
+def test_dist_adam_state_dict_roundtrip():
+    run_distributed(_sd_worker, world_size=2)
+
+
+def _lamb_worker(rank, world_size):
+        from apex_amd.contrib.optimizers import DistributedFusedLAMB
+        from apex_amd.optimizers import FusedLAMB
+
+        model = _make_model()
+        ref_params = [p.detach().clone().requires_grad_(True) for p in model.parameters()]
+        opt = DistributedFusedLAMB(model.parameters(), lr=1e-3, weight_decay=0.01,
+                                   max_grad_norm=1.0, bucket_cap_mb=1)
+        ref_opt = FusedLAMB(ref_params, lr=1e-3, weight_decay=0.01, max_grad_norm=1.0)
+        for it in range(5):
+            mean_grads = _set_rank_grads(model, rank, world_size, it)
+            for p in model.parameters():
+                opt._grad_copy(p)
+            opt.step()
+            for p, g in zip(ref_params, mean_grads):
+                p.grad = g
+            ref_opt.step()
+            for p, rp in zip(model.parameters(), ref_params):
+                torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_dist_lamb_matches_fused_lamb():
+    run_distributed(_lamb_worker, world_size=2)
+
+
+def test_fp16_optimizer_cpu():
+    from apex_amd.contrib.optimizers import FP16_Optimizer
+
+    torch.manual_seed(0)
+    model = torch.nn.Linear(8, 4).to(torch.bfloat16)
+    inner = torch.optim.SGD([p for p in model.parameters()], lr=0.1)
+    opt = FP16_Optimizer(inner, static_loss_scale=128.0)
+    x = torch.randn(4, 8, dtype=torch.bfloat16)
+    loss = model(x).float().sum()
+    opt.zero_grad()
+    opt.backward(loss)
+    before = [p.detach().clone() for p in model.parameters()]
+    opt.step()
+    for b, p in zip(before, model.parameters()):
+        assert not torch.equal(b, p.detach())
