@@ -60,6 +60,7 @@ ext_modules = [
     hip_ext("apex_amd._xentropy", ["csrc/xentropy.hip"]),
     hip_ext("apex_amd._focal_loss", ["csrc/focal_loss.hip"]),
     hip_ext("apex_amd._index_mul_2d", ["csrc/index_mul_2d.hip"]),
+    hip_ext("apex_amd._group_norm", ["csrc/group_norm.hip"]),
 ]
 
 setup(
@@ -75,6 +76,10 @@ setup(
         "apex_amd.contrib.focal_loss",
         "apex_amd.contrib.index_mul_2d",
         "apex_amd.contrib.optimizers",
+        "apex_amd.contrib.group_norm",
+        "apex_amd.contrib.groupbn",
+        "apex_amd.contrib.layer_norm",
+        "apex_amd.contrib.sparsity",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",

@@ -1,0 +1,134 @@
+"""CPU tests for contrib: sparsity ASP, xentropy fallback, group_norm
+fallback, focal loss reference, fast_layer_norm fallback."""
+
+import torch
+import pytest
+
+
+def test_create_mask_2to4():
+    from apex_amd.contrib.sparsity import create_mask
+
+    torch.manual_seed(0)
+    w = torch.randn(16, 32)
+    mask = create_mask(w, "m4n2_1d")
+    assert mask.shape == w.shape
+    groups = mask.reshape(-1, 4).sum(dim=1)
+    assert (groups == 2).all()
+    # kept entries are the 2 largest per group
+    wg = w.abs().reshape(-1, 4)
+    kept = wg[mask.reshape(-1, 4)].reshape(-1, 2).min(dim=1).values
+    dropped = wg[~mask.reshape(-1, 4)].reshape(-1, 2).max(dim=1).values
+    assert (kept >= dropped - 1e-6).all()
+
+
+def test_create_mask_2d():
+    from apex_amd.contrib.sparsity import create_mask
+
+    torch.manual_seed(1)
+    w = torch.randn(8, 8)
+    mask = create_mask(w, "m4n2_2d_best")
+    m = mask.reshape(2, 4, 2, 4)
+    # every 4-row and 4-col of each 4x4 block has exactly 2 kept
+    assert (mask.reshape(8, 2, 4).sum(-1) == 2).all()
+
+
+def test_asp_prune_and_step_reapplies_mask():
+    from apex_amd.contrib.sparsity import ASP
+
+    ASP._reset()
+    torch.manual_seed(2)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 8), torch.nn.ReLU(), torch.nn.Linear(8, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    ASP.prune_trained_model(model, opt)
+    w = model[0].weight
+    assert (w.detach().reshape(-1, 4) != 0).sum(dim=1).max() <= 2
+    # train a step; mask must be re-applied
+    x = torch.randn(4, 16)
+    loss = model(x).sum()
+    loss.backward()
+    opt.step()
+    assert ((w.detach().reshape(-1, 4) != 0).sum(dim=1) <= 2).all()
+    ASP._reset()
+
+
+def test_group_norm_cpu_fallback():
+    from apex_amd.contrib.group_norm import GroupNorm
+
+    torch.manual_seed(3)
+    gn = GroupNorm(4, 32)
+    ref = torch.nn.GroupNorm(4, 32)
+    with torch.no_grad():
+        ref.weight.copy_(gn.weight)
+        ref.bias.copy_(gn.bias)
+    x = torch.randn(2, 32, 8, 8)
+    torch.testing.assert_close(gn(x), ref(x))
+
+
+def test_group_norm_silu_cpu():
+    from apex_amd.contrib.group_norm import GroupNorm
+
+    gn = GroupNorm(2, 8, act="silu")
+    x = torch.randn(2, 8, 4, 4)
+    expected = torch.nn.functional.silu(
+        torch.nn.functional.group_norm(x, 2, gn.weight, gn.bias, gn.eps)
+    )
+    torch.testing.assert_close(gn(x), expected)
+
+
+def test_fast_layer_norm_cpu_fallback():
+    from apex_amd.contrib.layer_norm import FastLayerNorm
+
+    ln = FastLayerNorm(64)
+    x = torch.randn(4, 64)
+    ref = torch.nn.functional.layer_norm(x, (64,), ln.weight, ln.bias, ln.epsilon)
+    torch.testing.assert_close(ln(x), ref)
+
+
+def test_xentropy_cpu_fallback():
+    from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
+
+    torch.manual_seed(4)
+    logits = torch.randn(16, 32, requires_grad=True)
+    labels = torch.randint(1, 32, (16,))
+    losses = SoftmaxCrossEntropyLoss.apply(logits, labels, 0.1, 0, False)
+    ref = torch.nn.functional.cross_entropy(logits.detach(), labels, reduction="none",
+                                            label_smoothing=0.1)
+    torch.testing.assert_close(losses, ref, rtol=1e-5, atol=1e-6)
+    losses.sum().backward()
+    x2 = logits.detach().clone().requires_grad_(True)
+    torch.nn.functional.cross_entropy(x2, labels, reduction="none", label_smoothing=0.1).sum().backward()
+    torch.testing.assert_close(logits.grad, x2.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_focal_loss_reference_math():
+    from apex_amd.contrib.focal_loss import focal_loss
+
+    torch.manual_seed(5)
+    x = torch.randn(32, 16)
+    y = torch.randint(-2, 16, (32,))
+    nps = torch.tensor([max(float((y >= 0).sum()), 1.0)])
+    loss = focal_loss(x, y, nps, 16, 0.25, 2.0, 0.0)
+    assert torch.isfinite(loss)
+
+
+def test_groupbn_nhwc_cpu():
+    from apex_amd.contrib.groupbn import BatchNorm2d_NHWC
+
+    torch.manual_seed(6)
+    bn = BatchNorm2d_NHWC(8)
+    bn.train()
+    x = torch.randn(4, 5, 5, 8)
+    y = bn(x)
+    ref_bn = torch.nn.BatchNorm2d(8)
+    ref_bn.train()
+    with torch.no_grad():
+        ref_bn.weight.copy_(bn.weight)
+        ref_bn.bias.copy_(bn.bias)
+    ref = ref_bn(x.permute(0, 3, 1, 2)).permute(0, 2, 3, 1)
+    torch.testing.assert_close(y, ref, rtol=1e-4, atol=1e-5)
+    # add+relu fusion path
+    z = torch.randn_like(y)
+    y2 = BatchNorm2d_NHWC(8, fuse_relu=True)
+    y2.train()
+    out = y2(x, z)
+    assert (out >= 0).all()
