@@ -13,6 +13,50 @@ at::Tensor flat2d(const at::Tensor& t) {
   return t.contiguous().reshape({-1, t.size(-1)});
 }
 
+// erf-GELU elementwise fwd/bwd — fallback for dtypes where hipBLASLt has no
+// GELU_AUX_BIAS / DGELU_BGRAD algorithm (bf16 as of hipBLASLt 1.2).
+template <typename T>
+__global__ void __launch_bounds__(256) gelu_fwd_kernel(const T* __restrict__ z,
+                                                       T* __restrict__ y, long n) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float v = to_float(z[i]);
+    y[i] = from_float<T>(0.5f * v * (1.f + erff(v * 0.70710678118654752440f)));
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256) dgelu_kernel(T* __restrict__ d, const T* __restrict__ z,
+                                                    long n) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    const float v = to_float(z[i]);
+    const float cdf = 0.5f * (1.f + erff(v * 0.70710678118654752440f));
+    const float pdf = 0.3989422804014327f * __expf(-0.5f * v * v);
+    d[i] = from_float<T>(to_float(d[i]) * (cdf + v * pdf));
+  }
+}
+
+void gelu_fwd_inplace(const at::Tensor& z, at::Tensor& y) {
+  const long n = z.numel();
+  const int grid = (int)std::min<long>((n + 255) / 256, 8192);
+  APEX_DISPATCH_FLOAT_HALF_BF(z.scalar_type(), "gelu_fwd", ([&] {
+    hipLaunchKernelGGL((gelu_fwd_kernel<scalar_t>), dim3(grid), dim3(256), 0, current_stream(),
+                       (const scalar_t*)z.data_ptr(), (scalar_t*)y.data_ptr(), n);
+  }()));
+  HIP_CHECK(hipGetLastError());
+}
+
+void dgelu_inplace(at::Tensor& d, const at::Tensor& z) {
+  const long n = z.numel();
+  const int grid = (int)std::min<long>((n + 255) / 256, 8192);
+  APEX_DISPATCH_FLOAT_HALF_BF(z.scalar_type(), "dgelu", ([&] {
+    hipLaunchKernelGGL((dgelu_kernel<scalar_t>), dim3(grid), dim3(256), 0, current_stream(),
+                       (scalar_t*)d.data_ptr(), (const scalar_t*)z.data_ptr(), n);
+  }()));
+  HIP_CHECK(hipGetLastError());
+}
+
 }  // namespace
 
 at::Tensor linear_bias_forward(at::Tensor input, at::Tensor weight, at::Tensor bias) {
@@ -75,7 +119,12 @@ std::vector<at::Tensor> linear_gelu_linear_forward(at::Tensor input, at::Tensor 
   auto output1 = at::empty({m, n1}, x.options());   // GELU(X@W1^T + b1)
   auto gelu_in = at::empty({m, n1}, x.options());   // pre-GELU (post-bias) aux
   auto output2 = at::empty({m, n2}, x.options());
-  lt_linear(x, w1, output1, &b1, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, &gelu_in);
+  if (!lt_linear(x, w1, output1, &b1, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, &gelu_in,
+                 /*allow_fail=*/true)) {
+    // split epilogue: BIAS GEMM into gelu_in, then elementwise erf-GELU
+    lt_linear(x, w1, gelu_in, &b1, HIPBLASLT_EPILOGUE_BIAS, nullptr);
+    gelu_fwd_inplace(gelu_in, output1);
+  }
   lt_linear(output1, w2, output2, &b2, HIPBLASLT_EPILOGUE_BIAS, nullptr);
   auto sizes = input.sizes().vec();
   sizes.back() = n2;
@@ -100,7 +149,12 @@ std::vector<at::Tensor> linear_gelu_linear_backward(at::Tensor input, at::Tensor
   // d_gelu = dGELU(dY2 @ W2, gelu_in), db1 fused
   auto d_gelu = at::empty({m, n1}, x.options());
   auto db1 = at::empty({n1}, w1.options());
-  lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DGELU_BGRAD, &gi, &db1);
+  if (!lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DGELU_BGRAD, &gi, &db1,
+                       /*allow_fail=*/true)) {
+    lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr);
+    dgelu_inplace(d_gelu, gi);
+    db1.copy_(d_gelu.sum(0).to(db1.scalar_type()));
+  }
 
   auto dw1 = at::empty_like(w1);
   lt_linear_wgrad(x, d_gelu, dw1, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, 0.f);

@@ -52,11 +52,16 @@ struct LtAlgoKey {
 
 // col-major primitive: C[m,n] = alpha * opA(A) * opB(B) + beta * C
 // aux/bias pointers per hipblasLt epilogue semantics.
-inline void lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m, long n, long k,
+// Returns false (without launching) iff no algorithm exists for the
+// requested epilogue/dtype combo and `allow_fail` is set — callers fall back
+// to a split epilogue (e.g. bf16 GELU_AUX_BIAS is not implemented by
+// hipBLASLt 1.2).
+inline bool lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m, long n, long k,
                          const void* A, long lda, hipDataType typeA, const void* B, long ldb,
                          hipDataType typeB, void* C, long ldc, hipDataType typeC,
                          hipblasLtEpilogue_t epi, const void* bias, hipDataType bias_type,
-                         void* aux, long aux_ld, hipDataType aux_type, float alpha, float beta) {
+                         void* aux, long aux_ld, hipDataType aux_type, float alpha, float beta,
+                         bool allow_fail = false) {
   auto handle = lt_handle();
   auto stream = current_stream();
 
@@ -104,47 +109,56 @@ inline void lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m,
 
   hipblasLtMatmulHeuristicResult_t heur;
   int nheur = 0;
-  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(handle, desc, layA, layB, layC, layC, pref, 1, &heur,
-                                           &nheur));
-  TORCH_CHECK(nheur > 0, "hipBLASLt: no algorithm for this GEMM (m=", m, " n=", n, " k=", k,
-              " epi=", (int)epi, ")");
-
-  LT_CHECK(hipblasLtMatmul(handle, desc, &alpha, A, layA, B, layB, &beta, C, layC, C, layC,
-                           &heur.algo, ws.data_ptr(), ws_size, stream));
+  hipblasStatus_t hst =
+      hipblasLtMatmulAlgoGetHeuristic(handle, desc, layA, layB, layC, layC, pref, 1, &heur, &nheur);
+  bool ok = (hst == HIPBLAS_STATUS_SUCCESS) && nheur > 0;
+  if (!ok && !allow_fail) {
+    TORCH_CHECK(false, "hipBLASLt: no algorithm for this GEMM (m=", m, " n=", n, " k=", k,
+                " epi=", (int)epi, ")");
+  }
+  if (ok) {
+    LT_CHECK(hipblasLtMatmul(handle, desc, &alpha, A, layA, B, layB, &beta, C, layC, C, layC,
+                             &heur.algo, ws.data_ptr(), ws_size, stream));
+  }
 
   hipblasLtMatmulPreferenceDestroy(pref);
   hipblasLtMatrixLayoutDestroy(layA);
   hipblasLtMatrixLayoutDestroy(layB);
   hipblasLtMatrixLayoutDestroy(layC);
   hipblasLtMatmulDescDestroy(desc);
+  return ok;
 }
 
 // ---- row-major wrappers ----
 
 // out[m,n] = X[m,k] @ W[n,k]^T (+bias[n], + optional activation epilogue)
 // aux (if used) is row-major [m, n] with ld n.
-inline void lt_linear(const at::Tensor& X, const at::Tensor& W, at::Tensor& out,
-                      const at::Tensor* bias, hipblasLtEpilogue_t epi, at::Tensor* aux) {
+inline bool lt_linear(const at::Tensor& X, const at::Tensor& W, at::Tensor& out,
+                      const at::Tensor* bias, hipblasLtEpilogue_t epi, at::Tensor* aux,
+                      bool allow_fail = false) {
   const long m = X.size(0), k = X.size(1), n = W.size(0);
-  lt_matmul_cm(HIPBLAS_OP_T, HIPBLAS_OP_N, n, m, k, W.data_ptr(), k, lt_dtype(W.scalar_type()),
+  return lt_matmul_cm(HIPBLAS_OP_T, HIPBLAS_OP_N, n, m, k, W.data_ptr(), k,
+               lt_dtype(W.scalar_type()),
                X.data_ptr(), k, lt_dtype(X.scalar_type()), out.data_ptr(), n,
                lt_dtype(out.scalar_type()), epi, bias ? bias->data_ptr() : nullptr,
                bias ? lt_dtype(bias->scalar_type()) : HIP_R_32F,
                aux ? aux->data_ptr() : nullptr, aux ? n : 0,
-               aux ? lt_dtype(aux->scalar_type()) : HIP_R_32F, 1.f, 0.f);
+               aux ? lt_dtype(aux->scalar_type()) : HIP_R_32F, 1.f, 0.f, allow_fail);
 }
 
 // dX[m,k] = dY[m,n] @ W[n,k] (+ optional DGELU/DGELU_BGRAD with aux = gelu_in
 // row-major [m, k], dbias via bias pointer)
-inline void lt_linear_dgrad(const at::Tensor& dY, const at::Tensor& W, at::Tensor& dX,
-                            hipblasLtEpilogue_t epi, at::Tensor* aux, at::Tensor* dbias) {
+inline bool lt_linear_dgrad(const at::Tensor& dY, const at::Tensor& W, at::Tensor& dX,
+                            hipblasLtEpilogue_t epi, at::Tensor* aux, at::Tensor* dbias,
+                            bool allow_fail = false) {
   const long m = dY.size(0), n = dY.size(1), k = W.size(1);
-  lt_matmul_cm(HIPBLAS_OP_N, HIPBLAS_OP_N, k, m, n, W.data_ptr(), k, lt_dtype(W.scalar_type()),
+  return lt_matmul_cm(HIPBLAS_OP_N, HIPBLAS_OP_N, k, m, n, W.data_ptr(), k,
+               lt_dtype(W.scalar_type()),
                dY.data_ptr(), n, lt_dtype(dY.scalar_type()), dX.data_ptr(), k,
                lt_dtype(dX.scalar_type()), epi, dbias ? dbias->data_ptr() : nullptr,
                dbias ? lt_dtype(dbias->scalar_type()) : HIP_R_32F,
                aux ? aux->data_ptr() : nullptr, aux ? k : 0,
-               aux ? lt_dtype(aux->scalar_type()) : HIP_R_32F, 1.f, 0.f);
+               aux ? lt_dtype(aux->scalar_type()) : HIP_R_32F, 1.f, 0.f, allow_fail);
 }
 
 // dW[n,k] (+= if beta=1) dY[m,n]^T @ X[m,k]; optional BGRADB dbias[n].
