@@ -3,6 +3,12 @@ And how do the fused MLP grads compare against same-dtype torch references?"""
 
 import torch
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import apex_amd._fused_dense as fd
 
 

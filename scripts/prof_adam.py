@@ -3,6 +3,12 @@ profiling payload, small and fast)."""
 
 import torch
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 from apex_amd.optimizers import FusedAdam
 from apex_amd.normalization import FusedLayerNorm
 
