@@ -1,0 +1,328 @@
+// apex_amd._transducer — RNN-T joint and loss for gfx950.
+// Reference surface: apex/contrib/transducer (TransducerJoint: out[b,t,u,:] =
+// f[b,t,:] + g[b,u,:] with optional fused ReLU and length masking;
+// TransducerLoss: alpha/beta dynamic program over (T, U) with fused grad).
+//
+// MI355X design: joint fwd is a broadcast-add elementwise pass; joint bwd is
+// two sum-reductions (over u for grad_f, over t for grad_g) with one
+// workgroup per output row, lanes across H (coalesced). The loss DP walks
+// anti-diagonals with one workgroup per batch element (wave-parallel along
+// the diagonal, barrier per step); grads are emitted in one elementwise pass
+// from alpha+beta.
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+constexpr int TJ_BLOCK = 256;
+
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_kernel(
+    const T* __restrict__ f, const T* __restrict__ g, T* __restrict__ out,
+    const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
+    long H) {
+  const long total = B * Tm * U * H;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long h = i % H;
+    const long u = (i / H) % U;
+    const long t = (i / (H * U)) % Tm;
+    const long b = i / (H * U * Tm);
+    float v = 0.f;
+    if (t < f_len[b] && u < g_len[b]) {
+      v = to_float(f[(b * Tm + t) * H + h]) + to_float(g[(b * U + u) * H + h]);
+      if (RELU) v = fmaxf(v, 0.f);
+    }
+    out[i] = from_float<T>(v);
+  }
+}
+
+// grad_f[b,t,h] = sum_u dout[b,t,u,h] (masked); one block per (b,t) row.
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_f_kernel(
+    const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ df,
+    const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
+    long H) {
+  const long bt = blockIdx.x;
+  const long b = bt / Tm, t = bt % Tm;
+  const int ulen = (t < f_len[b]) ? g_len[b] : 0;
+  for (long h = threadIdx.x; h < H; h += blockDim.x) {
+    float acc = 0.f;
+    for (long u = 0; u < ulen; ++u) {
+      const long idx = ((b * Tm + t) * U + u) * H + h;
+      float d = to_float(dout[idx]);
+      if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
+      acc += d;
+    }
+    df[(b * Tm + t) * H + h] = from_float<T>(acc);
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_g_kernel(
+    const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ dg,
+    const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
+    long H) {
+  const long bu = blockIdx.x;
+  const long b = bu / U, u = bu % U;
+  const int tlen = (u < g_len[b]) ? f_len[b] : 0;
+  for (long h = threadIdx.x; h < H; h += blockDim.x) {
+    float acc = 0.f;
+    for (long t = 0; t < tlen; ++t) {
+      const long idx = ((b * Tm + t) * U + u) * H + h;
+      float d = to_float(dout[idx]);
+      if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
+      acc += d;
+    }
+    dg[(b * U + u) * H + h] = from_float<T>(acc);
+  }
+}
+
+__device__ __forceinline__ float log_add(float a, float b) {
+  if (a == -INFINITY) return b;
+  if (b == -INFINITY) return a;
+  const float mx = fmaxf(a, b);
+  return mx + __logf(__expf(a - mx) + __expf(b - mx));
+}
+
+// alpha DP: one block per batch element; anti-diagonal parallel.
+// x: log-probs [B, T, U, V]; label: [B, U-1]; alpha: [B, T, U] fp32.
+template <typename T>
+__global__ void __launch_bounds__(TJ_BLOCK) rnnt_alpha_kernel(
+    const T* __restrict__ x, const int* __restrict__ label, float* __restrict__ alpha,
+    float* __restrict__ losses, const int* __restrict__ f_len, const int* __restrict__ y_len,
+    long B, long Tm, long U, long V, int blank) {
+  const long b = blockIdx.x;
+  const int Tb = f_len[b];
+  const int Ub = y_len[b] + 1;
+  float* al = alpha + b * Tm * U;
+  const T* xb = x + b * Tm * U * V;
+
+  for (int d = 0; d < Tb + Ub - 1; ++d) {
+    for (int t = threadIdx.x; t <= d; t += blockDim.x) {
+      const int u = d - t;
+      if (t >= Tb || u >= Ub) continue;
+      float v;
+      if (t == 0 && u == 0) {
+        v = 0.f;
+      } else {
+        float from_blank = -INFINITY, from_label = -INFINITY;
+        if (t > 0) {
+          from_blank = al[(t - 1) * U + u] + to_float(xb[((long)(t - 1) * U + u) * V + blank]);
+        }
+        if (u > 0) {
+          const int y = label[b * (U - 1) + (u - 1)];
+          from_label = al[(long)t * U + (u - 1)] + to_float(xb[((long)t * U + (u - 1)) * V + y]);
+        }
+        v = log_add(from_blank, from_label);
+      }
+      al[(long)t * U + u] = v;
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const float last_blank = to_float(xb[((long)(Tb - 1) * U + (Ub - 1)) * V + blank]);
+    losses[b] = -(al[(long)(Tb - 1) * U + (Ub - 1)] + last_blank);
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(TJ_BLOCK) rnnt_beta_kernel(
+    const T* __restrict__ x, const int* __restrict__ label, float* __restrict__ beta,
+    const int* __restrict__ f_len, const int* __restrict__ y_len, long B, long Tm, long U, long V,
+    int blank) {
+  const long b = blockIdx.x;
+  const int Tb = f_len[b];
+  const int Ub = y_len[b] + 1;
+  float* be = beta + b * Tm * U;
+  const T* xb = x + b * Tm * U * V;
+
+  for (int d = Tb + Ub - 2; d >= 0; --d) {
+    for (int t = threadIdx.x; t <= d; t += blockDim.x) {
+      const int u = d - t;
+      if (t >= Tb || u >= Ub) continue;
+      float v;
+      if (t == Tb - 1 && u == Ub - 1) {
+        v = to_float(xb[((long)t * U + u) * V + blank]);
+      } else {
+        float via_blank = -INFINITY, via_label = -INFINITY;
+        if (t + 1 < Tb) {
+          via_blank = to_float(xb[((long)t * U + u) * V + blank]) + be[(long)(t + 1) * U + u];
+        }
+        if (u + 1 < Ub) {
+          const int y = label[b * (U - 1) + u];
+          via_label = to_float(xb[((long)t * U + u) * V + y]) + be[(long)t * U + (u + 1)];
+        }
+        v = log_add(via_blank, via_label);
+      }
+      be[(long)t * U + u] = v;
+    }
+    __syncthreads();
+  }
+}
+
+// grads over log-probs: nonzero only at blank and label entries.
+template <typename T>
+__global__ void __launch_bounds__(TJ_BLOCK) rnnt_grad_kernel(
+    const T* __restrict__ x, const int* __restrict__ label, const float* __restrict__ alpha,
+    const float* __restrict__ beta, const float* __restrict__ grad_loss, T* __restrict__ dx,
+    const int* __restrict__ f_len, const int* __restrict__ y_len, long B, long Tm, long U, long V,
+    int blank) {
+  const long total = B * Tm * U;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long u = i % U;
+    const long t = (i / U) % Tm;
+    const long b = i / (U * Tm);
+    const int Tb = f_len[b];
+    const int Ub = y_len[b] + 1;
+    if (t >= Tb || u >= Ub) continue;
+    const float lz = beta[b * Tm * U + 0];  // beta[0,0] == log Z
+    const float gl = grad_loss[b];
+    const T* xb = x + (b * Tm * U + t * U + u) * V;
+    T* dxb = dx + (b * Tm * U + t * U + u) * V;
+    const float a = alpha[b * Tm * U + t * U + u];
+    // blank transition -> (t+1, u) (or termination at (Tb-1, Ub-1))
+    float bnext;
+    if (t == Tb - 1 && u == Ub - 1) {
+      bnext = 0.f;
+    } else if (t + 1 < Tb) {
+      bnext = beta[b * Tm * U + (t + 1) * U + u];
+    } else {
+      bnext = -INFINITY;
+    }
+    if (bnext != -INFINITY) {
+      const float lp = to_float(xb[blank]);
+      dxb[blank] = from_float<T>(-__expf(a + lp + bnext - lz) * gl);
+    }
+    // label transition -> (t, u+1)
+    if (u + 1 < Ub) {
+      const int y = label[b * (U - 1) + u];
+      const float lp = to_float(xb[y]);
+      const float bn = beta[b * Tm * U + t * U + (u + 1)];
+      // accumulate (y may equal blank index in degenerate configs)
+      const float cur = to_float(dxb[y]);
+      dxb[y] = from_float<T>(cur - __expf(a + lp + bn - lz) * gl);
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> transducer_joint_forward(at::Tensor f, at::Tensor g, at::Tensor f_len,
+                                                 at::Tensor g_len, bool relu) {
+  auto fc = f.contiguous();
+  auto gc = g.contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto gl = g_len.to(at::kInt).contiguous();
+  const long B = fc.size(0), Tm = fc.size(1), H = fc.size(2), U = gc.size(1);
+  auto out = at::empty({B, Tm, U, H}, fc.options());
+  const long total = out.numel();
+  const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
+  APEX_DISPATCH_FLOAT_HALF_BF(fc.scalar_type(), "transducer_joint_forward", ([&] {
+    if (relu) {
+      hipLaunchKernelGGL((joint_fwd_kernel<scalar_t, true>), dim3(grid), dim3(TJ_BLOCK), 0,
+                         current_stream(), (const scalar_t*)fc.data_ptr(),
+                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+    } else {
+      hipLaunchKernelGGL((joint_fwd_kernel<scalar_t, false>), dim3(grid), dim3(TJ_BLOCK), 0,
+                         current_stream(), (const scalar_t*)fc.data_ptr(),
+                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+    }
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return {out};
+}
+
+std::vector<at::Tensor> transducer_joint_backward(at::Tensor grad_out, at::Tensor out,
+                                                  at::Tensor f_len, at::Tensor g_len, long B,
+                                                  long Tm, long U, long H, bool relu) {
+  auto dout = grad_out.contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto gl = g_len.to(at::kInt).contiguous();
+  auto df = at::empty({B, Tm, H}, dout.options());
+  auto dg = at::empty({B, U, H}, dout.options());
+  APEX_DISPATCH_FLOAT_HALF_BF(dout.scalar_type(), "transducer_joint_backward", ([&] {
+    if (relu) {
+      hipLaunchKernelGGL((joint_bwd_f_kernel<scalar_t, true>), dim3((uint32_t)(B * Tm)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+      hipLaunchKernelGGL((joint_bwd_g_kernel<scalar_t, true>), dim3((uint32_t)(B * U)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+    } else {
+      hipLaunchKernelGGL((joint_bwd_f_kernel<scalar_t, false>), dim3((uint32_t)(B * Tm)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+      hipLaunchKernelGGL((joint_bwd_g_kernel<scalar_t, false>), dim3((uint32_t)(B * U)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
+                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
+    }
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return {df, dg};
+}
+
+std::vector<at::Tensor> transducer_loss_forward(at::Tensor x, at::Tensor label, at::Tensor f_len,
+                                                at::Tensor y_len, long blank_idx) {
+  auto xc = x.contiguous();
+  auto lab = label.to(at::kInt).contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto yl = y_len.to(at::kInt).contiguous();
+  const long B = xc.size(0), Tm = xc.size(1), U = xc.size(2), V = xc.size(3);
+  auto fopts = xc.options().dtype(at::kFloat);
+  auto alpha = at::full({B, Tm, U}, -INFINITY, fopts);
+  auto losses = at::empty({B}, fopts);
+  APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "transducer_loss_forward", ([&] {
+    hipLaunchKernelGGL((rnnt_alpha_kernel<scalar_t>), dim3((uint32_t)B), dim3(TJ_BLOCK), 0,
+                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
+                       alpha.data_ptr<float>(), losses.data_ptr<float>(), fl.data_ptr<int>(),
+                       yl.data_ptr<int>(), B, Tm, U, V, (int)blank_idx);
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return {losses, alpha};
+}
+
+at::Tensor transducer_loss_backward(at::Tensor x, at::Tensor label, at::Tensor alpha,
+                                    at::Tensor grad_loss, at::Tensor f_len, at::Tensor y_len,
+                                    long blank_idx) {
+  auto xc = x.contiguous();
+  auto lab = label.to(at::kInt).contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto yl = y_len.to(at::kInt).contiguous();
+  auto gl = grad_loss.to(at::kFloat).contiguous();
+  const long B = xc.size(0), Tm = xc.size(1), U = xc.size(2), V = xc.size(3);
+  auto fopts = xc.options().dtype(at::kFloat);
+  auto beta = at::full({B, Tm, U}, -INFINITY, fopts);
+  auto dx = at::zeros_like(xc);
+  APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "transducer_loss_backward", ([&] {
+    hipLaunchKernelGGL((rnnt_beta_kernel<scalar_t>), dim3((uint32_t)B), dim3(TJ_BLOCK), 0,
+                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
+                       beta.data_ptr<float>(), fl.data_ptr<int>(), yl.data_ptr<int>(), B, Tm, U,
+                       V, (int)blank_idx);
+    HIP_CHECK(hipGetLastError());
+    const long total = B * Tm * U;
+    const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
+    hipLaunchKernelGGL((rnnt_grad_kernel<scalar_t>), dim3(grid), dim3(TJ_BLOCK), 0,
+                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
+                       alpha.data_ptr<float>(), beta.data_ptr<float>(), gl.data_ptr<float>(),
+                       (scalar_t*)dx.data_ptr(), fl.data_ptr<int>(), yl.data_ptr<int>(), B, Tm,
+                       U, V, (int)blank_idx);
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return dx;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("joint_forward", &transducer_joint_forward);
+  m.def("joint_backward", &transducer_joint_backward);
+  m.def("loss_forward", &transducer_loss_forward);
+  m.def("loss_backward", &transducer_loss_backward);
+}

@@ -61,6 +61,7 @@ ext_modules = [
     hip_ext("apex_amd._focal_loss", ["csrc/focal_loss.hip"]),
     hip_ext("apex_amd._index_mul_2d", ["csrc/index_mul_2d.hip"]),
     hip_ext("apex_amd._group_norm", ["csrc/group_norm.hip"]),
+    hip_ext("apex_amd._transducer", ["csrc/transducer.hip"]),
 ]
 
 setup(
@@ -80,6 +81,7 @@ setup(
         "apex_amd.contrib.groupbn",
         "apex_amd.contrib.layer_norm",
         "apex_amd.contrib.sparsity",
+        "apex_amd.contrib.transducer",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",

@@ -1,0 +1,90 @@
+"""GPU numerics for the transducer joint and RNN-T loss vs fp32 references."""
+
+import torch
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_transducer_joint_gpu(relu):
+    from apex_amd.contrib.transducer import TransducerJoint
+
+    torch.manual_seed(0)
+    B, T, U, H = 3, 7, 5, 32
+    f = torch.randn(B, T, H, device="cuda", requires_grad=True)
+    g = torch.randn(B, U, H, device="cuda", requires_grad=True)
+    f_len = torch.tensor([7, 5, 6], dtype=torch.int32, device="cuda")
+    g_len = torch.tensor([5, 3, 4], dtype=torch.int32, device="cuda")
+    joint = TransducerJoint(relu=relu)
+    out = joint(f, g, f_len, g_len)
+
+    fr = f.detach().clone().requires_grad_(True)
+    gr = g.detach().clone().requires_grad_(True)
+    ref = fr.unsqueeze(2) + gr.unsqueeze(1)
+    mask_t = torch.arange(T, device="cuda")[None, :, None, None] < f_len[:, None, None, None]
+    mask_u = torch.arange(U, device="cuda")[None, None, :, None] < g_len[:, None, None, None]
+    ref = ref * (mask_t & mask_u)
+    if relu:
+        ref = torch.relu(ref)
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+
+    go = torch.randn_like(out)
+    out.backward(go)
+    ref.backward(go)
+    torch.testing.assert_close(f.grad, fr.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(g.grad, gr.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_transducer_loss_gpu():
+    from apex_amd.contrib.transducer import TransducerLoss
+    from apex_amd.contrib.transducer.transducer import _ref_rnnt_loss
+
+    torch.manual_seed(1)
+    B, T, Umax, V = 2, 6, 4, 10  # Umax = max labels + 1
+    blank = 0
+    y_len = torch.tensor([3, 2], dtype=torch.int32, device="cuda")
+    f_len = torch.tensor([6, 4], dtype=torch.int32, device="cuda")
+    label = torch.randint(1, V, (B, Umax - 1), dtype=torch.int32, device="cuda")
+    logits = torch.randn(B, T, Umax, V, device="cuda", requires_grad=True)
+    x = torch.log_softmax(logits, dim=-1)
+
+    loss_mod = TransducerLoss()
+    losses = loss_mod(x, label, f_len, y_len, blank_idx=blank)
+    ref = _ref_rnnt_loss(x.detach().cpu(), label.cpu().long(), f_len.cpu(), y_len.cpu(), blank)
+    torch.testing.assert_close(losses.cpu(), ref, rtol=1e-4, atol=1e-4)
+
+    # grads vs autograd through the python DP
+    losses.sum().backward()
+    logits2 = logits.detach().cpu().requires_grad_(True)
+    x2 = torch.log_softmax(logits2, dim=-1)
+    _ref_rnnt_loss(x2, label.cpu().long(), f_len.cpu(), y_len.cpu(), blank).sum().backward()
+    torch.testing.assert_close(logits.grad.cpu(), logits2.grad, rtol=1e-3, atol=1e-4)
+
+
+def test_transducer_loss_decreases():
+    """End-to-end sanity: a tiny joint+loss setup trains."""
+    from apex_amd.contrib.transducer import TransducerJoint, TransducerLoss
+
+    torch.manual_seed(2)
+    B, T, Umax, H, V = 2, 5, 3, 16, 8
+    f = torch.randn(B, T, H, device="cuda", requires_grad=True)
+    g = torch.randn(B, Umax, H, device="cuda", requires_grad=True)
+    proj = torch.nn.Linear(H, V).cuda()
+    f_len = torch.full((B,), T, dtype=torch.int32, device="cuda")
+    g_len = torch.full((B,), Umax - 1, dtype=torch.int32, device="cuda")
+    y_len = g_len.clone()
+    label = torch.randint(1, V, (B, Umax - 1), dtype=torch.int32, device="cuda")
+    joint = TransducerJoint()
+    loss_mod = TransducerLoss()
+    opt = torch.optim.Adam([f, g] + list(proj.parameters()), lr=5e-2)
+    losses = []
+    for _ in range(25):
+        opt.zero_grad()
+        h = joint(f, g, f_len, g_len + 1)
+        x = torch.log_softmax(proj(h), dim=-1)
+        loss = loss_mod(x, label, f_len, y_len).mean()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.5
