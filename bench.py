@@ -55,6 +55,10 @@ def main():
     ap.add_argument("--batch", type=int, default=128, help="per-GPU batch size")
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--skip-adam-bench", action="store_true")
+    ap.add_argument("--model", default="resnet50",
+                    choices=["resnet50", "bert", "gpt2"],
+                    help="resnet50 = headline config #2; bert/gpt2 = BASELINE configs #3/#4")
+    ap.add_argument("--seq-len", type=int, default=512)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -83,27 +87,73 @@ def main():
     if distributed:
         dist.barrier()
 
-    # --- part 2: ResNet-50 AMP img/s ---
+    # --- part 2: flagship training step ---
     torch.manual_seed(1234)
-    model = resnet50(num_classes=1000)
-    use_syncbn = has_ext("syncbn")
-    if use_syncbn:
-        model = convert_syncbn_model(model)
-    model = model.to(device)
-    opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
-    model, opt = amp.initialize(model, opt, opt_level="O1",
-                                cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
+    use_syncbn = False
+    if args.model == "resnet50":
+        model = resnet50(num_classes=1000)
+        use_syncbn = has_ext("syncbn")
+        if use_syncbn:
+            model = convert_syncbn_model(model)
+        model = model.to(device)
+        opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
+        model, opt = amp.initialize(model, opt, opt_level="O1",
+                                    cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
+        x = torch.randn(args.batch, 3, args.image_size, args.image_size, device=device)
+        y = torch.randint(0, 1000, (args.batch,), device=device)
+        criterion = torch.nn.CrossEntropyLoss()
+
+        def fwd_loss():
+            return criterion(model(x).float(), y)
+
+        units_per_step = args.batch  # images
+        metric, unit = "resnet50_amp_imgs_per_s", "img/s"
+        config_model = "resnet50"
+    else:
+        # BASELINE configs #3/#4: BERT-base (FusedAdam + FusedLayerNorm +
+        # scaled_masked_softmax) / GPT-2 345M (fused_dense GEMM+bias+GELU +
+        # FusedRMSNorm + FusedLAMB + causal softmax), token throughput.
+        from apex_amd.models.transformer import (
+            BertModel, GPTModel, bert_base_config, gpt2_345m_config,
+        )
+        from apex_amd.optimizers import FusedAdam, FusedLAMB
+        from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
+
+        if args.model == "bert":
+            cfg = bert_base_config(seq_len=args.seq_len)
+            model = BertModel(cfg).to(device)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            config_model = "bert-base"
+        else:
+            cfg = gpt2_345m_config(seq_len=min(args.seq_len, 1024))
+            model = GPTModel(cfg).to(device)
+            opt = FusedLAMB(model.parameters(), lr=1e-4, weight_decay=0.01)
+            config_model = "gpt2-345m"
+        model, opt = amp.initialize(model, opt, opt_level="O2",
+                                    cast_model_type=torch.bfloat16, loss_scale=1.0,
+                                    keep_batchnorm_fp32=False, verbosity=0)
+        tokens = torch.randint(0, cfg.vocab_size, (args.batch, cfg.seq_len), device=device)
+        mask = None
+        if not cfg.causal:
+            mask = torch.zeros(args.batch, 1, cfg.seq_len, cfg.seq_len, dtype=torch.bool,
+                               device=device)
+
+        def fwd_loss():
+            logits = model(tokens, mask) if mask is not None else model(tokens)
+            losses = SoftmaxCrossEntropyLoss.apply(
+                logits.reshape(-1, cfg.vocab_size).contiguous(), tokens.reshape(-1), 0.0, -1, True
+            )
+            return losses.mean()
+
+        units_per_step = args.batch * cfg.seq_len  # tokens
+        metric, unit = f"{args.model}_amp_tokens_per_s", "tokens/s"
+
     if distributed:
         model = DDP(model, message_size=16_000_000)
 
-    x = torch.randn(args.batch, 3, args.image_size, args.image_size, device=device)
-    y = torch.randint(0, 1000, (args.batch,), device=device)
-    criterion = torch.nn.CrossEntropyLoss()
-
     def step():
         opt.zero_grad()
-        out = model(x)
-        loss = criterion(out.float(), y)
+        loss = fwd_loss()
         with amp.scale_loss(loss, opt) as scaled:
             scaled.backward()
         opt.step()
@@ -129,14 +179,14 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    imgs_per_s = world * args.batch * args.steps / elapsed
+    value = world * units_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
         result = {
-            "metric": "resnet50_amp_imgs_per_s",
-            "value": imgs_per_s,
-            "unit": "img/s",
+            "metric": metric,
+            "value": value,
+            "unit": unit,
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -148,11 +198,13 @@ def main():
             "data": "synthetic",
             "fusedadam_350m_ms": adam_ms,
             "config": {
-                "model": "resnet50",
+                "model": config_model,
                 "global_batch": world * args.batch,
-                "image_size": args.image_size,
-                "amp": "O1-bf16",
-                "optimizer": "FusedSGD(momentum=0.9)",
+                "image_size": args.image_size if args.model == "resnet50" else None,
+                "seq_len": None if args.model == "resnet50" else args.seq_len,
+                "amp": "O1-bf16" if args.model == "resnet50" else "O2-bf16",
+                "optimizer": {"resnet50": "FusedSGD(momentum=0.9)", "bert": "FusedAdam",
+                              "gpt2": "FusedLAMB"}[args.model],
                 "syncbn": use_syncbn,
                 "parallelism": f"dp{world}",
             },
