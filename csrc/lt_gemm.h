@@ -107,11 +107,34 @@ inline bool lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m,
   LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES,
                                                  &ws_size, sizeof(ws_size)));
 
+  // algo heuristic cache: one query per (shape, dtype, epilogue) per process
+  // (the heuristic is host-side and would otherwise stall every launch)
+  static std::map<LtAlgoKey, hipblasLtMatmulHeuristicResult_t> algo_cache;
+  static std::mutex algo_mu;
+  LtAlgoKey key{m, n, k, (int)opA, (int)opB, (int)typeA, (int)typeB, (int)typeC, (int)epi};
+
   hipblasLtMatmulHeuristicResult_t heur;
-  int nheur = 0;
-  hipblasStatus_t hst =
-      hipblasLtMatmulAlgoGetHeuristic(handle, desc, layA, layB, layC, layC, pref, 1, &heur, &nheur);
-  bool ok = (hst == HIPBLAS_STATUS_SUCCESS) && nheur > 0;
+  bool ok = false;
+  bool cached = false;
+  {
+    std::lock_guard<std::mutex> lk(algo_mu);
+    auto it = algo_cache.find(key);
+    if (it != algo_cache.end()) {
+      heur = it->second;
+      ok = true;
+      cached = true;
+    }
+  }
+  if (!cached) {
+    int nheur = 0;
+    hipblasStatus_t hst = hipblasLtMatmulAlgoGetHeuristic(handle, desc, layA, layB, layC, layC,
+                                                          pref, 1, &heur, &nheur);
+    ok = (hst == HIPBLAS_STATUS_SUCCESS) && nheur > 0;
+    if (ok) {
+      std::lock_guard<std::mutex> lk(algo_mu);
+      algo_cache.emplace(key, heur);
+    }
+  }
   if (!ok && !allow_fail) {
     TORCH_CHECK(false, "hipBLASLt: no algorithm for this GEMM (m=", m, " n=", n, " k=", k,
                 " epi=", (int)epi, ")");

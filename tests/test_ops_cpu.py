@@ -44,7 +44,9 @@ def test_fused_dense_gelu_dense_parity():
     x2 = x1.detach().clone().requires_grad_(True)
     y1 = m(x1)
     ref = torch.nn.functional.linear(
-        torch.nn.functional.gelu(torch.nn.functional.linear(x2, m.weight1, m.bias1)), m.weight2, m.bias2
+        torch.nn.functional.gelu(torch.nn.functional.linear(x2, m.weight1, m.bias1),
+                                 approximate="tanh"),
+        m.weight2, m.bias2,
     )
     torch.testing.assert_close(y1, ref)
     y1.sum().backward()

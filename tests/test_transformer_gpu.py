@@ -205,22 +205,23 @@ def test_fused_dense_gelu_dense_gpu(dtype):
     w2 = m.weight2.detach().float().clone().requires_grad_(True)
     b2 = m.bias2.detach().float().clone().requires_grad_(True)
     y = m(x)
+    # hipBLASLt's GELU epilogue is the tanh approximation (probed)
     y_ref = torch.nn.functional.linear(
-        torch.nn.functional.gelu(torch.nn.functional.linear(xr, w1, b1)), w2, b2
+        torch.nn.functional.gelu(torch.nn.functional.linear(xr, w1, b1), approximate="tanh"),
+        w2, b2,
     )
     torch.testing.assert_close(y.float(), y_ref, **tol_for(dtype))
     g = torch.randn_like(y)
     y.backward(g)
     y_ref.backward(g.float())
-    # dGELU runs on the low-precision gelu_in saved by the epilogue (and the
-    # library may use the tanh approximation) → allow a few percent of
-    # moderate outliers on the grads flowing through it.
-    wtol = {k: max(v * 16, 0.03) for k, v in tol_for(dtype).items()}
-    assert_mostly_close(x.grad.float(), xr.grad, wtol["rtol"], wtol["atol"], 0.05)
-    assert_mostly_close(m.weight1.grad.float(), w1.grad, wtol["rtol"], wtol["atol"], 0.05)
-    assert_mostly_close(m.bias1.grad.float(), b1.grad, 0.05, 0.5, 0.05)
-    assert_mostly_close(m.weight2.grad.float(), w2.grad, wtol["rtol"], wtol["atol"], 0.02)
-    assert_mostly_close(m.bias2.grad.float(), b2.grad, wtol["rtol"], wtol["atol"], 0.02)
+    # dGELU runs on the low-precision gelu_in saved by the epilogue → a few
+    # grad elements near the GELU knee legitimately exceed the tolerance.
+    wtol = {k: v * 16 for k, v in tol_for(dtype).items()}
+    assert_mostly_close(x.grad.float(), xr.grad, wtol["rtol"], wtol["atol"], 0.01)
+    assert_mostly_close(m.weight1.grad.float(), w1.grad, wtol["rtol"], wtol["atol"], 0.01)
+    assert_mostly_close(m.bias1.grad.float(), b1.grad, wtol["rtol"], max(wtol["atol"], 0.1), 0.02)
+    assert_mostly_close(m.weight2.grad.float(), w2.grad, wtol["rtol"], wtol["atol"], 0.01)
+    assert_mostly_close(m.bias2.grad.float(), b2.grad, wtol["rtol"], wtol["atol"], 0.01)
 
 
 @pytest.mark.parametrize("dtype", [torch.float32, torch.float16, torch.bfloat16])
