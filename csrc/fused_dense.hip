@@ -4,6 +4,7 @@
 // backward, linear_gelu_linear_forward/backward) and
 // csrc/megatron/fused_weight_gradient_dense.cpp:11-13 (wgrad_gemm_accum_*).
 #include "lt_gemm.h"
+#include "multi_tensor_apply.h"  // VecPack
 
 #include <vector>
 
@@ -21,26 +22,58 @@ at::Tensor flat2d(const at::Tensor& t) {
 constexpr float kGeluC = 0.7978845608028654f;   // sqrt(2/pi)
 constexpr float kGeluA = 0.044715f;
 
+__device__ __forceinline__ float gelu_tanh(float v) {
+  const float t = tanhf(kGeluC * (v + kGeluA * v * v * v));
+  return 0.5f * v * (1.f + t);
+}
+
+__device__ __forceinline__ float dgelu_tanh(float v) {
+  const float t = tanhf(kGeluC * (v + kGeluA * v * v * v));
+  const float dt = (1.f - t * t) * kGeluC * (1.f + 3.f * kGeluA * v * v);
+  return 0.5f * (1.f + t) + 0.5f * v * dt;
+}
+
 template <typename T>
 __global__ void __launch_bounds__(256) gelu_fwd_kernel(const T* __restrict__ z,
                                                        T* __restrict__ y, long n) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x) {
-    const float v = to_float(z[i]);
-    const float t = tanhf(kGeluC * (v + kGeluA * v * v * v));
-    y[i] = from_float<T>(0.5f * v * (1.f + t));
+  constexpr int W = VecPack<T>::width;
+  if ((n % W) == 0) {
+    for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * W; i < n;
+         i += (long)gridDim.x * blockDim.x * W) {
+      VecPack<T> v, r;
+      load_pack(v, z + i);
+#pragma unroll
+      for (int j = 0; j < W; ++j) r.a[j] = from_float<T>(gelu_tanh(to_float(v.a[j])));
+      store_pack(y + i, r);
+    }
+  } else {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+      y[i] = from_float<T>(gelu_tanh(to_float(z[i])));
+    }
   }
 }
 
 template <typename T>
 __global__ void __launch_bounds__(256) dgelu_kernel(T* __restrict__ d, const T* __restrict__ z,
                                                     long n) {
-  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (long)gridDim.x * blockDim.x) {
-    const float v = to_float(z[i]);
-    const float t = tanhf(kGeluC * (v + kGeluA * v * v * v));
-    const float dt = (1.f - t * t) * kGeluC * (1.f + 3.f * kGeluA * v * v);
-    d[i] = from_float<T>(to_float(d[i]) * (0.5f * (1.f + t) + 0.5f * v * dt));
+  constexpr int W = VecPack<T>::width;
+  if ((n % W) == 0) {
+    for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * W; i < n;
+         i += (long)gridDim.x * blockDim.x * W) {
+      VecPack<T> vd, vz;
+      load_pack(vd, d + i);
+      load_pack(vz, z + i);
+#pragma unroll
+      for (int j = 0; j < W; ++j)
+        vd.a[j] = from_float<T>(to_float(vd.a[j]) * dgelu_tanh(to_float(vz.a[j])));
+      store_pack(d + i, vd);
+    }
+  } else {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+      d[i] = from_float<T>(to_float(d[i]) * dgelu_tanh(to_float(z[i])));
+    }
   }
 }
 

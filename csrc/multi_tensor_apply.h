@@ -154,3 +154,25 @@ template <typename T>
 __device__ __host__ __forceinline__ bool is_vec4_aligned(const void* p) {
   return (reinterpret_cast<uintptr_t>(p) & (sizeof(T) * 4 - 1)) == 0;
 }
+
+// 16-byte pack: 8 elements for 2-byte types, 4 for fp32 (G13 sweet spot).
+template <typename T>
+struct VecPack {
+  static constexpr int width = 16 / sizeof(T);
+  alignas(16) T a[width];
+};
+
+template <typename T>
+__device__ __forceinline__ void load_pack(VecPack<T>& dst, const T* src) {
+  *reinterpret_cast<uint4*>(dst.a) = *reinterpret_cast<const uint4*>(src);
+}
+
+template <typename T>
+__device__ __forceinline__ void store_pack(T* dst, const VecPack<T>& src) {
+  *reinterpret_cast<uint4*>(dst) = *reinterpret_cast<const uint4*>(src.a);
+}
+
+template <typename T>
+__device__ __host__ __forceinline__ bool is_pack_aligned(const void* p) {
+  return (reinterpret_cast<uintptr_t>(p) & 15) == 0;
+}

@@ -10,6 +10,7 @@
 // is bounded at sk<=16384; the block-streaming form handles any sk, so
 // "generic" binds to the same kernels). Masked positions write exact 0.
 #include "common.h"
+#include "multi_tensor_apply_hip.h"  // VecPack 16B loads
 
 namespace {
 
@@ -58,11 +59,14 @@ __device__ void block_online_sm(OnlineSM& o, float* smem /* 2*nwaves */) {
 }
 
 // MODE: 0 = plain, 1 = additive bool mask [b,1,sq,sk], 2 = causal upper-tri
-template <typename T, int MODE>
+// VEC: 16-byte packed loads (8 x bf16/fp16 per lane — G13); the scalar
+// variant (VEC=false) covers odd sk / unaligned rows.
+template <typename T, int MODE, bool VEC>
 __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_kernel(
     const T* __restrict__ in, T* __restrict__ out, const uint8_t* __restrict__ mask, float scale,
     long rows, long sk, long np, long sq) {
   __shared__ float smem[2 * (SM_BLOCK / WAVE_SIZE)];
+  constexpr int W = VecPack<T>::width;
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* x = in + row * sk;
     T* y = out + row * sk;
@@ -79,44 +83,100 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_kernel(
     }
 
     OnlineSM o;
-    for (long i = threadIdx.x; i < limit; i += blockDim.x) {
-      float v = to_float(x[i]) * scale;
-      if (MODE == 1 && mrow[i]) v = -10000.0f;
-      o.add(v);
+    if (VEC) {
+      for (long i = (long)threadIdx.x * W; i < sk; i += (long)blockDim.x * W) {
+        VecPack<T> v;
+        load_pack(v, x + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          if (MODE == 2 && i + j >= limit) break;
+          float f = to_float(v.a[j]) * scale;
+          if (MODE == 1 && mrow[i + j]) f = -10000.0f;
+          o.add(f);
+        }
+      }
+    } else {
+      for (long i = threadIdx.x; i < limit; i += blockDim.x) {
+        float v = to_float(x[i]) * scale;
+        if (MODE == 1 && mrow[i]) v = -10000.0f;
+        o.add(v);
+      }
     }
     block_online_sm(o, smem);
     const float inv_s = o.s > 0.f ? 1.f / o.s : 0.f;
 
-    for (long i = threadIdx.x; i < sk; i += blockDim.x) {
-      float r = 0.f;
-      if (i < limit) {
-        float v = to_float(x[i]) * scale;
-        if (MODE == 1 && mrow[i]) v = -10000.0f;
-        r = __expf(v - o.m) * inv_s;
+    if (VEC) {
+      for (long i = (long)threadIdx.x * W; i < sk; i += (long)blockDim.x * W) {
+        VecPack<T> v, r;
+        load_pack(v, x + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float f = 0.f;
+          if (MODE != 2 || i + j < limit) {
+            float vv = to_float(v.a[j]) * scale;
+            if (MODE == 1 && mrow[i + j]) vv = -10000.0f;
+            f = __expf(vv - o.m) * inv_s;
+          }
+          r.a[j] = from_float<T>(f);
+        }
+        store_pack(y + i, r);
       }
-      y[i] = from_float<T>(r);
+    } else {
+      for (long i = threadIdx.x; i < sk; i += blockDim.x) {
+        float r = 0.f;
+        if (i < limit) {
+          float v = to_float(x[i]) * scale;
+          if (MODE == 1 && mrow[i]) v = -10000.0f;
+          r = __expf(v - o.m) * inv_s;
+        }
+        y[i] = from_float<T>(r);
+      }
     }
     __syncthreads();
   }
 }
 
 // grad = y * (dy - sum(dy*y)) * scale
-template <typename T>
+template <typename T, bool VEC>
 __global__ void __launch_bounds__(SM_BLOCK) softmax_bwd_kernel(
     const T* __restrict__ dy_ptr, const T* __restrict__ y_ptr, T* __restrict__ dx_ptr,
     float scale, long rows, long sk) {
   __shared__ float smem[SM_BLOCK / WAVE_SIZE];
+  constexpr int W = VecPack<T>::width;
   for (long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dy = dy_ptr + row * sk;
     const T* y = y_ptr + row * sk;
     T* dx = dx_ptr + row * sk;
     float acc = 0.f;
-    for (long i = threadIdx.x; i < sk; i += blockDim.x)
-      acc = fmaf(to_float(dy[i]), to_float(y[i]), acc);
+    if (VEC) {
+      for (long i = (long)threadIdx.x * W; i < sk; i += (long)blockDim.x * W) {
+        VecPack<T> vd, vy;
+        load_pack(vd, dy + i);
+        load_pack(vy, y + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) acc = fmaf(to_float(vd.a[j]), to_float(vy.a[j]), acc);
+      }
+    } else {
+      for (long i = threadIdx.x; i < sk; i += blockDim.x)
+        acc = fmaf(to_float(dy[i]), to_float(y[i]), acc);
+    }
     float dot = block_reduce_sum(acc, smem);
-    for (long i = threadIdx.x; i < sk; i += blockDim.x) {
-      float yv = to_float(y[i]);
-      dx[i] = from_float<T>(yv * (to_float(dy[i]) - dot) * scale);
+    if (VEC) {
+      for (long i = (long)threadIdx.x * W; i < sk; i += (long)blockDim.x * W) {
+        VecPack<T> vd, vy, r;
+        load_pack(vd, dy + i);
+        load_pack(vy, y + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          r.a[j] = from_float<T>(to_float(vy.a[j]) * (to_float(vd.a[j]) - dot) * scale);
+        }
+        store_pack(dx + i, r);
+      }
+    } else {
+      for (long i = threadIdx.x; i < sk; i += blockDim.x) {
+        float yv = to_float(y[i]);
+        dx[i] = from_float<T>(yv * (to_float(dy[i]) - dot) * scale);
+      }
     }
     __syncthreads();
   }
@@ -134,11 +194,18 @@ at::Tensor fwd_impl(const at::Tensor& input, const c10::optional<at::Tensor>& ma
   at::Tensor m8;
   if (MODE == 1) m8 = mask->to(at::kByte).contiguous();
   APEX_DISPATCH_FLOAT_HALF_BF(x.scalar_type(), "scaled_softmax_forward", ([&] {
-    hipLaunchKernelGGL((softmax_fwd_kernel<scalar_t, MODE>), dim3(sm_grid(rows)), dim3(SM_BLOCK),
-                       0, current_stream(), (const scalar_t*)x.data_ptr(),
-                       (scalar_t*)y.data_ptr(),
-                       MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr, (float)scale, rows, sk, np,
-                       sq);
+    const bool vec = (sk % VecPack<scalar_t>::width == 0) && is_pack_aligned<scalar_t>(x.data_ptr());
+    if (vec) {
+      hipLaunchKernelGGL((softmax_fwd_kernel<scalar_t, MODE, true>), dim3(sm_grid(rows)),
+                         dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)x.data_ptr(),
+                         (scalar_t*)y.data_ptr(), MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr,
+                         (float)scale, rows, sk, np, sq);
+    } else {
+      hipLaunchKernelGGL((softmax_fwd_kernel<scalar_t, MODE, false>), dim3(sm_grid(rows)),
+                         dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)x.data_ptr(),
+                         (scalar_t*)y.data_ptr(), MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr,
+                         (float)scale, rows, sk, np, sq);
+    }
   }()));
   HIP_CHECK(hipGetLastError());
   return y;
@@ -151,10 +218,20 @@ at::Tensor bwd_impl(const at::Tensor& grad_out, const at::Tensor& softmax_out, d
   const long sk = y.size(-1);
   const long rows = y.numel() / sk;
   APEX_DISPATCH_FLOAT_HALF_BF(y.scalar_type(), "scaled_softmax_backward", ([&] {
-    hipLaunchKernelGGL((softmax_bwd_kernel<scalar_t>), dim3(sm_grid(rows)), dim3(SM_BLOCK), 0,
-                       current_stream(), (const scalar_t*)dy.data_ptr(),
-                       (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
-                       rows, sk);
+    const bool vec = (sk % VecPack<scalar_t>::width == 0) &&
+                     is_pack_aligned<scalar_t>(y.data_ptr()) &&
+                     is_pack_aligned<scalar_t>(dy.data_ptr());
+    if (vec) {
+      hipLaunchKernelGGL((softmax_bwd_kernel<scalar_t, true>), dim3(sm_grid(rows)),
+                         dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)dy.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
+                         rows, sk);
+    } else {
+      hipLaunchKernelGGL((softmax_bwd_kernel<scalar_t, false>), dim3(sm_grid(rows)),
+                         dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)dy.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
+                         rows, sk);
+    }
   }()));
   HIP_CHECK(hipGetLastError());
   return dx;

@@ -38,14 +38,16 @@ def main():
         opt.step()
 
     import time
-    for _ in range(3):
+    nwarm = int(os.environ.get("PROF_WARMUP", "3"))
+    nsteps = int(os.environ.get("PROF_STEPS", "5"))
+    for _ in range(nwarm):
         step()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(5):
+    for _ in range(nsteps):
         step()
     torch.cuda.synchronize()
-    print(f"bert step: {(time.perf_counter() - t0) / 5 * 1000:.1f} ms")
+    print(f"bert step: {(time.perf_counter() - t0) / nsteps * 1000:.1f} ms")
 
 
 if __name__ == "__main__":
