@@ -10,8 +10,15 @@ def main(d, prefix):
     kt = list(csv.DictReader(open(f"{d}/{prefix}_kernel_trace.csv")))
     evs = sorted((int(r["Start_Timestamp"]), int(r["End_Timestamp"]), r["Kernel_Name"]) for r in kt)
     adam = [e for s, e, n in evs if "AdamFunctor" in n]
-    if len(adam) >= 2:
-        w0, w1 = adam[-2], adam[-1]
+    # optimizer launches cluster at step ends; keep the last of each cluster
+    step_ends = []
+    for t in adam:
+        if not step_ends or t - step_ends[-1] > 5_000_000:  # >5 ms apart
+            step_ends.append(t)
+        else:
+            step_ends[-1] = t
+    if len(step_ends) >= 2:
+        w0, w1 = step_ends[-2], step_ends[-1]
     else:
         w0, w1 = evs[0][0], evs[-1][1]
     window = [(s, e, n) for s, e, n in evs if s >= w0 and e <= w1]
