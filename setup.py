@@ -62,6 +62,9 @@ ext_modules = [
     hip_ext("apex_amd._index_mul_2d", ["csrc/index_mul_2d.hip"]),
     hip_ext("apex_amd._group_norm", ["csrc/group_norm.hip"]),
     hip_ext("apex_amd._transducer", ["csrc/transducer.hip"]),
+    hip_ext("apex_amd._peer_memory", ["csrc/peer_memory.hip"]),
+    hip_ext("apex_amd._rccl_p2p", ["csrc/rccl_p2p.hip"], libraries=["rccl"]),
+    hip_ext("apex_amd._rccl_allocator", ["csrc/rccl_allocator.cpp"], libraries=["rccl"]),
 ]
 
 setup(
@@ -82,6 +85,9 @@ setup(
         "apex_amd.contrib.layer_norm",
         "apex_amd.contrib.sparsity",
         "apex_amd.contrib.transducer",
+        "apex_amd.contrib.peer_memory",
+        "apex_amd.contrib.nccl_allocator",
+        "apex_amd.contrib.rccl_p2p",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",
