@@ -108,8 +108,11 @@ inline bool lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m,
                                                  &ws_size, sizeof(ws_size)));
 
   // algo heuristic cache: one query per (shape, dtype, epilogue) per process
-  // (the heuristic is host-side and would otherwise stall every launch)
-  static std::map<LtAlgoKey, hipblasLtMatmulHeuristicResult_t> algo_cache;
+  // (the heuristic is host-side and would otherwise stall every launch).
+  // Failures are cached too — an unsupported epilogue (e.g. bf16
+  // GELU_AUX_BIAS) costs a ~27 ms Tensile solution sweep per MISS, and the
+  // caller's fallback path would otherwise pay it on every step.
+  static std::map<LtAlgoKey, std::pair<bool, hipblasLtMatmulHeuristicResult_t>> algo_cache;
   static std::mutex algo_mu;
   LtAlgoKey key{m, n, k, (int)opA, (int)opB, (int)typeA, (int)typeB, (int)typeC, (int)epi};
 
@@ -120,8 +123,8 @@ inline bool lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m,
     std::lock_guard<std::mutex> lk(algo_mu);
     auto it = algo_cache.find(key);
     if (it != algo_cache.end()) {
-      heur = it->second;
-      ok = true;
+      ok = it->second.first;
+      heur = it->second.second;
       cached = true;
     }
   }
@@ -130,10 +133,8 @@ inline bool lt_matmul_cm(hipblasOperation_t opA, hipblasOperation_t opB, long m,
     hipblasStatus_t hst = hipblasLtMatmulAlgoGetHeuristic(handle, desc, layA, layB, layC, layC,
                                                           pref, 1, &heur, &nheur);
     ok = (hst == HIPBLAS_STATUS_SUCCESS) && nheur > 0;
-    if (ok) {
-      std::lock_guard<std::mutex> lk(algo_mu);
-      algo_cache.emplace(key, heur);
-    }
+    std::lock_guard<std::mutex> lk(algo_mu);
+    algo_cache.emplace(key, std::make_pair(ok, heur));
   }
   if (!ok && !allow_fail) {
     TORCH_CHECK(false, "hipBLASLt: no algorithm for this GEMM (m=", m, " n=", n, " k=", k,
