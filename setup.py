@@ -88,6 +88,9 @@ setup(
         "apex_amd.contrib.peer_memory",
         "apex_amd.contrib.nccl_allocator",
         "apex_amd.contrib.rccl_p2p",
+        "apex_amd.contrib.conv_bias_relu",
+        "apex_amd.contrib.bottleneck",
+        "apex_amd.contrib.gbn",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",
