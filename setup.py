@@ -91,6 +91,8 @@ setup(
         "apex_amd.contrib.conv_bias_relu",
         "apex_amd.contrib.bottleneck",
         "apex_amd.contrib.gbn",
+        "apex_amd.contrib.openfold",
+        "apex_amd.contrib.torchsched",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",

@@ -1,0 +1,67 @@
+"""CPU tests for contrib.openfold and torchsched surfaces."""
+
+import math
+
+import torch
+import pytest
+
+
+def test_fused_adam_swa_cpu():
+    from apex_amd.contrib.openfold import FusedAdamSWA
+
+    torch.manual_seed(0)
+    ps = [torch.randn(16, 16, requires_grad=True), torch.randn(8, requires_grad=True)]
+    opt = FusedAdamSWA(ps, swa_decay_rate=0.9, lr=1e-2)
+    swa_before = [s.clone() for s in opt.swa_params]
+    for p in ps:
+        p.grad = torch.randn_like(p)
+    opt.step()
+    for s, s0, p in zip(opt.swa_params, swa_before, ps):
+        expected = 0.9 * s0 + 0.1 * p.detach()
+        torch.testing.assert_close(s, expected, rtol=1e-5, atol=1e-6)
+
+
+def test_openfold_mha_cpu():
+    from apex_amd.contrib.openfold import AttnTri
+
+    torch.manual_seed(1)
+    b, h, s, d = 2, 4, 16, 32
+    q = torch.randn(b, h, s, d)
+    k = torch.randn(b, h, s, d)
+    v = torch.randn(b, h, s, d)
+    out = AttnTri(q, k, v, mask=None)
+    ref = torch.softmax(q @ k.transpose(-2, -1) / math.sqrt(d), dim=-1) @ v
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+
+
+def test_openfold_layer_norm_cpu():
+    from apex_amd.contrib.openfold import LayerNormSmallShapeOptImpl
+
+    # CPU path: module-level fallback exercises the same entry points
+    from apex_amd.normalization import FusedLayerNorm
+
+    ln = FusedLayerNorm(32)
+    x = torch.randn(4, 32)
+    ref = torch.nn.functional.layer_norm(x, (32,), ln.weight, ln.bias, ln.eps)
+    torch.testing.assert_close(ln(x), ref)
+
+
+def test_torchsched_backend_registered():
+    from apex_amd.contrib import torchsched
+
+    backend = torchsched.get_backend()
+    assert callable(backend)
+    assert torchsched.set_default_backend("torchsched") == "torchsched"
+
+
+def test_torchsched_compiles_a_function():
+    import torch._dynamo as dynamo
+
+    dynamo.reset()
+
+    def f(a, b):
+        return torch.relu(a @ b) + 1
+
+    compiled = torch.compile(f, backend="torchsched")
+    a, b = torch.randn(4, 8), torch.randn(8, 4)
+    torch.testing.assert_close(compiled(a, b), f(a, b))
