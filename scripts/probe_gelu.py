@@ -1,47 +1,78 @@
-"""Probe: which GELU does hipBLASLt's epilogue implement (erf vs tanh)?
-And how do the fused MLP grads compare against same-dtype torch references?"""
-
-import torch
+"""Probe: per-tensor error of linear_gelu_linear fwd/bwd vs fp32 tanh-GELU
+chain (bf16 path uses the split-epilogue fallback), and the O0/O2/O3 ResNet
+first-loss comparison."""
 
 import os
 import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+import torch
 
 import apex_amd._fused_dense as fd
 
 
-def main():
-    torch.manual_seed(0)
-    n = 256
-    x = torch.linspace(-4, 4, n, device="cuda", dtype=torch.float32).reshape(1, n).t().contiguous()  # [n,1]
-    w = torch.ones(1, 1, device="cuda")
-    b = torch.zeros(1, device="cuda")
-    w2 = torch.ones(1, 1, device="cuda")
-    b2 = torch.zeros(1, device="cuda")
-    o1, o2, gelu_in = fd.linear_gelu_linear_forward(x, w, b, w2, b2)
-    y = o1.flatten()
-    xin = x.flatten()
-    erf = torch.nn.functional.gelu(xin)
-    tanh = torch.nn.functional.gelu(xin, approximate="tanh")
-    print("max |y-erf| :", (y - erf).abs().max().item())
-    print("max |y-tanh|:", (y - tanh).abs().max().item())
+def probe_gelu(dtype):
+    torch.manual_seed(8)
+    m, nin, nh, nout = 32, 64, 256, 48
+    x = torch.randn(m, nin, device="cuda", dtype=dtype)
+    w1 = torch.randn(nh, nin, device="cuda", dtype=dtype) * 0.1
+    b1 = torch.randn(nh, device="cuda", dtype=dtype)
+    w2 = torch.randn(nout, nh, device="cuda", dtype=dtype) * 0.1
+    b2 = torch.randn(nout, device="cuda", dtype=dtype)
+    o1, o2, gi = fd.linear_gelu_linear_forward(x, w1, b1, w2, b2)
 
-    # dgelu probe via linear_gelu_linear_backward
-    gi = xin.reshape(-1, 1)
-    dy = torch.ones_like(gi)
-    out = fd.linear_gelu_linear_backward(gi, gi, o1, w, w2, dy)
-    dx = out[0].flatten()
-    xg = xin.clone().requires_grad_(True)
-    torch.nn.functional.gelu(xg).sum().backward()
-    derf = xg.grad.clone()
-    xg2 = xin.clone().requires_grad_(True)
-    torch.nn.functional.gelu(xg2, approximate="tanh").sum().backward()
-    dtanh = xg2.grad.clone()
-    print("max |dx-derf| :", (dx - derf).abs().max().item())
-    print("max |dx-dtanh|:", (dx - dtanh).abs().max().item())
+    xf = x.float().requires_grad_(True)
+    w1f = w1.float().requires_grad_(True)
+    b1f = b1.float().requires_grad_(True)
+    w2f = w2.float().requires_grad_(True)
+    b2f = b2.float().requires_grad_(True)
+    z1 = torch.nn.functional.linear(xf, w1f, b1f)
+    o1f = torch.nn.functional.gelu(z1, approximate="tanh")
+    o2f = torch.nn.functional.linear(o1f, w2f, b2f)
+
+    print(f"[{dtype}] fwd: |gi-z1| {(gi.float()-z1).abs().max():.4f} "
+          f"|o1-ref| {(o1.float()-o1f).abs().max():.4f} |o2-ref| {(o2.float()-o2f).abs().max():.4f}")
+
+    dy = torch.randn(m, nout, device="cuda", dtype=dtype)
+    dx, dw1, db1, dw2, db2 = fd.linear_gelu_linear_backward(x, gi, o1, w1, w2, dy)
+    o2f.backward(dy.float())
+    for name, a, b in [("dx", dx, xf.grad), ("dw1", dw1, w1f.grad), ("db1", db1, b1f.grad),
+                       ("dw2", dw2, w2f.grad), ("db2", db2, b2f.grad)]:
+        err = (a.float() - b).abs().max().item()
+        rel = err / (b.abs().max().item() + 1e-9)
+        print(f"[{dtype}] bwd {name}: max abs {err:.5f} rel {rel:.4f}")
+
+
+def probe_resnet_o2():
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from apex_amd.models import resnet50
+
+    torch.manual_seed(7)
+    model = resnet50(num_classes=100).cuda()
+    gen = torch.Generator().manual_seed(7)
+    x = torch.randn(16, 3, 96, 96, generator=gen).cuda()
+    y = torch.randint(0, 100, (16,), generator=gen).cuda()
+
+    with torch.no_grad():
+        loss0 = torch.nn.functional.cross_entropy(model(x).float(), y)
+        # O3-style: full bf16
+        m3 = resnet50(num_classes=100)
+        torch.manual_seed(7)
+        m3 = resnet50(num_classes=100).cuda().to(torch.bfloat16)
+        loss3 = torch.nn.functional.cross_entropy(m3(x.bfloat16()).float(), y)
+        # O2-style: bf16 except BN fp32
+        torch.manual_seed(7)
+        m2 = resnet50(num_classes=100).cuda().to(torch.bfloat16)
+        for mod in m2.modules():
+            if isinstance(mod, torch.nn.modules.batchnorm._BatchNorm):
+                mod.float()
+        loss2 = torch.nn.functional.cross_entropy(m2(x.bfloat16()).float(), y)
+    print(f"resnet first loss: O0 {loss0:.4f}  O3-bf16 {loss3:.4f}  O2-bf16+bnfp32 {loss2:.4f}")
 
 
 if __name__ == "__main__":
-    main()
+    probe_gelu(torch.bfloat16)
+    probe_gelu(torch.float16)
+    probe_gelu(torch.float32)
+    probe_resnet_o2()
