@@ -159,9 +159,12 @@ std::vector<at::Tensor> linear_gelu_linear_forward(at::Tensor input, at::Tensor 
   auto output1 = at::empty({m, n1}, x.options());   // GELU(X@W1^T + b1)
   auto gelu_in = at::empty({m, n1}, x.options());   // pre-GELU (post-bias) aux
   auto output2 = at::empty({m, n2}, x.options());
-  if (!lt_linear(x, w1, output1, &b1, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, &gelu_in,
-                 /*allow_fail=*/true)) {
-    // split epilogue: BIAS GEMM into gelu_in, then elementwise erf-GELU
+  // GELU_AUX_BIAS trusted for fp16 only (see the DGELU note in backward).
+  const bool fused_gelu = x.scalar_type() == at::ScalarType::Half &&
+                          lt_linear(x, w1, output1, &b1, HIPBLASLT_EPILOGUE_GELU_AUX_BIAS,
+                                    &gelu_in, /*allow_fail=*/true);
+  if (!fused_gelu) {
+    // split epilogue: BIAS GEMM into gelu_in, then elementwise tanh-GELU
     lt_linear(x, w1, gelu_in, &b1, HIPBLASLT_EPILOGUE_BIAS, nullptr);
     gelu_fwd_inplace(gelu_in, output1);
   }
@@ -186,11 +189,18 @@ std::vector<at::Tensor> linear_gelu_linear_backward(at::Tensor input, at::Tensor
   auto db2 = at::empty({w2.size(0)}, w2.options());
   lt_linear_wgrad(o1, dy2, dw2, HIPBLASLT_EPILOGUE_BGRADB, &db2, 0.f);
 
-  // d_gelu = dGELU(dY2 @ W2, gelu_in), db1 fused
+  // d_gelu = dGELU(dY2 @ W2, gelu_in), db1 fused.
+  // The DGELU_BGRAD epilogue is only trusted for fp16: for bf16 the
+  // heuristic reports algorithms but the aux (gelu_in) is misinterpreted and
+  // the output is wrong (probed on MI355X: dx rel-err 0.8); fp32 has no
+  // algorithms. Other dtypes take the split path (plain dgrad GEMM +
+  // vectorized tanh-dGELU pass + bias-grad reduction).
   auto d_gelu = at::empty({m, n1}, x.options());
   auto db1 = at::empty({n1}, w1.options());
-  if (!lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DGELU_BGRAD, &gi, &db1,
-                       /*allow_fail=*/true)) {
+  const bool fused_dgelu = x.scalar_type() == at::ScalarType::Half &&
+                           lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DGELU_BGRAD, &gi,
+                                           &db1, /*allow_fail=*/true);
+  if (!fused_dgelu) {
     lt_linear_dgrad(dy2, w2, d_gelu, HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr);
     dgelu_inplace(d_gelu, gi);
     db1.copy_(d_gelu.sum(0).to(db1.scalar_type()));

@@ -71,8 +71,42 @@ def probe_resnet_o2():
     print(f"resnet first loss: O0 {loss0:.4f}  O3-bf16 {loss3:.4f}  O2-bf16+bnfp32 {loss2:.4f}")
 
 
-if __name__ == "__main__":
+def run_all():
     probe_gelu(torch.bfloat16)
     probe_gelu(torch.float16)
     probe_gelu(torch.float32)
     probe_resnet_o2()
+
+
+def probe_train_o2():
+    """Replicate the integration test's O2 _train first loss, bisecting amp."""
+    from apex_amd import amp
+    from apex_amd.amp._amp_state import _amp_state
+    from apex_amd.models import resnet50
+    from apex_amd.optimizers import FusedSGD
+
+    def first_loss(use_amp, opt_level="O2"):
+        _amp_state.reset()
+        torch.manual_seed(7)
+        torch.backends.cudnn.deterministic = True
+        model = resnet50(num_classes=100).cuda()
+        opt = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
+        if use_amp:
+            model, opt = amp.initialize(model, opt, opt_level=opt_level,
+                                        cast_model_type=None if opt_level in ("O0","O1") else torch.bfloat16,
+                                        loss_scale=128.0, verbosity=0)
+        gen = torch.Generator().manual_seed(7)
+        x = torch.randn(16, 3, 96, 96, generator=gen).cuda()
+        y = torch.randint(0, 100, (16,), generator=gen).cuda()
+        out = model(x)
+        return float(torch.nn.functional.cross_entropy(out.float(), y))
+
+    print("train-first-loss O0:", first_loss(True, "O0"))
+    print("train-first-loss O2 amp:", first_loss(True, "O2"))
+    print("train-first-loss O3 amp:", first_loss(True, "O3"))
+    print("train-first-loss no-amp fp32:", first_loss(False))
+
+
+if __name__ == "__main__":
+    run_all()
+    probe_train_o2()
