@@ -220,6 +220,18 @@ def _patch_step_for_skip_and_copy(optimizer, needs_master_copy):
     optimizer.step = new_step
     optimizer._amp_step_patched = True
 
+    if hasattr(optimizer, "_amp_stash"):
+        old_zero = optimizer.zero_grad
+
+        @functools.wraps(old_zero)
+        def new_zero_grad(set_to_none=True):
+            out = old_zero(set_to_none)
+            for p in optimizer._amp_stash.all_fp16_params:
+                p.grad = None
+            return out
+
+        optimizer.zero_grad = new_zero_grad
+
 
 def initialize(
     models,
@@ -353,6 +365,11 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
         if hasattr(opt, "_amp_stash"):  # O2: unscale fp16 grads into masters
             model_grads, master_grads = _materialize_master_grads(opt, loss_scale)
             ov = scaler.unscale_grads(model_grads, master_grads)
+            # model grads are consumed into the masters here; clear them so
+            # the next backward doesn't accumulate stale gradients (the
+            # optimizer's zero_grad only sees the master params)
+            for p in opt._amp_stash.all_fp16_params:
+                p.grad = None
             # fp32 params' grads unscaled in place
             stash = opt._amp_stash
             fp32_grads = [p.grad for p in stash.all_fp32_from_fp32_params if p.grad is not None]

@@ -55,14 +55,20 @@ def test_o0_reproducible():
                                                   ("O2", "dynamic"), ("O2", 128.0),
                                                   ("O3", 1.0)])
 def test_opt_levels_track_o0(opt_level, loss_scale):
-    """Mixed-precision loss curves must track the fp32 baseline closely."""
-    base = _train("O0", None)
-    test = _train(opt_level, loss_scale)
+    """Mixed-precision loss curves must track the fp32 baseline closely.
+    Dynamic scaling legitimately skips the first ~dozen steps while the
+    scale backs off from 2^16, so those runs get more iterations and a
+    looser end-point check."""
+    dynamic = loss_scale == "dynamic"
+    iters = 40 if dynamic else 12
+    base = _train("O0", None, iters=iters)
+    test = _train(opt_level, loss_scale, iters=iters)
     assert len(base) == len(test)
-    # same starting loss (fwd in reduced precision ~1% off), same downward trend
+    # same starting loss (fwd in reduced precision slightly off)
     assert abs(base[0] - test[0]) / base[0] < 0.05
-    assert test[-1] < test[0]
-    assert abs(base[-1] - test[-1]) / base[-1] < 0.25
+    assert test[-1] < test[0]  # training proceeds
+    tol = 0.5 if dynamic else 0.25
+    assert abs(base[-1] - test[-1]) / base[-1] < tol
 
 
 def test_o2_master_params_match_model():
