@@ -68,7 +68,9 @@ def test_opt_levels_track_o0(opt_level, loss_scale):
     assert abs(base[0] - test[0]) / base[0] < 0.05
     assert test[-1] < test[0]  # training proceeds
     tol = 0.5 if dynamic else 0.25
-    assert abs(base[-1] - test[-1]) / base[-1] < tol
+    # relative where losses are O(1), absolute floor once both have converged
+    # to near-zero (relative comparisons of 1e-3 losses are meaningless)
+    assert abs(base[-1] - test[-1]) < max(tol * base[-1], 0.05)
 
 
 def test_o2_master_params_match_model():
