@@ -136,6 +136,102 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_kernel(
   }
 }
 
+// Wave-per-row variant for short rows (sk <= 64*W): each wave64 owns one
+// row — no LDS, no barriers, one vector pass for the online stats and one
+// for the write (4 rows per 256-thread workgroup).
+template <typename T, int MODE>
+__global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_wave_kernel(
+    const T* __restrict__ in, T* __restrict__ out, const uint8_t* __restrict__ mask, float scale,
+    long rows, long sk, long np, long sq) {
+  constexpr int W = VecPack<T>::width;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int waves_per_block = SM_BLOCK / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * waves_per_block + wid; row < rows;
+       row += (long)gridDim.x * waves_per_block) {
+    const T* x = in + row * sk;
+    T* y = out + row * sk;
+    const uint8_t* mrow = nullptr;
+    long limit = sk;
+    if (MODE == 1) {
+      const long q = row % sq;
+      const long b = row / (np * sq);
+      mrow = mask + (b * sq + q) * sk;
+    } else if (MODE == 2) {
+      limit = (row % sq) + 1;
+    }
+
+    OnlineSM o;
+    const long i0 = (long)lane * W;
+    VecPack<T> v;
+    const bool active = i0 < sk;
+    if (active) {
+      load_pack(v, x + i0);
+#pragma unroll
+      for (int j = 0; j < W; ++j) {
+        if (MODE == 2 && i0 + j >= limit) break;
+        float f = to_float(v.a[j]) * scale;
+        if (MODE == 1 && mrow[i0 + j]) f = -10000.0f;
+        o.add(f);
+      }
+    }
+#pragma unroll
+    for (int off = WAVE_SIZE / 2; off > 0; off >>= 1) {
+      o.combine(__shfl_xor(o.m, off), __shfl_xor(o.s, off));
+    }
+    const float inv_s = o.s > 0.f ? 1.f / o.s : 0.f;
+    if (active) {
+      VecPack<T> r;
+#pragma unroll
+      for (int j = 0; j < W; ++j) {
+        float f = 0.f;
+        if (MODE != 2 || i0 + j < limit) {
+          float vv = to_float(v.a[j]) * scale;
+          if (MODE == 1 && mrow[i0 + j]) vv = -10000.0f;
+          f = __expf(vv - o.m) * inv_s;
+        }
+        r.a[j] = from_float<T>(f);
+      }
+      store_pack(y + i0, r);
+    }
+  }
+}
+
+template <typename T>
+__global__ void __launch_bounds__(SM_BLOCK) softmax_bwd_wave_kernel(
+    const T* __restrict__ dy_ptr, const T* __restrict__ y_ptr, T* __restrict__ dx_ptr,
+    float scale, long rows, long sk) {
+  constexpr int W = VecPack<T>::width;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const int waves_per_block = SM_BLOCK / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * waves_per_block + wid; row < rows;
+       row += (long)gridDim.x * waves_per_block) {
+    const T* dy = dy_ptr + row * sk;
+    const T* y = y_ptr + row * sk;
+    T* dx = dx_ptr + row * sk;
+    const long i0 = (long)lane * W;
+    const bool active = i0 < sk;
+    VecPack<T> vd, vy;
+    float acc = 0.f;
+    if (active) {
+      load_pack(vd, dy + i0);
+      load_pack(vy, y + i0);
+#pragma unroll
+      for (int j = 0; j < W; ++j) acc = fmaf(to_float(vd.a[j]), to_float(vy.a[j]), acc);
+    }
+    float dot = wave_reduce_sum(acc);
+    if (active) {
+      VecPack<T> r;
+#pragma unroll
+      for (int j = 0; j < W; ++j) {
+        r.a[j] = from_float<T>(to_float(vy.a[j]) * (to_float(vd.a[j]) - dot) * scale);
+      }
+      store_pack(dx + i0, r);
+    }
+  }
+}
+
 // grad = y * (dy - sum(dy*y)) * scale
 template <typename T, bool VEC>
 __global__ void __launch_bounds__(SM_BLOCK) softmax_bwd_kernel(
@@ -195,7 +291,15 @@ at::Tensor fwd_impl(const at::Tensor& input, const c10::optional<at::Tensor>& ma
   if (MODE == 1) m8 = mask->to(at::kByte).contiguous();
   APEX_DISPATCH_FLOAT_HALF_BF(x.scalar_type(), "scaled_softmax_forward", ([&] {
     const bool vec = (sk % VecPack<scalar_t>::width == 0) && is_pack_aligned<scalar_t>(x.data_ptr());
-    if (vec) {
+    if (vec && sk <= WAVE_SIZE * VecPack<scalar_t>::width) {
+      // short rows: one wave64 per row, zero barriers
+      const int wpb = SM_BLOCK / WAVE_SIZE;
+      const int grid = (int)std::min<long>((rows + wpb - 1) / wpb, 32768);
+      hipLaunchKernelGGL((softmax_fwd_wave_kernel<scalar_t, MODE>), dim3(grid), dim3(SM_BLOCK),
+                         0, current_stream(), (const scalar_t*)x.data_ptr(),
+                         (scalar_t*)y.data_ptr(), MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr,
+                         (float)scale, rows, sk, np, sq);
+    } else if (vec) {
       hipLaunchKernelGGL((softmax_fwd_kernel<scalar_t, MODE, true>), dim3(sm_grid(rows)),
                          dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)x.data_ptr(),
                          (scalar_t*)y.data_ptr(), MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr,
@@ -221,7 +325,14 @@ at::Tensor bwd_impl(const at::Tensor& grad_out, const at::Tensor& softmax_out, d
     const bool vec = (sk % VecPack<scalar_t>::width == 0) &&
                      is_pack_aligned<scalar_t>(y.data_ptr()) &&
                      is_pack_aligned<scalar_t>(dy.data_ptr());
-    if (vec) {
+    if (vec && sk <= WAVE_SIZE * VecPack<scalar_t>::width) {
+      const int wpb = SM_BLOCK / WAVE_SIZE;
+      const int grid = (int)std::min<long>((rows + wpb - 1) / wpb, 32768);
+      hipLaunchKernelGGL((softmax_bwd_wave_kernel<scalar_t>), dim3(grid), dim3(SM_BLOCK), 0,
+                         current_stream(), (const scalar_t*)dy.data_ptr(),
+                         (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
+                         rows, sk);
+    } else if (vec) {
       hipLaunchKernelGGL((softmax_bwd_kernel<scalar_t, true>), dim3(sm_grid(rows)),
                          dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)dy.data_ptr(),
                          (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
