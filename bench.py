@@ -71,6 +71,9 @@ def main():
         torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank)
     torch.cuda.set_device(device)
+    # let MIOpen pick tuned conv solutions (throughput bench, not the
+    # determinism harness — that one sets deterministic mode instead)
+    torch.backends.cudnn.benchmark = True
 
     from apex_amd import amp
     from apex_amd._ext import has_ext
