@@ -71,6 +71,154 @@ __device__ __forceinline__ bool row_vec_ok(const T* base, long n2) {
   return ((reinterpret_cast<uintptr_t>(base) & (sizeof(T) * 4 - 1)) == 0) && ((n2 & 3) == 0);
 }
 
+// ---------------- wave-per-row forward (narrow rows) ----------------
+// One wave64 per row, the whole row held in registers (<= NPACK 16-byte
+// packs per lane): single global read, zero barriers. Covers the common
+// transformer hidden sizes (bf16: n2 <= 2048 at NPACK=4).
+template <typename T, typename WT, bool RMS, bool AFFINE, int NPACK>
+__global__ void __launch_bounds__(LN_BLOCK) ln_fwd_wave_kernel(
+    const T* __restrict__ input, T* __restrict__ output, float* __restrict__ mean_out,
+    float* __restrict__ invvar_out, const WT* __restrict__ gamma, const WT* __restrict__ beta,
+    long n1, long n2, float eps) {
+  constexpr int W = VecPack<T>::width;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * WPB + wid; row < n1; row += (long)gridDim.x * WPB) {
+    const T* x = input + row * n2;
+    T* y = output + row * n2;
+    VecPack<T> xs[NPACK];
+    Welford w;
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        load_pack(xs[k], x + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) w.add(to_float(xs[k].a[j]));
+      }
+    }
+#pragma unroll
+    for (int off = WAVE_SIZE / 2; off > 0; off >>= 1) {
+      float mb = __shfl_xor(w.mean, off);
+      float m2b = __shfl_xor(w.m2, off);
+      float nb = __shfl_xor(w.count, off);
+      w.combine(mb, m2b, nb);
+    }
+    const float mean = RMS ? 0.f : w.mean;
+    // for RMS, m2 accumulated via Welford gives sum((x-mean)^2); recompute
+    // sum(x^2) = m2 + n*mean^2
+    const float sumsq = w.m2 + w.count * w.mean * w.mean;
+    const float invvar = RMS ? rsqrtf(sumsq / n2 + eps) : rsqrtf(w.m2 / n2 + eps);
+    if (lane == 0) {
+      if (!RMS && mean_out) mean_out[row] = mean;
+      invvar_out[row] = invvar;
+    }
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        VecPack<T> o;
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float xhat = (to_float(xs[k].a[j]) - mean) * invvar;
+          float r = xhat;
+          if (AFFINE) {
+            r = xhat * to_float(gamma[i + j]);
+            if (!RMS) r += to_float(beta[i + j]);
+          }
+          o.a[j] = from_float<T>(r);
+        }
+        store_pack(y + i, o);
+      }
+    }
+  }
+}
+
+// wave-per-row backward dx: dy and x(/y) rows in registers, two wave sums.
+template <typename T, typename WT, bool RMS, bool AFFINE, bool MEMEFF, int NPACK>
+__global__ void __launch_bounds__(LN_BLOCK) ln_bwd_dx_wave_kernel(
+    const T* __restrict__ dy_ptr, const T* __restrict__ io, const float* __restrict__ mean_ptr,
+    const float* __restrict__ invvar_ptr, const WT* __restrict__ gamma,
+    const WT* __restrict__ beta, T* __restrict__ dx_ptr, long n1, long n2) {
+  constexpr int W = VecPack<T>::width;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * WPB + wid; row < n1; row += (long)gridDim.x * WPB) {
+    const T* dy = dy_ptr + row * n2;
+    const T* xr = io + row * n2;
+    T* dx = dx_ptr + row * n2;
+    const float invvar = invvar_ptr[row];
+    const float mean = (RMS || MEMEFF) ? 0.f : mean_ptr[row];
+
+    VecPack<T> vd[NPACK], vx[NPACK];
+    float a1 = 0.f, a2 = 0.f;
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        load_pack(vd[k], dy + i);
+        load_pack(vx[k], xr + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float g = AFFINE ? to_float(gamma[i + j]) : 1.f;
+          float dyf = to_float(vd[k].a[j]) * g;
+          float xhat;
+          if (MEMEFF) {
+            float yv = to_float(vx[k].a[j]);
+            if (AFFINE) {
+              float gv = to_float(gamma[i + j]);
+              xhat = RMS ? yv / gv : (yv - to_float(beta[i + j])) / gv;
+            } else {
+              xhat = yv;
+            }
+          } else {
+            xhat = (to_float(vx[k].a[j]) - mean) * invvar;
+          }
+          a1 = fmaf(dyf, xhat, a1);
+          a2 += dyf;
+        }
+      }
+    }
+    float s1 = wave_reduce_sum(a1);
+    float s2 = RMS ? 0.f : wave_reduce_sum(a2);
+    const float inv_n = 1.f / (float)n2;
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        VecPack<T> o;
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float g = AFFINE ? to_float(gamma[i + j]) : 1.f;
+          float dyf = to_float(vd[k].a[j]) * g;
+          float xhat;
+          if (MEMEFF) {
+            float yv = to_float(vx[k].a[j]);
+            if (AFFINE) {
+              float gv = to_float(gamma[i + j]);
+              xhat = RMS ? yv / gv : (yv - to_float(beta[i + j])) / gv;
+            } else {
+              xhat = yv;
+            }
+          } else {
+            xhat = (to_float(vx[k].a[j]) - mean) * invvar;
+          }
+          float r;
+          if (RMS) {
+            r = invvar * (dyf - xhat * s1 * inv_n);
+          } else {
+            r = invvar * (dyf - s2 * inv_n - xhat * s1 * inv_n);
+          }
+          o.a[j] = from_float<T>(r);
+        }
+        store_pack(dx + i, o);
+      }
+    }
+  }
+}
+
 // ---------------- forward ----------------
 // RMS: no mean; affine: gamma (and beta for LN) applied.
 template <typename T, typename WT, bool RMS, bool AFFINE>
@@ -318,18 +466,47 @@ std::vector<at::Tensor> norm_fwd(const at::Tensor& input,
     const auto w_type = affine ? gamma->scalar_type() : in.scalar_type();
     APEX_DISPATCH_FLOAT_HALF_BF(w_type, "fused_norm_fwd", ([&] {
       using w_t = scalar_t;
-      if (affine) {
-        hipLaunchKernelGGL((ln_fwd_kernel<in_t, w_t, RMS, true>), dim3(fwd_grid(n1)),
+      constexpr int W = VecPack<in_t>::width;
+      const bool wave_ok = (n2 % W == 0) && is_pack_aligned<in_t>(in.data_ptr()) &&
+                           n2 <= (long)WAVE_SIZE * W * 4;
+      const int npack = wave_ok ? (int)((n2 + WAVE_SIZE * W - 1) / (WAVE_SIZE * W)) : 0;
+      const w_t* g_ptr = affine ? (const w_t*)gamma->data_ptr() : nullptr;
+      const w_t* b_ptr = (affine && !RMS) ? (const w_t*)beta->data_ptr() : nullptr;
+      float* mean_ptr = RMS ? nullptr : mean.data_ptr<float>();
+
+      auto launch_wave = [&](auto aff, auto np_tag) {
+        constexpr bool AFF = decltype(aff)::value;
+        constexpr int NP = decltype(np_tag)::value;
+        constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+        const int grid = (int)std::min<long>((n1 + WPB - 1) / WPB, 32768);
+        hipLaunchKernelGGL((ln_fwd_wave_kernel<in_t, w_t, RMS, AFF, NP>), dim3(grid),
                            dim3(LN_BLOCK), 0, stream, (const in_t*)in.data_ptr(),
-                           (in_t*)out.data_ptr(), RMS ? nullptr : mean.data_ptr<float>(),
-                           invvar.data_ptr<float>(), (const w_t*)gamma->data_ptr(),
-                           RMS ? nullptr : (const w_t*)beta->data_ptr(), n1, n2, (float)eps);
+                           (in_t*)out.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr,
+                           b_ptr, n1, n2, (float)eps);
+      };
+      auto launch_block = [&](auto aff) {
+        constexpr bool AFF = decltype(aff)::value;
+        hipLaunchKernelGGL((ln_fwd_kernel<in_t, w_t, RMS, AFF>), dim3(fwd_grid(n1)),
+                           dim3(LN_BLOCK), 0, stream, (const in_t*)in.data_ptr(),
+                           (in_t*)out.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr,
+                           b_ptr, n1, n2, (float)eps);
+      };
+      using Tt = std::true_type;
+      using Ff = std::false_type;
+      if (wave_ok) {
+        auto dispatch_np = [&](auto aff) {
+          switch (npack) {
+            case 1: launch_wave(aff, std::integral_constant<int, 1>{}); break;
+            case 2: launch_wave(aff, std::integral_constant<int, 2>{}); break;
+            case 3: launch_wave(aff, std::integral_constant<int, 3>{}); break;
+            default: launch_wave(aff, std::integral_constant<int, 4>{}); break;
+          }
+        };
+        if (affine) dispatch_np(Tt{});
+        else dispatch_np(Ff{});
       } else {
-        hipLaunchKernelGGL((ln_fwd_kernel<in_t, w_t, RMS, false>), dim3(fwd_grid(n1)),
-                           dim3(LN_BLOCK), 0, stream, (const in_t*)in.data_ptr(),
-                           (in_t*)out.data_ptr(), RMS ? nullptr : mean.data_ptr<float>(),
-                           invvar.data_ptr<float>(), (const w_t*)nullptr, (const w_t*)nullptr,
-                           n1, n2, (float)eps);
+        if (affine) launch_block(Tt{});
+        else launch_block(Ff{});
       }
       HIP_CHECK(hipGetLastError());
     }()));
@@ -344,10 +521,32 @@ void launch_bwd_impl(const at::Tensor& dy, const at::Tensor& io, const float* me
                      at::Tensor& dx, at::Tensor& part_gw, at::Tensor& part_gb,
                      at::Tensor& grad_gamma, at::Tensor& grad_beta, long tiles, long n1, long n2,
                      bool affine, hipStream_t stream) {
-  hipLaunchKernelGGL((ln_bwd_dx_kernel<in_t, w_t, RMS, AFF, MEFF>), dim3(fwd_grid(n1)),
-                     dim3(LN_BLOCK), 0, stream, (const in_t*)dy.data_ptr(),
-                     (const in_t*)io.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr, b_ptr,
-                     (in_t*)dx.data_ptr(), n1, n2);
+  constexpr int W = VecPack<in_t>::width;
+  const bool wave_ok = (n2 % W == 0) && is_pack_aligned<in_t>(dy.data_ptr()) &&
+                       is_pack_aligned<in_t>(io.data_ptr()) && n2 <= (long)WAVE_SIZE * W * 4;
+  if (wave_ok) {
+    const int npack = (int)((n2 + WAVE_SIZE * W - 1) / (WAVE_SIZE * W));
+    constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+    const int grid = (int)std::min<long>((n1 + WPB - 1) / WPB, 32768);
+    auto lw = [&](auto np_tag) {
+      constexpr int NP = decltype(np_tag)::value;
+      hipLaunchKernelGGL((ln_bwd_dx_wave_kernel<in_t, w_t, RMS, AFF, MEFF, NP>), dim3(grid),
+                         dim3(LN_BLOCK), 0, stream, (const in_t*)dy.data_ptr(),
+                         (const in_t*)io.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr,
+                         b_ptr, (in_t*)dx.data_ptr(), n1, n2);
+    };
+    switch (npack) {
+      case 1: lw(std::integral_constant<int, 1>{}); break;
+      case 2: lw(std::integral_constant<int, 2>{}); break;
+      case 3: lw(std::integral_constant<int, 3>{}); break;
+      default: lw(std::integral_constant<int, 4>{}); break;
+    }
+  } else {
+    hipLaunchKernelGGL((ln_bwd_dx_kernel<in_t, w_t, RMS, AFF, MEFF>), dim3(fwd_grid(n1)),
+                       dim3(LN_BLOCK), 0, stream, (const in_t*)dy.data_ptr(),
+                       (const in_t*)io.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr,
+                       b_ptr, (in_t*)dx.data_ptr(), n1, n2);
+  }
   HIP_CHECK(hipGetLastError());
   if (affine) {
     dim3 pgrid((uint32_t)std::min<long>((n2 + LN_BLOCK - 1) / LN_BLOCK, 1024), (uint32_t)tiles);
