@@ -59,21 +59,28 @@ def main():
                     choices=["resnet50", "bert", "gpt2"],
                     help="resnet50 = headline config #2; bert/gpt2 = BASELINE configs #3/#4")
     ap.add_argument("--seq-len", type=int, default=512)
+    ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
+                    help="cpu + gloo is a debug mode for validating the distributed flow")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
+    use_cpu = args.device == "cpu"
     distributed = world > 1
     if distributed:
-        dist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
-    torch.cuda.set_device(device)
-    # let MIOpen pick tuned conv solutions (throughput bench, not the
-    # determinism harness — that one sets deterministic mode instead)
-    torch.backends.cudnn.benchmark = True
+        dist.init_process_group(backend="gloo" if use_cpu else "nccl")
+        if not use_cpu:
+            torch.cuda.set_device(local_rank)
+    if use_cpu:
+        device = torch.device("cpu")
+    else:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+        # let MIOpen pick tuned conv solutions (throughput bench, not the
+        # determinism harness — that one sets deterministic mode instead)
+        torch.backends.cudnn.benchmark = True
 
     from apex_amd import amp
     from apex_amd._ext import has_ext
@@ -84,7 +91,7 @@ def main():
 
     # --- part 1: FusedAdam 350M step time (rank 0, 1 GPU) ---
     adam_ms = None
-    if rank == 0 and not args.skip_adam_bench:
+    if rank == 0 and not args.skip_adam_bench and not use_cpu:
         adam_ms = bench_fused_adam_350m(device, steps=max(10, args.steps // 2), warmup=args.warmup)
 
     if distributed:
@@ -167,11 +174,13 @@ def main():
 
     if distributed:
         dist.barrier()
-    torch.cuda.synchronize()
+    if not use_cpu:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
-    torch.cuda.synchronize()
+    if not use_cpu:
+        torch.cuda.synchronize()
     if distributed:
         dist.barrier()
     elapsed = time.perf_counter() - t0

@@ -1,0 +1,34 @@
+"""End-to-end bench.py flow validation on CPU/gloo (2 ranks via torchrun) —
+de-risks the driver's multi-GPU invocation: rendezvous, DDP wrap, barriers,
+MAX-over-ranks, single JSON line from rank 0."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(600)
+def test_bench_cpu_gloo_two_ranks():
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1", "--master-port", "29611",
+        "bench.py", "--device", "cpu", "--batch", "2", "--image-size", "64",
+        "--steps", "2", "--warmup", "1", "--skip-adam-bench",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=560)
+    assert out.returncode == 0, f"bench failed:\n{out.stdout[-2000:]}\n{out.stderr[-2000:]}"
+    json_lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1, f"expected exactly one JSON line, got: {json_lines}"
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["value"] > 0
+    assert rec["config"]["parallelism"] == "dp2"
+    for key in ["metric", "unit", "steps", "warmup", "ms_per_step", "higher_is_better",
+                "scaling", "vs_baseline", "dtype", "data", "config"]:
+        assert key in rec
