@@ -89,7 +89,13 @@ void multi_tensor_apply(long chunk_size, const at::Tensor& noop_flag,
     while (t < ntensors && nt < max_t) {
       const long numel = tensor_lists[0][t].numel();
       for (int d = 0; d < depth; ++d) {
-        TORCH_CHECK(tensor_lists[d][t].is_contiguous(), "multi_tensor_apply: tensor not contiguous");
+        // any dense non-overlapping layout is fine for elementwise work
+        // (e.g. channels_last conv params) as long as every tensor in the
+        // tuple shares the SAME element order
+        TORCH_CHECK(tensor_lists[d][t].is_non_overlapping_and_dense(),
+                    "multi_tensor_apply: tensor not dense/contiguous");
+        TORCH_CHECK(numel <= 1 || tensor_lists[d][t].strides() == tensor_lists[0][t].strides(),
+                    "multi_tensor_apply: layout mismatch within tensor tuple");
         TORCH_CHECK(tensor_lists[d][t].numel() == numel, "multi_tensor_apply: size mismatch within tuple");
         meta.addrs[d][nt] = tensor_lists[d][t].data_ptr();
       }
