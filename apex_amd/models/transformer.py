@@ -138,3 +138,43 @@ class BertModel(_TransformerLM):
 class GPTModel(_TransformerLM):
     def __init__(self, cfg=None):
         super().__init__(cfg or gpt2_345m_config())
+
+
+def transformer_large_config(seq_len=512):
+    return TransformerLMConfig(vocab_size=32768, hidden=1024, layers=24, heads=16,
+                               seq_len=seq_len, causal=True, norm="layernorm")
+
+
+class TransformerLargeModel(nn.Module):
+    """Transformer-large LM built on contrib.fast_multihead_attn (BASELINE
+    config #5: fast-MHA equivalent + contrib.xentropy)."""
+
+    def __init__(self, cfg=None):
+        super().__init__()
+        from ..contrib.fast_multihead_attn import SelfMultiheadAttn
+
+        cfg = cfg or transformer_large_config()
+        self.cfg = cfg
+        norm_cls = FusedLayerNorm
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.pos_emb = nn.Embedding(cfg.seq_len, cfg.hidden)
+        self.attns = nn.ModuleList([
+            SelfMultiheadAttn(cfg.hidden, cfg.heads, dropout=0.0) for _ in range(cfg.layers)
+        ])
+        self.ln1 = nn.ModuleList([norm_cls(cfg.hidden) for _ in range(cfg.layers)])
+        self.ln2 = nn.ModuleList([norm_cls(cfg.hidden) for _ in range(cfg.layers)])
+        self.mlps = nn.ModuleList([FusedMLPBlock(cfg) for _ in range(cfg.layers)])
+        self.final_norm = norm_cls(cfg.hidden)
+        nn.init.normal_(self.tok_emb.weight, std=0.02)
+        nn.init.normal_(self.pos_emb.weight, std=0.02)
+
+    def forward(self, tokens):
+        b, s = tokens.shape
+        pos = torch.arange(s, device=tokens.device).unsqueeze(0)
+        x = (self.tok_emb(tokens) + self.pos_emb(pos)).transpose(0, 1)  # [s, b, h]
+        for attn, l1, l2, mlp in zip(self.attns, self.ln1, self.ln2, self.mlps):
+            a, _ = attn(l1(x), attn_mask="causal")
+            x = x + a
+            x = x + mlp(l2(x))
+        x = self.final_norm(x).transpose(0, 1)
+        return torch.matmul(x, self.tok_emb.weight.t())

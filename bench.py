@@ -56,7 +56,7 @@ def main():
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--skip-adam-bench", action="store_true")
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "bert", "gpt2"],
+                    choices=["resnet50", "bert", "gpt2", "transformer_lg"],
                     help="resnet50 = headline config #2; bert/gpt2 = BASELINE configs #3/#4")
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
@@ -124,7 +124,8 @@ def main():
         # scaled_masked_softmax) / GPT-2 345M (fused_dense GEMM+bias+GELU +
         # FusedRMSNorm + FusedLAMB + causal softmax), token throughput.
         from apex_amd.models.transformer import (
-            BertModel, GPTModel, bert_base_config, gpt2_345m_config,
+            BertModel, GPTModel, TransformerLargeModel,
+            bert_base_config, gpt2_345m_config, transformer_large_config,
         )
         from apex_amd.optimizers import FusedAdam, FusedLAMB
         from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
@@ -134,6 +135,11 @@ def main():
             model = BertModel(cfg).to(device)
             opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
             config_model = "bert-base"
+        elif args.model == "transformer_lg":
+            cfg = transformer_large_config(seq_len=args.seq_len)
+            model = TransformerLargeModel(cfg).to(device)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            config_model = "transformer-large(fast_multihead_attn)"
         else:
             cfg = gpt2_345m_config(seq_len=min(args.seq_len, 1024))
             model = GPTModel(cfg).to(device)
@@ -144,7 +150,7 @@ def main():
                                     keep_batchnorm_fp32=False, verbosity=0)
         tokens = torch.randint(0, cfg.vocab_size, (args.batch, cfg.seq_len), device=device)
         mask = None
-        if not cfg.causal:
+        if not cfg.causal and args.model == "bert":
             mask = torch.zeros(args.batch, 1, cfg.seq_len, cfg.seq_len, dtype=torch.bool,
                                device=device)
 
@@ -216,7 +222,8 @@ def main():
                 "seq_len": None if args.model == "resnet50" else args.seq_len,
                 "amp": "O1-bf16" if args.model == "resnet50" else "O2-bf16",
                 "optimizer": {"resnet50": "FusedSGD(momentum=0.9)", "bert": "FusedAdam",
-                              "gpt2": "FusedLAMB"}[args.model],
+                              "gpt2": "FusedLAMB",
+                              "transformer_lg": "FusedAdam"}[args.model],
                 "syncbn": use_syncbn,
                 "parallelism": f"dp{world}",
             },

@@ -93,6 +93,7 @@ setup(
         "apex_amd.contrib.gbn",
         "apex_amd.contrib.openfold",
         "apex_amd.contrib.torchsched",
+        "apex_amd.contrib.fast_multihead_attn",
         "apex_amd.fused_dense",
         "apex_amd.mlp",
         "apex_amd.models",
