@@ -135,10 +135,10 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_kernel(
   }
 }
 
-// Wave-per-row variant for short rows (sk <= 64*W): each wave64 owns one
-// row — no LDS, no barriers, one vector pass for the online stats and one
-// for the write (4 rows per 256-thread workgroup).
-template <typename T, int MODE>
+// Wave-per-row variant for short rows (sk <= 64*W*NPACK): each wave64 owns
+// one row held in registers — no LDS, no barriers, one global read per
+// element (4 rows per 256-thread workgroup).
+template <typename T, int MODE, int NPACK>
 __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_wave_kernel(
     const T* __restrict__ in, T* __restrict__ out, const uint8_t* __restrict__ mask, float scale,
     long rows, long sk, long np, long sq) {
@@ -161,17 +161,19 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_wave_kernel(
     }
 
     OnlineSM o;
-    const long i0 = (long)lane * W;
-    VecPack<T> v;
-    const bool active = i0 < sk;
-    if (active) {
-      load_pack(v, x + i0);
+    VecPack<T> v[NPACK];
 #pragma unroll
-      for (int j = 0; j < W; ++j) {
-        if (MODE == 2 && i0 + j >= limit) break;
-        float f = to_float(v.a[j]) * scale;
-        if (MODE == 1 && mrow[i0 + j]) f = -10000.0f;
-        o.add(f);
+    for (int k = 0; k < NPACK; ++k) {
+      const long i0 = (long)(k * WAVE_SIZE + lane) * W;
+      if (i0 < sk) {
+        load_pack(v[k], x + i0);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          if (MODE == 2 && i0 + j >= limit) break;
+          float f = to_float(v[k].a[j]) * scale;
+          if (MODE == 1 && mrow[i0 + j]) f = -10000.0f;
+          o.add(f);
+        }
       }
     }
 #pragma unroll
@@ -179,24 +181,28 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_fwd_wave_kernel(
       o.combine(__shfl_xor(o.m, off), __shfl_xor(o.s, off));
     }
     const float inv_s = o.s > 0.f ? 1.f / o.s : 0.f;
-    if (active) {
-      VecPack<T> r;
 #pragma unroll
-      for (int j = 0; j < W; ++j) {
-        float f = 0.f;
-        if (MODE != 2 || i0 + j < limit) {
-          float vv = to_float(v.a[j]) * scale;
-          if (MODE == 1 && mrow[i0 + j]) vv = -10000.0f;
-          f = __expf(vv - o.m) * inv_s;
+    for (int k = 0; k < NPACK; ++k) {
+      const long i0 = (long)(k * WAVE_SIZE + lane) * W;
+      if (i0 < sk) {
+        VecPack<T> r;
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float f = 0.f;
+          if (MODE != 2 || i0 + j < limit) {
+            float vv = to_float(v[k].a[j]) * scale;
+            if (MODE == 1 && mrow[i0 + j]) vv = -10000.0f;
+            f = __expf(vv - o.m) * inv_s;
+          }
+          r.a[j] = from_float<T>(f);
         }
-        r.a[j] = from_float<T>(f);
+        store_pack(y + i0, r);
       }
-      store_pack(y + i0, r);
     }
   }
 }
 
-template <typename T>
+template <typename T, int NPACK>
 __global__ void __launch_bounds__(SM_BLOCK) softmax_bwd_wave_kernel(
     const T* __restrict__ dy_ptr, const T* __restrict__ y_ptr, T* __restrict__ dx_ptr,
     float scale, long rows, long sk) {
@@ -209,24 +215,30 @@ __global__ void __launch_bounds__(SM_BLOCK) softmax_bwd_wave_kernel(
     const T* dy = dy_ptr + row * sk;
     const T* y = y_ptr + row * sk;
     T* dx = dx_ptr + row * sk;
-    const long i0 = (long)lane * W;
-    const bool active = i0 < sk;
-    VecPack<T> vd, vy;
+    VecPack<T> vd[NPACK], vy[NPACK];
     float acc = 0.f;
-    if (active) {
-      load_pack(vd, dy + i0);
-      load_pack(vy, y + i0);
 #pragma unroll
-      for (int j = 0; j < W; ++j) acc = fmaf(to_float(vd.a[j]), to_float(vy.a[j]), acc);
+    for (int k = 0; k < NPACK; ++k) {
+      const long i0 = (long)(k * WAVE_SIZE + lane) * W;
+      if (i0 < sk) {
+        load_pack(vd[k], dy + i0);
+        load_pack(vy[k], y + i0);
+#pragma unroll
+        for (int j = 0; j < W; ++j) acc = fmaf(to_float(vd[k].a[j]), to_float(vy[k].a[j]), acc);
+      }
     }
     float dot = wave_reduce_sum(acc);
-    if (active) {
-      VecPack<T> r;
 #pragma unroll
-      for (int j = 0; j < W; ++j) {
-        r.a[j] = from_float<T>(to_float(vy.a[j]) * (to_float(vd.a[j]) - dot) * scale);
+    for (int k = 0; k < NPACK; ++k) {
+      const long i0 = (long)(k * WAVE_SIZE + lane) * W;
+      if (i0 < sk) {
+        VecPack<T> r;
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          r.a[j] = from_float<T>(to_float(vy[k].a[j]) * (to_float(vd[k].a[j]) - dot) * scale);
+        }
+        store_pack(dx + i0, r);
       }
-      store_pack(dx + i0, r);
     }
   }
 }
@@ -289,15 +301,27 @@ at::Tensor fwd_impl(const at::Tensor& input, const c10::optional<at::Tensor>& ma
   at::Tensor m8;
   if (MODE == 1) m8 = mask->to(at::kByte).contiguous();
   APEX_DISPATCH_FLOAT_HALF_BF(x.scalar_type(), "scaled_softmax_forward", ([&] {
-    const bool vec = (sk % VecPack<scalar_t>::width == 0) && is_pack_aligned<scalar_t>(x.data_ptr());
-    if (vec && sk <= WAVE_SIZE * VecPack<scalar_t>::width) {
-      // short rows: one wave64 per row, zero barriers
+    constexpr int PW = 16 / sizeof(scalar_t);
+    const bool vec = (sk % PW == 0) && is_pack_aligned<scalar_t>(x.data_ptr());
+    const long wave_span = (long)WAVE_SIZE * PW;
+    if (vec && sk <= wave_span * 4) {
+      // short rows: one wave64 per row (row in registers), zero barriers
       const int wpb = SM_BLOCK / WAVE_SIZE;
       const int grid = (int)std::min<long>((rows + wpb - 1) / wpb, 32768);
-      hipLaunchKernelGGL((softmax_fwd_wave_kernel<scalar_t, MODE>), dim3(grid), dim3(SM_BLOCK),
-                         0, current_stream(), (const scalar_t*)x.data_ptr(),
-                         (scalar_t*)y.data_ptr(), MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr,
-                         (float)scale, rows, sk, np, sq);
+      const int npack = (int)((sk + wave_span - 1) / wave_span);
+      auto lw = [&](auto np_tag) {
+        hipLaunchKernelGGL((softmax_fwd_wave_kernel<scalar_t, MODE, decltype(np_tag)::value>),
+                           dim3(grid), dim3(SM_BLOCK), 0, current_stream(),
+                           (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+                           MODE == 1 ? m8.data_ptr<uint8_t>() : nullptr, (float)scale, rows, sk,
+                           np, sq);
+      };
+      switch (npack) {
+        case 1: lw(std::integral_constant<int, 1>{}); break;
+        case 2: lw(std::integral_constant<int, 2>{}); break;
+        case 3: lw(std::integral_constant<int, 3>{}); break;
+        default: lw(std::integral_constant<int, 4>{}); break;
+      }
     } else if (vec) {
       hipLaunchKernelGGL((softmax_fwd_kernel<scalar_t, MODE, true>), dim3(sm_grid(rows)),
                          dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)x.data_ptr(),
@@ -321,16 +345,24 @@ at::Tensor bwd_impl(const at::Tensor& grad_out, const at::Tensor& softmax_out, d
   const long sk = y.size(-1);
   const long rows = y.numel() / sk;
   APEX_DISPATCH_FLOAT_HALF_BF(y.scalar_type(), "scaled_softmax_backward", ([&] {
-    const bool vec = (sk % VecPack<scalar_t>::width == 0) &&
+    constexpr int PW = 16 / sizeof(scalar_t);
+    const bool vec = (sk % PW == 0) &&
                      is_pack_aligned<scalar_t>(y.data_ptr()) &&
                      is_pack_aligned<scalar_t>(dy.data_ptr());
-    if (vec && sk <= WAVE_SIZE * VecPack<scalar_t>::width) {
+    const long wave_span = (long)WAVE_SIZE * PW;
+    if (vec && sk <= wave_span * 2) {
+      // NPACK capped at 2 for bwd (2 operand rows live in registers)
       const int wpb = SM_BLOCK / WAVE_SIZE;
       const int grid = (int)std::min<long>((rows + wpb - 1) / wpb, 32768);
-      hipLaunchKernelGGL((softmax_bwd_wave_kernel<scalar_t>), dim3(grid), dim3(SM_BLOCK), 0,
-                         current_stream(), (const scalar_t*)dy.data_ptr(),
-                         (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(), (float)scale,
-                         rows, sk);
+      const int npack = (int)((sk + wave_span - 1) / wave_span);
+      auto lw = [&](auto np_tag) {
+        hipLaunchKernelGGL((softmax_bwd_wave_kernel<scalar_t, decltype(np_tag)::value>),
+                           dim3(grid), dim3(SM_BLOCK), 0, current_stream(),
+                           (const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+                           (scalar_t*)dx.data_ptr(), (float)scale, rows, sk);
+      };
+      if (npack == 1) lw(std::integral_constant<int, 1>{});
+      else lw(std::integral_constant<int, 2>{});
     } else if (vec) {
       hipLaunchKernelGGL((softmax_bwd_kernel<scalar_t, true>), dim3(sm_grid(rows)),
                          dim3(SM_BLOCK), 0, current_stream(), (const scalar_t*)dy.data_ptr(),
