@@ -5,6 +5,11 @@ examples/dcgan/main_amp.py:214-253 — the surface that pins
 
 import argparse
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", ".."))
+
 import torch
 import torch.nn as nn
 

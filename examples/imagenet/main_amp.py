@@ -14,6 +14,11 @@ import json
 import os
 import time
 
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", ".."))
+
 import torch
 import torch.distributed as dist
 
