@@ -1,0 +1,71 @@
+"""Verify MFMA fragment layouts + hand-written GEMM correctness and speed."""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+import apex_amd._mfma as mfma
+
+
+def main():
+    torch.manual_seed(0)
+    # 1. single-tile layout verification (asymmetric random inputs, G9)
+    A = torch.randn(16, 32, device="cuda", dtype=torch.bfloat16)
+    B = torch.randn(32, 16, device="cuda", dtype=torch.bfloat16)
+    D = mfma.mfma_tile_probe(A.view(torch.int16), B.view(torch.int16))
+    ref = A.float() @ B.float()
+    err = (D - ref).abs().max().item()
+    print(f"tile probe max err: {err:.5f}  ({'OK' if err < 0.1 else 'LAYOUT WRONG'})")
+
+    # 2. full GEMM refcheck: gelu(X @ W^T + b)
+    M, N, K = 512, 256, 128
+    X = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    W = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    b = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    out, gi = mfma.gemm_bias_gelu(X, W, b, True)
+    z = X.float() @ W.float().t() + b.float()
+    ref_out = torch.nn.functional.gelu(z, approximate="tanh")
+    e1 = (gi.float() - z).abs().max().item()
+    e2 = (out.float() - ref_out).abs().max().item()
+    print(f"gemm refcheck: |gelu_in - z| {e1:.4f}  |out - ref| {e2:.4f}")
+
+    # 3. perf at the BERT mlp1 shape vs the hipBLASLt split path
+    import apex_amd._fused_dense as fd
+
+    M, N, K = 16384, 3072, 768
+    X = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    W = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.02
+    b = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    W2 = torch.randn(768, N, device="cuda", dtype=torch.bfloat16)
+    b2 = torch.randn(768, device="cuda", dtype=torch.bfloat16)
+
+    def t(fn, iters=30):
+        for _ in range(5):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / iters * 1e3
+
+    tm = t(lambda: mfma.gemm_bias_gelu(X, W, b, True))
+    tl = t(lambda: fd.linear_gelu_linear_forward(X, W, b, W2, b2))  # incl. 2nd GEMM
+    t1 = t(lambda: mfma.gemm_bias(X, W, b))
+    flops = 2.0 * M * N * K
+    print(f"mfma gemm_bias_gelu {tm:.3f} ms ({flops/tm/1e9:.0f} TF)  "
+          f"mfma gemm_bias {t1:.3f} ms ({flops/t1/1e9:.0f} TF)")
+    print(f"hipBLASLt linear_gelu_linear (both GEMMs) {tl:.3f} ms")
+    # correctness at the big shape too
+    out, gi = mfma.gemm_bias_gelu(X, W, b, True)
+    z = (X.float() @ W.float().t() + b.float())
+    rel = (gi.float() - z).abs().max().item() / z.abs().max().item()
+    print(f"big-shape gelu_in rel err: {rel:.5f}")
+
+
+if __name__ == "__main__":
+    main()
