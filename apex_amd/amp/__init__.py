@@ -309,8 +309,8 @@ def initialize(
         for opt in optimizer_list:
             _process_optimizer_o2(opt, props["cast_model_type"])
             if type(opt).__name__ == "FusedSGD":
-                opt._amp_handles_param_copy = getattr(opt, "materialize_master_grads", True) is not None and True
-                # FusedSGD's 4-list kernel writes the fp16 copy itself
+                # FusedSGD's 4-list kernel writes the low-precision param
+                # copy itself — amp must not copy masters back after step
                 opt._amp_handles_param_copy = True
             _patch_step_for_skip_and_copy(opt, needs_master_copy=True)
     else:

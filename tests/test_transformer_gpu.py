@@ -68,12 +68,15 @@ def test_scaled_masked_softmax(dtype):
     torch.testing.assert_close(x.grad.float(), xr.grad, **tol_for(dtype))
 
 
-@pytest.mark.parametrize("dtype", [torch.float16, torch.bfloat16])
-def test_scaled_upper_triang_masked_softmax(dtype):
+@pytest.mark.parametrize("dtype,sq", [(torch.float16, 129), (torch.bfloat16, 129),
+                                      (torch.bfloat16, 512), (torch.bfloat16, 1024)])
+def test_scaled_upper_triang_masked_softmax(dtype, sq):
+    """sq=129 exercises the scalar block path; 512/1024 the wave-per-row
+    register path (NPACK 1 and 2)."""
     from apex_amd.transformer import scaled_upper_triang_masked_softmax
 
     torch.manual_seed(2)
-    ab, sq = 8, 129
+    ab = 4
     x = torch.randn(ab, sq, sq, device="cuda", dtype=dtype, requires_grad=True)
     xr = x.detach().float().clone().requires_grad_(True)
     scale = 0.5
