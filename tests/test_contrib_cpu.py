@@ -132,3 +132,23 @@ def test_groupbn_nhwc_cpu():
     y2.train()
     out = y2(x, z)
     assert (out >= 0).all()
+
+
+def test_permutation_search_improves_kept_magnitude():
+    from apex_amd.contrib.sparsity.permutation_search import (
+        efficacy, search_for_good_permutation, apply_permutation_in_place,
+    )
+
+    torch.manual_seed(7)
+    # adversarial: each 4-group has correlated magnitudes so permutation helps
+    w = torch.randn(32, 64)
+    w[:, ::4] *= 5.0  # big channels clustered into the same group positions
+    before = efficacy(w)
+    perm = search_for_good_permutation(w, max_iters=20)
+    after = efficacy(w[:, perm])
+    assert after >= before  # never worse
+    lin = torch.nn.Linear(64, 32)
+    with torch.no_grad():
+        lin.weight.copy_(w)
+    apply_permutation_in_place(lin, perm)
+    torch.testing.assert_close(lin.weight.detach(), w[:, perm])
