@@ -30,6 +30,14 @@ from ..._ext import get_ext
 from ...multi_tensor_apply import multi_tensor_applier
 
 
+class _null:
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *a):
+        return False
+
+
 def _backend_supports_rs(group):
     try:
         return dist.get_backend(group) == "nccl"
@@ -271,19 +279,35 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             else:
                 self._adam_ref(group, bias_correction, beta1, beta2, b)
 
-        # param sync: shard -> param bucket, then all_gather
+        # param sync: shard -> param bucket, then all_gather. On GPU the
+        # gathers run async on the comm stream with a stream-order dependency
+        # back to the main stream — the next forward's kernels queue behind
+        # them while the host races ahead (overlap_param_sync semantics
+        # without model hooks; per-bucket lazy waits land in a later round).
+        use_comm_stream = self._use_stream and self.world_size > 1
+        if use_comm_stream:
+            self._comm_stream.wait_stream(torch.cuda.current_stream())
         for b in self.buckets:
             lo = self.rank * b.shard_size
             shard = b.param_data[lo:lo + b.shard_size]
             shard.copy_(b.master_shard.to(b.param_data.dtype))
-            if self.world_size > 1:
-                if _backend_supports_rs(self.process_group):
-                    dist.all_gather_into_tensor(b.param_data, shard, group=self.process_group)
-                else:
-                    chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
-                    dist.all_gather(chunks, shard, group=self.process_group)
-                    for r, c in enumerate(chunks):
-                        b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
+        if self.world_size > 1:
+            ctx = torch.cuda.stream(self._comm_stream) if use_comm_stream else _null()
+            with ctx:
+                for b in self.buckets:
+                    lo = self.rank * b.shard_size
+                    shard = b.param_data[lo:lo + b.shard_size]
+                    if _backend_supports_rs(self.process_group):
+                        dist.all_gather_into_tensor(b.param_data, shard,
+                                                    group=self.process_group)
+                    else:
+                        chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                        dist.all_gather(chunks, shard, group=self.process_group)
+                        for r, c in enumerate(chunks):
+                            b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
+            if use_comm_stream:
+                torch.cuda.current_stream().wait_stream(self._comm_stream)
+        for b in self.buckets:
             # reset for the next iteration
             b.grad_data.zero_()
             b.ready_params.clear()
