@@ -3,7 +3,8 @@
 API parity with the reference ``apex.contrib.transducer``
 (apex/contrib/transducer/transducer.py: TransducerJoint:6, TransducerLoss:88).
 Round-1 scope notes (documented gaps, not silent fallbacks):
-- packed layouts (``pack_output`` / ``packed_input``) raise NotImplementedError;
+- packed layouts are supported (``pack_output`` with ``batch_offset`` =
+  inclusive cumsum of f_len*g_len, and ``packed_input`` for the loss);
 - fused dropout inside the joint runs as a torch dropout on the joint output;
 - ``fuse_softmax_backward`` is accepted; the loss consumes log-probs and
   returns grads w.r.t. them (the log_softmax backward is chained by autograd
@@ -17,9 +18,16 @@ from ..._ext import get_ext
 
 class TransducerJointFunc(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, f, g, f_len, g_len, relu):
+    def forward(ctx, f, g, f_len, g_len, relu, batch_offset=None, packed_batch=0):
         ext = get_ext("transducer")
-        (out,) = ext.joint_forward(f, g, f_len, g_len, relu)
+        if batch_offset is not None:
+            # exclusive offsets from the reference's inclusive cumsum
+            off = torch.cat([batch_offset.new_zeros(1), batch_offset[:-1]])
+            (out,) = ext.joint_forward_packed(f, g, f_len, g_len, off, packed_batch, relu)
+            ctx.off = off
+        else:
+            (out,) = ext.joint_forward(f, g, f_len, g_len, relu)
+            ctx.off = None
         ctx.save_for_backward(out, f_len, g_len)
         ctx.dims = (f.size(0), f.size(1), g.size(1), f.size(2))
         ctx.relu = relu
@@ -30,22 +38,33 @@ class TransducerJointFunc(torch.autograd.Function):
         ext = get_ext("transducer")
         out, f_len, g_len = ctx.saved_tensors
         B, T, U, H = ctx.dims
-        df, dg = ext.joint_backward(grad_out, out, f_len, g_len, B, T, U, H, ctx.relu)
-        return df, dg, None, None, None
+        if ctx.off is not None:
+            df, dg = ext.joint_backward_packed(grad_out, out, f_len, g_len, ctx.off, B, T, U, H,
+                                               ctx.relu)
+        else:
+            df, dg = ext.joint_backward(grad_out, out, f_len, g_len, B, T, U, H, ctx.relu)
+        return df, dg, None, None, None, None, None
 
 
 class TransducerJoint(torch.nn.Module):
     def __init__(self, pack_output=False, relu=False, dropout=False, opt=1, fwd_tile_size=4,
                  dropout_prob=0.0, probe_mask=False):
         super().__init__()
-        if pack_output:
-            raise NotImplementedError("packed joint output lands in a later round")
+        self.pack_output = pack_output
         self.relu = relu
         self.dropout = dropout
         self.dropout_prob = dropout_prob
         self.mask_probe = [] if (relu or dropout) and probe_mask else None
 
     def forward(self, f, g, f_len, g_len, batch_offset=None, packed_batch=0):
+        if self.pack_output:
+            assert batch_offset is not None and packed_batch > 0, \
+                "pack_output needs batch_offset (cumsum of f_len*g_len) and packed_batch"
+            out = TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, batch_offset,
+                                            packed_batch)
+            if self.dropout and self.training:
+                out = torch.nn.functional.dropout(out, p=self.dropout_prob)
+            return out
         if not f.is_cuda:
             # reference math on CPU
             out = f.unsqueeze(2) + g.unsqueeze(1)
@@ -55,7 +74,7 @@ class TransducerJoint(torch.nn.Module):
             if self.relu:
                 out = torch.relu(out)
         else:
-            out = TransducerJointFunc.apply(f, g, f_len, g_len, self.relu)
+            out = TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, None, 0)
         if self.dropout and self.training:
             out = torch.nn.functional.dropout(out, p=self.dropout_prob)
         return out
@@ -63,11 +82,16 @@ class TransducerJoint(torch.nn.Module):
 
 class TransducerLossFunc(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, label, f_len, y_len, blank_idx):
+    def forward(ctx, x, label, f_len, y_len, blank_idx, batch_offset=None, max_f_len=0):
         ext = get_ext("transducer")
-        losses, alpha = ext.loss_forward(x, label, f_len, y_len, blank_idx)
+        off = None
+        if batch_offset is not None:
+            off = torch.cat([batch_offset.new_zeros(1), batch_offset[:-1]])
+        losses, alpha = ext.loss_forward(x, label, f_len, y_len, blank_idx, off, max_f_len)
         ctx.save_for_backward(x, label, alpha, f_len, y_len)
         ctx.blank_idx = blank_idx
+        ctx.off = off
+        ctx.max_f_len = max_f_len
         return losses
 
     @staticmethod
@@ -75,8 +99,8 @@ class TransducerLossFunc(torch.autograd.Function):
         ext = get_ext("transducer")
         x, label, alpha, f_len, y_len = ctx.saved_tensors
         dx = ext.loss_backward(x, label, alpha, grad_loss.contiguous(), f_len, y_len,
-                               ctx.blank_idx)
-        return dx, None, None, None, None
+                               ctx.blank_idx, ctx.off, ctx.max_f_len)
+        return dx, None, None, None, None, None, None
 
 
 def _ref_rnnt_loss(x, label, f_len, y_len, blank):
@@ -106,13 +130,18 @@ def _ref_rnnt_loss(x, label, f_len, y_len, blank):
 class TransducerLoss(torch.nn.Module):
     def __init__(self, fuse_softmax_backward=True, opt=1, packed_input=False):
         super().__init__()
-        if packed_input:
-            raise NotImplementedError("packed loss input lands in a later round")
+        self.packed_input = packed_input
         self.fuse_softmax_backward = fuse_softmax_backward
 
     def forward(self, x, label, f_len, y_len, blank_idx=0, batch_offset=None, max_f_len=None,
                 debug_list=None):
-        """x: log-probs [B, T, U, V]; label: [B, U-1]; f_len, y_len per batch."""
+        """x: log-probs [B, T, U, V] (dense) or [packed_rows, V] with
+        batch_offset = cumsum(f_len*(y_len+1)); label: [B, U-1]."""
+        if self.packed_input:
+            assert batch_offset is not None and max_f_len is not None, \
+                "packed_input needs batch_offset (cumsum of f_len*(y_len+1)) and max_f_len"
+            return TransducerLossFunc.apply(x, label, f_len, y_len, blank_idx, batch_offset,
+                                            max_f_len)
         if not x.is_cuda:
             return _ref_rnnt_loss(x, label, f_len, y_len, blank_idx)
-        return TransducerLossFunc.apply(x, label, f_len, y_len, blank_idx)
+        return TransducerLossFunc.apply(x, label, f_len, y_len, blank_idx, None, 0)

@@ -79,6 +79,76 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_g_kernel(
   }
 }
 
+// ---- packed joint: out rows = sum_b f_len[b]*g_len[b] (batch_offset is
+// the EXCLUSIVE prefix of f_len*g_len) ----
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_packed_kernel(
+    const T* __restrict__ f, const T* __restrict__ g, T* __restrict__ out,
+    const int* __restrict__ f_len, const int* __restrict__ g_len,
+    const long* __restrict__ off /* [B+1] exclusive */, long B, long Tm, long U, long H,
+    long total_rows) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_rows * H;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / H;
+    const long h = i % H;
+    // binary search: largest b with off[b] <= row
+    long lo = 0, hi = B - 1;
+    while (lo < hi) {
+      const long mid = (lo + hi + 1) >> 1;
+      if (off[mid] <= row) lo = mid; else hi = mid - 1;
+    }
+    const long b = lo;
+    const long local = row - off[b];
+    const long gl = g_len[b];
+    const long t = local / gl;
+    const long u = local % gl;
+    float v = to_float(f[(b * Tm + t) * H + h]) + to_float(g[(b * U + u) * H + h]);
+    if (RELU) v = fmaxf(v, 0.f);
+    out[i] = from_float<T>(v);
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_f_kernel(
+    const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ df,
+    const int* __restrict__ f_len, const int* __restrict__ g_len,
+    const long* __restrict__ off, long B, long Tm, long U, long H) {
+  const long bt = blockIdx.x;
+  const long b = bt / Tm, t = bt % Tm;
+  const int ulen = (t < f_len[b]) ? g_len[b] : 0;
+  const long base = off[b] + t * (long)g_len[b];
+  for (long h = threadIdx.x; h < H; h += blockDim.x) {
+    float acc = 0.f;
+    for (long u = 0; u < ulen; ++u) {
+      const long idx = (base + u) * H + h;
+      float d = to_float(dout[idx]);
+      if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
+      acc += d;
+    }
+    df[(b * Tm + t) * H + h] = from_float<T>(acc);
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_g_kernel(
+    const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ dg,
+    const int* __restrict__ f_len, const int* __restrict__ g_len,
+    const long* __restrict__ off, long B, long Tm, long U, long H) {
+  const long bu = blockIdx.x;
+  const long b = bu / U, u = bu % U;
+  const int tlen = (u < g_len[b]) ? f_len[b] : 0;
+  for (long h = threadIdx.x; h < H; h += blockDim.x) {
+    float acc = 0.f;
+    for (long t = 0; t < tlen; ++t) {
+      const long idx = (off[b] + t * (long)g_len[b] + u) * H + h;
+      float d = to_float(dout[idx]);
+      if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
+      acc += d;
+    }
+    dg[(b * U + u) * H + h] = from_float<T>(acc);
+  }
+}
+
 __device__ __forceinline__ float log_add(float a, float b) {
   if (a == -INFINITY) return b;
   if (b == -INFINITY) return a;
@@ -88,16 +158,17 @@ __device__ __forceinline__ float log_add(float a, float b) {
 
 // alpha DP: one block per batch element; anti-diagonal parallel.
 // x: log-probs [B, T, U, V]; label: [B, U-1]; alpha: [B, T, U] fp32.
-template <typename T>
+template <typename T, bool PACKED>
 __global__ void __launch_bounds__(TJ_BLOCK) rnnt_alpha_kernel(
     const T* __restrict__ x, const int* __restrict__ label, float* __restrict__ alpha,
     float* __restrict__ losses, const int* __restrict__ f_len, const int* __restrict__ y_len,
-    long B, long Tm, long U, long V, int blank) {
+    const long* __restrict__ off, long B, long Tm, long U, long V, int blank) {
   const long b = blockIdx.x;
   const int Tb = f_len[b];
   const int Ub = y_len[b] + 1;
   float* al = alpha + b * Tm * U;
-  const T* xb = x + b * Tm * U * V;
+  const T* xb = PACKED ? x + off[b] * V : x + b * Tm * U * V;
+  const long su = PACKED ? Ub : U;  // u-stride of the x rows
 
   for (int d = 0; d < Tb + Ub - 1; ++d) {
     for (int t = threadIdx.x; t <= d; t += blockDim.x) {
@@ -109,11 +180,11 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_alpha_kernel(
       } else {
         float from_blank = -INFINITY, from_label = -INFINITY;
         if (t > 0) {
-          from_blank = al[(t - 1) * U + u] + to_float(xb[((long)(t - 1) * U + u) * V + blank]);
+          from_blank = al[(t - 1) * U + u] + to_float(xb[((long)(t - 1) * su + u) * V + blank]);
         }
         if (u > 0) {
           const int y = label[b * (U - 1) + (u - 1)];
-          from_label = al[(long)t * U + (u - 1)] + to_float(xb[((long)t * U + (u - 1)) * V + y]);
+          from_label = al[(long)t * U + (u - 1)] + to_float(xb[((long)t * su + (u - 1)) * V + y]);
         }
         v = log_add(from_blank, from_label);
       }
@@ -122,21 +193,22 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_alpha_kernel(
     __syncthreads();
   }
   if (threadIdx.x == 0) {
-    const float last_blank = to_float(xb[((long)(Tb - 1) * U + (Ub - 1)) * V + blank]);
+    const float last_blank = to_float(xb[((long)(Tb - 1) * su + (Ub - 1)) * V + blank]);
     losses[b] = -(al[(long)(Tb - 1) * U + (Ub - 1)] + last_blank);
   }
 }
 
-template <typename T>
+template <typename T, bool PACKED>
 __global__ void __launch_bounds__(TJ_BLOCK) rnnt_beta_kernel(
     const T* __restrict__ x, const int* __restrict__ label, float* __restrict__ beta,
-    const int* __restrict__ f_len, const int* __restrict__ y_len, long B, long Tm, long U, long V,
-    int blank) {
+    const int* __restrict__ f_len, const int* __restrict__ y_len,
+    const long* __restrict__ off, long B, long Tm, long U, long V, int blank) {
   const long b = blockIdx.x;
   const int Tb = f_len[b];
   const int Ub = y_len[b] + 1;
   float* be = beta + b * Tm * U;
-  const T* xb = x + b * Tm * U * V;
+  const T* xb = PACKED ? x + off[b] * V : x + b * Tm * U * V;
+  const long su = PACKED ? Ub : U;
 
   for (int d = Tb + Ub - 2; d >= 0; --d) {
     for (int t = threadIdx.x; t <= d; t += blockDim.x) {
@@ -144,15 +216,15 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_beta_kernel(
       if (t >= Tb || u >= Ub) continue;
       float v;
       if (t == Tb - 1 && u == Ub - 1) {
-        v = to_float(xb[((long)t * U + u) * V + blank]);
+        v = to_float(xb[((long)t * su + u) * V + blank]);
       } else {
         float via_blank = -INFINITY, via_label = -INFINITY;
         if (t + 1 < Tb) {
-          via_blank = to_float(xb[((long)t * U + u) * V + blank]) + be[(long)(t + 1) * U + u];
+          via_blank = to_float(xb[((long)t * su + u) * V + blank]) + be[(long)(t + 1) * U + u];
         }
         if (u + 1 < Ub) {
           const int y = label[b * (U - 1) + u];
-          via_label = to_float(xb[((long)t * U + u) * V + y]) + be[(long)t * U + (u + 1)];
+          via_label = to_float(xb[((long)t * su + u) * V + y]) + be[(long)t * U + (u + 1)];
         }
         v = log_add(via_blank, via_label);
       }
@@ -163,12 +235,12 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_beta_kernel(
 }
 
 // grads over log-probs: nonzero only at blank and label entries.
-template <typename T>
+template <typename T, bool PACKED>
 __global__ void __launch_bounds__(TJ_BLOCK) rnnt_grad_kernel(
     const T* __restrict__ x, const int* __restrict__ label, const float* __restrict__ alpha,
     const float* __restrict__ beta, const float* __restrict__ grad_loss, T* __restrict__ dx,
-    const int* __restrict__ f_len, const int* __restrict__ y_len, long B, long Tm, long U, long V,
-    int blank) {
+    const int* __restrict__ f_len, const int* __restrict__ y_len,
+    const long* __restrict__ off, long B, long Tm, long U, long V, int blank) {
   const long total = B * Tm * U;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
@@ -180,8 +252,9 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_grad_kernel(
     if (t >= Tb || u >= Ub) continue;
     const float lz = beta[b * Tm * U + 0];  // beta[0,0] == log Z
     const float gl = grad_loss[b];
-    const T* xb = x + (b * Tm * U + t * U + u) * V;
-    T* dxb = dx + (b * Tm * U + t * U + u) * V;
+    const long xrow = PACKED ? (off[b] + t * (long)Ub + u) : (b * Tm * U + t * U + u);
+    const T* xb = x + xrow * V;
+    T* dxb = dx + xrow * V;
     const float a = alpha[b * Tm * U + t * U + u];
     // blank transition -> (t+1, u) (or termination at (Tb-1, Ub-1))
     float bnext;
@@ -271,20 +344,37 @@ std::vector<at::Tensor> transducer_joint_backward(at::Tensor grad_out, at::Tenso
 }
 
 std::vector<at::Tensor> transducer_loss_forward(at::Tensor x, at::Tensor label, at::Tensor f_len,
-                                                at::Tensor y_len, long blank_idx) {
+                                                at::Tensor y_len, long blank_idx,
+                                                c10::optional<at::Tensor> batch_offset,
+                                                long max_f_len) {
   auto xc = x.contiguous();
   auto lab = label.to(at::kInt).contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
   auto yl = y_len.to(at::kInt).contiguous();
-  const long B = xc.size(0), Tm = xc.size(1), U = xc.size(2), V = xc.size(3);
+  const bool packed = batch_offset.has_value();
+  const long B = fl.numel();
+  const long U = lab.size(1) + 1;
+  const long Tm = packed ? max_f_len : xc.size(1);
+  const long V = xc.size(-1);
+  at::Tensor off;
+  if (packed) off = batch_offset->to(at::kLong).contiguous();
   auto fopts = xc.options().dtype(at::kFloat);
   auto alpha = at::full({B, Tm, U}, -INFINITY, fopts);
   auto losses = at::empty({B}, fopts);
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "transducer_loss_forward", ([&] {
-    hipLaunchKernelGGL((rnnt_alpha_kernel<scalar_t>), dim3((uint32_t)B), dim3(TJ_BLOCK), 0,
-                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
-                       alpha.data_ptr<float>(), losses.data_ptr<float>(), fl.data_ptr<int>(),
-                       yl.data_ptr<int>(), B, Tm, U, V, (int)blank_idx);
+    if (packed) {
+      hipLaunchKernelGGL((rnnt_alpha_kernel<scalar_t, true>), dim3((uint32_t)B), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)xc.data_ptr(),
+                         lab.data_ptr<int>(), alpha.data_ptr<float>(), losses.data_ptr<float>(),
+                         fl.data_ptr<int>(), yl.data_ptr<int>(), off.data_ptr<long>(), B, Tm, U,
+                         V, (int)blank_idx);
+    } else {
+      hipLaunchKernelGGL((rnnt_alpha_kernel<scalar_t, false>), dim3((uint32_t)B), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)xc.data_ptr(),
+                         lab.data_ptr<int>(), alpha.data_ptr<float>(), losses.data_ptr<float>(),
+                         fl.data_ptr<int>(), yl.data_ptr<int>(), (const long*)nullptr, B, Tm, U,
+                         V, (int)blank_idx);
+    }
   }()));
   HIP_CHECK(hipGetLastError());
   return {losses, alpha};
@@ -292,32 +382,124 @@ std::vector<at::Tensor> transducer_loss_forward(at::Tensor x, at::Tensor label, 
 
 at::Tensor transducer_loss_backward(at::Tensor x, at::Tensor label, at::Tensor alpha,
                                     at::Tensor grad_loss, at::Tensor f_len, at::Tensor y_len,
-                                    long blank_idx) {
+                                    long blank_idx, c10::optional<at::Tensor> batch_offset,
+                                    long max_f_len) {
   auto xc = x.contiguous();
   auto lab = label.to(at::kInt).contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
   auto yl = y_len.to(at::kInt).contiguous();
   auto gl = grad_loss.to(at::kFloat).contiguous();
-  const long B = xc.size(0), Tm = xc.size(1), U = xc.size(2), V = xc.size(3);
+  const bool packed = batch_offset.has_value();
+  const long B = fl.numel();
+  const long U = lab.size(1) + 1;
+  const long Tm = packed ? max_f_len : xc.size(1);
+  const long V = xc.size(-1);
+  at::Tensor off;
+  if (packed) off = batch_offset->to(at::kLong).contiguous();
   auto fopts = xc.options().dtype(at::kFloat);
   auto beta = at::full({B, Tm, U}, -INFINITY, fopts);
   auto dx = at::zeros_like(xc);
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "transducer_loss_backward", ([&] {
-    hipLaunchKernelGGL((rnnt_beta_kernel<scalar_t>), dim3((uint32_t)B), dim3(TJ_BLOCK), 0,
-                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
-                       beta.data_ptr<float>(), fl.data_ptr<int>(), yl.data_ptr<int>(), B, Tm, U,
-                       V, (int)blank_idx);
-    HIP_CHECK(hipGetLastError());
+    const long* offp = packed ? off.data_ptr<long>() : nullptr;
     const long total = B * Tm * U;
     const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
-    hipLaunchKernelGGL((rnnt_grad_kernel<scalar_t>), dim3(grid), dim3(TJ_BLOCK), 0,
-                       current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
-                       alpha.data_ptr<float>(), beta.data_ptr<float>(), gl.data_ptr<float>(),
-                       (scalar_t*)dx.data_ptr(), fl.data_ptr<int>(), yl.data_ptr<int>(), B, Tm,
-                       U, V, (int)blank_idx);
+    if (packed) {
+      hipLaunchKernelGGL((rnnt_beta_kernel<scalar_t, true>), dim3((uint32_t)B), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)xc.data_ptr(),
+                         lab.data_ptr<int>(), beta.data_ptr<float>(), fl.data_ptr<int>(),
+                         yl.data_ptr<int>(), offp, B, Tm, U, V, (int)blank_idx);
+      HIP_CHECK(hipGetLastError());
+      hipLaunchKernelGGL((rnnt_grad_kernel<scalar_t, true>), dim3(grid), dim3(TJ_BLOCK), 0,
+                         current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
+                         alpha.data_ptr<float>(), beta.data_ptr<float>(), gl.data_ptr<float>(),
+                         (scalar_t*)dx.data_ptr(), fl.data_ptr<int>(), yl.data_ptr<int>(), offp,
+                         B, Tm, U, V, (int)blank_idx);
+    } else {
+      hipLaunchKernelGGL((rnnt_beta_kernel<scalar_t, false>), dim3((uint32_t)B), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)xc.data_ptr(),
+                         lab.data_ptr<int>(), beta.data_ptr<float>(), fl.data_ptr<int>(),
+                         yl.data_ptr<int>(), offp, B, Tm, U, V, (int)blank_idx);
+      HIP_CHECK(hipGetLastError());
+      hipLaunchKernelGGL((rnnt_grad_kernel<scalar_t, false>), dim3(grid), dim3(TJ_BLOCK), 0,
+                         current_stream(), (const scalar_t*)xc.data_ptr(), lab.data_ptr<int>(),
+                         alpha.data_ptr<float>(), beta.data_ptr<float>(), gl.data_ptr<float>(),
+                         (scalar_t*)dx.data_ptr(), fl.data_ptr<int>(), yl.data_ptr<int>(), offp,
+                         B, Tm, U, V, (int)blank_idx);
+    }
   }()));
   HIP_CHECK(hipGetLastError());
   return dx;
+}
+
+std::vector<at::Tensor> transducer_joint_forward_packed(at::Tensor f, at::Tensor g,
+                                                        at::Tensor f_len, at::Tensor g_len,
+                                                        at::Tensor batch_offset,
+                                                        long packed_batch, bool relu) {
+  auto fc = f.contiguous();
+  auto gc = g.contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto glen = g_len.to(at::kInt).contiguous();
+  auto off = batch_offset.to(at::kLong).contiguous();
+  const long B = fc.size(0), Tm = fc.size(1), H = fc.size(2), U = gc.size(1);
+  auto out = at::empty({packed_batch, H}, fc.options());
+  const long total = packed_batch * H;
+  const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
+  APEX_DISPATCH_FLOAT_HALF_BF(fc.scalar_type(), "transducer_joint_forward_packed", ([&] {
+    if (relu) {
+      hipLaunchKernelGGL((joint_fwd_packed_kernel<scalar_t, true>), dim3(grid), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)fc.data_ptr(),
+                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H, packed_batch);
+    } else {
+      hipLaunchKernelGGL((joint_fwd_packed_kernel<scalar_t, false>), dim3(grid), dim3(TJ_BLOCK),
+                         0, current_stream(), (const scalar_t*)fc.data_ptr(),
+                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H, packed_batch);
+    }
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return {out};
+}
+
+std::vector<at::Tensor> transducer_joint_backward_packed(at::Tensor grad_out, at::Tensor out,
+                                                         at::Tensor f_len, at::Tensor g_len,
+                                                         at::Tensor batch_offset, long B, long Tm,
+                                                         long U, long H, bool relu) {
+  auto dout = grad_out.contiguous();
+  auto fl = f_len.to(at::kInt).contiguous();
+  auto glen = g_len.to(at::kInt).contiguous();
+  auto off = batch_offset.to(at::kLong).contiguous();
+  auto df = at::empty({B, Tm, H}, dout.options());
+  auto dg = at::empty({B, U, H}, dout.options());
+  APEX_DISPATCH_FLOAT_HALF_BF(dout.scalar_type(), "transducer_joint_backward_packed", ([&] {
+    if (relu) {
+      hipLaunchKernelGGL((joint_bwd_packed_f_kernel<scalar_t, true>), dim3((uint32_t)(B * Tm)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H);
+      hipLaunchKernelGGL((joint_bwd_packed_g_kernel<scalar_t, true>), dim3((uint32_t)(B * U)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H);
+    } else {
+      hipLaunchKernelGGL((joint_bwd_packed_f_kernel<scalar_t, false>), dim3((uint32_t)(B * Tm)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H);
+      hipLaunchKernelGGL((joint_bwd_packed_g_kernel<scalar_t, false>), dim3((uint32_t)(B * U)),
+                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
+                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
+                         U, H);
+    }
+  }()));
+  HIP_CHECK(hipGetLastError());
+  return {df, dg};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -325,4 +507,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("joint_backward", &transducer_joint_backward);
   m.def("loss_forward", &transducer_loss_forward);
   m.def("loss_backward", &transducer_loss_backward);
+  m.def("joint_forward_packed", &transducer_joint_forward_packed);
+  m.def("joint_backward_packed", &transducer_joint_backward_packed);
 }
