@@ -99,6 +99,19 @@ class DistributedDataParallel(torch.nn.Module):
             self._streams = []
         self._next_stream = 0
 
+        # one communicator per all-reduce stream: collectives on a single
+        # process group serialize on its internal stream, so genuine overlap
+        # of concurrent bucket reductions needs distinct communicators
+        # (reference: allreduce_communicators / num_allreduce_streams)
+        if allreduce_communicators is not None:
+            self._process_groups = list(allreduce_communicators)
+        elif num_allreduce_streams > 1 and dist.is_initialized():
+            ranks = list(range(self.world_size))
+            self._process_groups = [dist.new_group(ranks) for _ in range(num_allreduce_streams)]
+        else:
+            self._process_groups = [None]
+        self._next_pg = 0
+
         # in-flight (work, bucket_grads, flat, stream) records
         self._pending = []
         self._active_params = []
@@ -169,6 +182,8 @@ class DistributedDataParallel(torch.nn.Module):
             self._next_stream = (self._next_stream + 1) % len(self._streams)
             stream.wait_stream(torch.cuda.current_stream())
 
+        pg = self._process_groups[self._next_pg]
+        self._next_pg = (self._next_pg + 1) % len(self._process_groups)
         ctx = torch.cuda.stream(stream) if stream is not None else _nullcontext()
         with ctx:
             flat = flatten(grads)
@@ -177,7 +192,7 @@ class DistributedDataParallel(torch.nn.Module):
                 flat = flat.float()
             if self.gradient_predivide_factor != 1.0:
                 flat.div_(self.gradient_predivide_factor)
-            work = dist.all_reduce(flat, async_op=True)
+            work = dist.all_reduce(flat, async_op=True, group=pg)
         self._pending.append((work, grads, flat, orig_dtype, stream))
 
     def _finish_backward(self):
@@ -214,6 +229,10 @@ class DistributedDataParallel(torch.nn.Module):
 
     def forward(self, *args, **kwargs):
         self._active_params = []
+        # reset round-robin state so stream/communicator assignment is
+        # identical across ranks every iteration (collective matching)
+        self._next_stream = 0
+        self._next_pg = 0
         return self.module(*args, **kwargs)
 
     def state_dict(self, *args, **kwargs):
