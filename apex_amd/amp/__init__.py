@@ -31,7 +31,6 @@ monkey-patching for O1):
 import contextlib
 import functools
 import itertools
-import warnings
 
 import torch
 

@@ -14,9 +14,8 @@ MIOpen (torch conv2d) with the scale/bias/ReLU epilogue composed around them
 """
 
 import torch
-import torch.distributed as dist
 
-from ..conv_bias_relu import ConvBiasReLU, ConvFrozenScaleBiasReLU
+from ..conv_bias_relu import ConvFrozenScaleBiasReLU
 
 
 class FrozenBatchNorm2d(torch.nn.Module):

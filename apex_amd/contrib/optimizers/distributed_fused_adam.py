@@ -20,7 +20,6 @@ differences from a line-by-line port, for MI355X:
   all_reduce / all_gather-list — the sharding logic is identical.
 """
 
-import math
 from collections import defaultdict
 
 import torch

@@ -6,7 +6,6 @@ file-store init, per-rank ``dist.init_process_group``. ``NcclDistributedTestBase
 requires GPUs (RCCL); ``GlooDistributedTestBase`` runs on CPU CI.
 """
 
-import os
 
 import torch
 import torch.distributed as dist

@@ -12,7 +12,6 @@ Device path: one HIP multi-tensor kernel per dtype group
 the no-GPU CI and as the numerics oracle).
 """
 
-import math
 
 import torch
 
