@@ -120,6 +120,27 @@ def _wrap_forward_input_cast(model, dtype):
     model._amp_input_cast_wrapped = True
 
 
+def _wrap_forward_output_cast(model, dtype):
+    """cast_model_outputs: force the forward outputs to a dtype (reference
+    kwarg on amp.initialize)."""
+    old_forward = model.forward
+
+    def cast_tree(x):
+        if torch.is_tensor(x) and x.is_floating_point():
+            return x.to(dtype)
+        if isinstance(x, (list, tuple)):
+            return type(x)(cast_tree(v) for v in x)
+        if isinstance(x, dict):
+            return {k: cast_tree(v) for k, v in x.items()}
+        return x
+
+    @functools.wraps(old_forward)
+    def new_forward(*args, **kwargs):
+        return cast_tree(old_forward(*args, **kwargs))
+
+    model.forward = new_forward
+
+
 def _process_optimizer_o2(optimizer, cast_type, verbose=False):
     """Build fp32 masters inside the optimizer (the amp-O2 master contract,
     evidenced by apex/optimizers/fused_sgd.py:165-230)."""
@@ -302,6 +323,9 @@ def initialize(
         cast_dtype = torch.float16 if cast_model_type is None else cast_model_type
         for model in model_list:
             _wrap_forward_autocast(model, cast_dtype)
+    if cast_model_outputs is not None:
+        for model in model_list:
+            _wrap_forward_output_cast(model, cast_model_outputs)
 
     # --- optimizers ---
     if props["master_weights"]:

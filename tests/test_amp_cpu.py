@@ -128,3 +128,47 @@ def test_state_dict_roundtrip():
     sd = amp.state_dict()
     assert "loss_scaler0" in sd
     amp.load_state_dict(sd)
+
+
+def test_cast_model_outputs():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O1", cast_model_type=torch.bfloat16,
+                                cast_model_outputs=torch.float32, verbosity=0)
+    out = model(torch.randn(4, 16))
+    assert out.dtype == torch.float32  # autocast produced bf16, output cast back
+
+
+def test_delay_unscale():
+    model = make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O2",
+                                cast_model_type=torch.bfloat16, loss_scale=64.0, verbosity=0)
+    x = torch.randn(4, 16).bfloat16()
+    opt.zero_grad()
+    # two backwards accumulating, unscale only on the second
+    loss1 = model(x).float().sum()
+    with amp.scale_loss(loss1, opt, delay_unscale=True) as s1:
+        s1.backward()
+    loss2 = model(x).float().sum()
+    with amp.scale_loss(loss2, opt) as s2:
+        s2.backward()
+    opt.step()  # must not blow up; masters received accumulated unscaled grads
+    for p in amp.master_params(opt):
+        assert torch.isfinite(p).all()
+
+
+def test_fp16_optimizer_dynamic_scale_backoff():
+    from apex_amd.contrib.optimizers import FP16_Optimizer
+
+    model = torch.nn.Linear(8, 4).to(torch.bfloat16)
+    inner = torch.optim.SGD(model.parameters(), lr=0.1)
+    opt = FP16_Optimizer(inner, dynamic_loss_scale=True,
+                         dynamic_loss_args={"init_scale": 2.0 ** 8})
+    before = opt.loss_scale
+    x = torch.full((4, 8), 1e30, dtype=torch.bfloat16)
+    loss = model(x).float().sum() * 1e30
+    opt.zero_grad()
+    opt.backward(loss)
+    opt.step()  # overflow → skip + backoff
+    assert opt.loss_scale < before
