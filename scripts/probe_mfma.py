@@ -56,6 +56,12 @@ def main():
     tm = t(lambda: mfma.gemm_bias_gelu(X, W, b, True))
     tl = t(lambda: fd.linear_gelu_linear_forward(X, W, b, W2, b2))  # incl. 2nd GEMM
     t1 = t(lambda: mfma.gemm_bias(X, W, b))
+    t2 = t(lambda: mfma.gemm_bias_v2(X, W, b))
+    # v2 refcheck
+    o2v = mfma.gemm_bias_v2(X, W, b)
+    zv = X.float() @ W.float().t() + b.float()
+    print(f"v2 rel err {(o2v.float()-zv).abs().max().item()/zv.abs().max().item():.5f}  "
+          f"v2 gemm_bias { t2:.3f} ms ({2.0*16384*3072*768/t2/1e9:.0f} TF)")
     flops = 2.0 * M * N * K
     print(f"mfma gemm_bias_gelu {tm:.3f} ms ({flops/tm/1e9:.0f} TF)  "
           f"mfma gemm_bias {t1:.3f} ms ({flops/t1/1e9:.0f} TF)")
