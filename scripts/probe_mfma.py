@@ -62,6 +62,13 @@ def main():
     zv = X.float() @ W.float().t() + b.float()
     print(f"v2 rel err {(o2v.float()-zv).abs().max().item()/zv.abs().max().item():.5f}  "
           f"v2 gemm_bias { t2:.3f} ms ({2.0*16384*3072*768/t2/1e9:.0f} TF)")
+    t2g = t(lambda: mfma.gemm_bias_gelu_v2(X, W, b, True))
+    ts1 = t(lambda: fd.linear_bias_forward(X, W, b))  # split-path stage-1 GEMM alone
+    og, gg = mfma.gemm_bias_gelu_v2(X, W, b, True)
+    refg = torch.nn.functional.gelu(zv, approximate="tanh")
+    print(f"v2 gemm_bias_gelu {t2g:.3f} ms (gelu rel err "
+          f"{(og.float()-refg).abs().max().item()/zv.abs().max().item():.5f}); "
+          f"hipBLASLt stage-1 BIAS GEMM alone {ts1:.3f} ms (+ separate gelu pass on top)")
     flops = 2.0 * M * N * K
     print(f"mfma gemm_bias_gelu {tm:.3f} ms ({flops/tm/1e9:.0f} TF)  "
           f"mfma gemm_bias {t1:.3f} ms ({flops/t1/1e9:.0f} TF)")
