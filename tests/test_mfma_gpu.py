@@ -46,3 +46,22 @@ def test_mfma_gemm_bias():
     out = mfma.gemm_bias(X, W, b)
     ref = X.float() @ W.float().t() + b.float()
     assert ((out.float() - ref).abs().max() / ref.abs().max()) < 2e-2
+
+
+@pytest.mark.parametrize("shape", [(256, 128, 128), (512, 384, 768)])
+def test_mfma_gemm_v2(shape):
+    import apex_amd._mfma as mfma
+
+    M, N, K = shape
+    torch.manual_seed(3)
+    X = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    W = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    b = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    out = mfma.gemm_bias_v2(X, W, b)
+    ref = X.float() @ W.float().t() + b.float()
+    assert ((out.float() - ref).abs().max() / ref.abs().max()) < 2e-2
+    og, gi = mfma.gemm_bias_gelu_v2(X, W, b, True)
+    refg = torch.nn.functional.gelu(ref, approximate="tanh")
+    scale = ref.abs().max()
+    assert ((gi.float() - ref).abs().max() / scale) < 2e-2
+    assert ((og.float() - refg).abs().max() / scale) < 2e-2
