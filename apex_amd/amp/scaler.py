@@ -76,7 +76,13 @@ class LossScaler:
 
         Returns True if an overflow was detected (host-synchronizing on GPU —
         one int read per iteration, matching the reference contract).
+
+        The dynamic-scale state ticks ONCE per optimizer step: callers that
+        unscale several grad sets in one step (amp O2 does masters then the
+        fp32 group) pass ``scale_override`` on the extra calls, which skips
+        the update.
         """
+        update_state = scale_override is None
         scale = self._loss_scale if scale_override is None else scale_override
         if len(grads_in) == 0:
             self._has_overflow = False
@@ -92,7 +98,7 @@ class LossScaler:
                 [grads_in, grads_out],
                 1.0 / scale,
             )
-            if self.dynamic:
+            if self.dynamic and update_state:
                 # on-device scale update; host reads only the skip decision
                 amp_C.update_scale_hysteresis(
                     self._scale_t,
@@ -105,7 +111,7 @@ class LossScaler:
                     self._hysteresis,
                 )
             self._has_overflow = bool(self._overflow_buf.item())
-            if self.dynamic:
+            if self.dynamic and update_state:
                 self._loss_scale = float(self._scale_t.item())
                 self._loss_scale = min(self._max_loss_scale, max(self._min_loss_scale, self._loss_scale))
                 self._scale_t.fill_(self._loss_scale)
@@ -118,7 +124,8 @@ class LossScaler:
                     overflow = True
                 go.copy_(gf.to(go.dtype))
             self._has_overflow = overflow
-            self.update_scale()
+            if update_state:
+                self.update_scale()
         return self._has_overflow
 
     def has_overflow(self):

@@ -172,3 +172,14 @@ def test_fp16_optimizer_dynamic_scale_backoff():
     opt.backward(loss)
     opt.step()  # overflow → skip + backoff
     assert opt.loss_scale < before
+
+
+def test_dynamic_scaler_single_tick_per_step():
+    from apex_amd.amp.scaler import LossScaler
+
+    s = LossScaler("dynamic", init_scale=2.0 ** 10, scale_window=3)
+    g = [torch.ones(4)]
+    for _ in range(3):
+        s.unscale_grads(g, g)                       # main tick
+        s.unscale_grads(g, g, scale_override=2.0 ** 10)  # extra set: no tick
+    assert s.loss_scale() == 2.0 ** 11  # exactly one growth after 3 steps
