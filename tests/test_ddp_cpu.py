@@ -88,3 +88,10 @@ def _param_sync_worker(rank, world_size):
 
 def test_ddp_broadcasts_initial_params():
     run_distributed(_param_sync_worker, world_size=2)
+
+def test_ddp_multiple_allreduce_pgs():
+    # num_allreduce_streams>1 creates one process group per stream and
+    # round-robins buckets across them; on CPU the stream list is empty but
+    # the multi-PG rotation is the same code path as on GPU.
+    run_distributed(_ddp_worker, world_size=2,
+                    args=({"num_allreduce_streams": 2, "message_size": 1},))
