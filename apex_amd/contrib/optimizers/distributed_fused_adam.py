@@ -56,6 +56,7 @@ class _Bucket:
         self.grad_data = torch.zeros(self.numel, dtype=grad_dtype, device=device)
         # sharded fp32 state (created lazily once param values are final)
         self.master_shard = None
+        self.param_remainder = None  # int16 low bits when store_param_remainders
         self.exp_avg = None
         self.exp_avg_sq = None
         self.grad_shard = None
@@ -67,7 +68,17 @@ class _Bucket:
 class DistributedFusedAdam(torch.optim.Optimizer):
     """ZeRO-2 Adam. Supported reference knobs: lr, bias_correction, betas,
     eps, weight_decay, adam_w_mode, bucket_cap_mb, overlap_grad_sync,
-    grad_sync_dtype, process_group, set_grad_none, average_grad_sync.
+    grad_sync_dtype, process_group, set_grad_none, average_grad_sync,
+    store_param_remainders, and ``state_dict(gather_on_root=True)``.
+
+    ``store_param_remainders`` (bf16 params only) keeps the sharded fp32
+    master implicitly as (bf16 param bits << 16) | int16 remainder — 2 bytes
+    of optimizer state per element instead of a 4-byte fp32 copy, with the
+    master value preserved bit-exactly (reference:
+    distributed_fused_adam.py store_param_remainders). The visible bf16
+    param is the truncated top half of the master rather than
+    round-to-nearest; the optimizer trajectory (which reads the exact
+    master) is unchanged.
     """
 
     def __init__(
@@ -86,6 +97,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         grad_sync_dtype=None,
         process_group=None,
         set_grad_none=True,
+        store_param_remainders=False,
     ):
         if amsgrad:
             raise RuntimeError("DistributedFusedAdam does not support AMSGrad")
@@ -101,6 +113,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
         self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
+        self.store_param_remainders = store_param_remainders
         self._step = 0
 
         device = self.param_groups[0]["params"][0].device
@@ -151,11 +164,43 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             offset += n
         lo = self.rank * b.shard_size
         hi = lo + b.shard_size
-        b.master_shard = b.param_data[lo:hi].float().clone()
-        b.exp_avg = torch.zeros_like(b.master_shard)
-        b.exp_avg_sq = torch.zeros_like(b.master_shard)
+        if self.store_param_remainders and dtype == torch.bfloat16:
+            # implicit master: bf16 param top bits + int16 remainder (zero at
+            # init — bf16→fp32 widening is exact, so the remainder is 0)
+            b.param_remainder = torch.zeros(b.shard_size, dtype=torch.int16,
+                                            device=self.device)
+        else:
+            b.master_shard = b.param_data[lo:hi].float().clone()
+        b.exp_avg = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
+        b.exp_avg_sq = torch.zeros_like(b.exp_avg)
         b.grad_shard = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
         self.buckets.append(b)
+
+    # ---------- implicit-master helpers (store_param_remainders) ----------
+    def _get_master(self, b):
+        """The sharded fp32 master — stored, or reconstructed bit-exactly
+        from (bf16 param bits, int16 remainder)."""
+        if b.master_shard is not None:
+            return b.master_shard
+        lo = self.rank * b.shard_size
+        top = b.param_data[lo:lo + b.shard_size].view(torch.int16).to(torch.int32)
+        low = b.param_remainder.to(torch.int32) & 0xFFFF
+        return ((top << 16) | low).view(torch.float32)
+
+    def _set_master(self, b, master):
+        """Write an updated fp32 master back; in remainder mode this also
+        updates the visible bf16 param shard (top 16 bits, truncated)."""
+        if b.master_shard is not None:
+            if master is not b.master_shard:
+                b.master_shard.copy_(master)
+            return
+        bits = master.contiguous().view(torch.int32)
+        lo = self.rank * b.shard_size
+        b.param_data[lo:lo + b.shard_size].view(torch.int16).copy_(
+            (bits >> 16).to(torch.int16))
+        low = bits & 0xFFFF
+        low = low - ((low >> 15) << 16)  # map [32768,65535] -> negative int16
+        b.param_remainder.copy_(low.to(torch.int16))
 
     def _register_hooks(self):
         self._hook_handles = []
@@ -267,16 +312,18 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             group = b.group
             beta1, beta2 = group["betas"]
             bias_correction = 1 if group["bias_correction"] else 0
+            master = self._get_master(b)
             if self.device.type == "cuda":
                 amp_C = get_ext("amp_C")
                 multi_tensor_applier(
                     amp_C.multi_tensor_adam, self._noop,
-                    [[b.grad_shard], [b.master_shard], [b.exp_avg], [b.exp_avg_sq]],
+                    [[b.grad_shard], [master], [b.exp_avg], [b.exp_avg_sq]],
                     group["lr"], beta1, beta2, group["eps"], self._step,
                     self.adam_w_mode, bias_correction, group["weight_decay"],
                 )
             else:
-                self._adam_ref(group, bias_correction, beta1, beta2, b)
+                self._adam_ref(group, bias_correction, beta1, beta2, b, master)
+            self._set_master(b, master)
 
         # param sync: shard -> param bucket, then all_gather. On GPU the
         # gathers run async on the comm stream with a stream-order dependency
@@ -287,6 +334,8 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         if use_comm_stream:
             self._comm_stream.wait_stream(torch.cuda.current_stream())
         for b in self.buckets:
+            if b.master_shard is None:
+                continue  # remainder mode: _set_master already wrote the shard
             lo = self.rank * b.shard_size
             shard = b.param_data[lo:lo + b.shard_size]
             shard.copy_(b.master_shard.to(b.param_data.dtype))
@@ -314,34 +363,56 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             b.sync_work = None
         return loss
 
-    def _adam_ref(self, group, bias_correction, beta1, beta2, b):
+    def _adam_ref(self, group, bias_correction, beta1, beta2, b, master):
         step = self._step
         bc1 = 1 - beta1 ** step if bias_correction else 1.0
         bc2 = 1 - beta2 ** step if bias_correction else 1.0
         lr, wd, eps = group["lr"], group["weight_decay"], group["eps"]
         g = b.grad_shard
         if self.adam_w_mode == 0 and wd != 0:
-            g = g + wd * b.master_shard
+            g = g + wd * master
         b.exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
         b.exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
         update = (b.exp_avg / bc1) / ((b.exp_avg_sq / bc2).sqrt() + eps)
         if self.adam_w_mode == 1 and wd != 0:
-            update = update + wd * b.master_shard
-        b.master_shard.add_(update, alpha=-lr)
+            update = update + wd * master
+        master.add_(update, alpha=-lr)
 
-    # ---------- checkpoint (shard-local, v2-style) ----------
+    # ---------- checkpoint ----------
     def state_dict(self, gather_on_root=False):
+        """Shard-local (v2-style) by default; ``gather_on_root=True`` gathers
+        the full fp32 master + moments per bucket (reference v1-style) —
+        every rank returns the same world-size-independent dict, which
+        ``load_state_dict`` can reshard onto any world size with the same
+        param list and bucket_cap."""
+        groups_sd = [
+            {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
+        ]
         if gather_on_root:
-            raise NotImplementedError("gather_on_root checkpointing lands in a later round")
+            buckets_sd = []
+            for b in self.buckets:
+                full = {}
+                for name, shard in (("master", self._get_master(b)),
+                                    ("exp_avg", b.exp_avg),
+                                    ("exp_avg_sq", b.exp_avg_sq)):
+                    if self.world_size > 1:
+                        chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                        dist.all_gather(chunks, shard.contiguous(),
+                                        group=self.process_group)
+                        t = torch.cat(chunks)[:b.numel_unpadded]
+                    else:
+                        t = shard[:b.numel_unpadded].clone()
+                    full[name] = t
+                buckets_sd.append(full)
+            return {"step": self._step, "gathered": True,
+                    "param_groups": groups_sd, "buckets": buckets_sd}
         return {
             "step": self._step,
             "world_size": self.world_size,
-            "param_groups": [
-                {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
-            ],
+            "param_groups": groups_sd,
             "buckets": [
                 {
-                    "master_shard": b.master_shard,
+                    "master_shard": self._get_master(b),
                     "exp_avg": b.exp_avg,
                     "exp_avg_sq": b.exp_avg_sq,
                 }
@@ -351,19 +422,40 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
     def load_state_dict(self, sd):
         if "buckets" not in sd:
-            raise ValueError("expected a DistributedFusedAdam shard-local state dict")
-        assert sd["world_size"] == self.world_size, (
-            "world size changed; resharding load is not supported yet"
-        )
+            raise ValueError("expected a DistributedFusedAdam state dict")
         self._step = sd["step"]
         for g, gsd in zip(self.param_groups, sd["param_groups"]):
             g.update(gsd)
-        for b, bsd in zip(self.buckets, sd["buckets"]):
-            b.master_shard.copy_(bsd["master_shard"])
-            b.exp_avg.copy_(bsd["exp_avg"])
-            b.exp_avg_sq.copy_(bsd["exp_avg_sq"])
-            lo = self.rank * b.shard_size
-            b.param_data[lo:lo + b.shard_size].copy_(b.master_shard.to(b.param_data.dtype))
+        if sd.get("gathered"):
+            # reshard a gathered (full) checkpoint onto this world size
+            for b, bsd in zip(self.buckets, sd["buckets"]):
+                lo = self.rank * b.shard_size
+                hi = lo + b.shard_size
+                for name, dst in (("master", None), ("exp_avg", b.exp_avg),
+                                  ("exp_avg_sq", b.exp_avg_sq)):
+                    full = bsd[name].to(self.device, torch.float32)
+                    pad = b.numel - full.numel()
+                    if pad:
+                        full = torch.cat([full, full.new_zeros(pad)])
+                    if name == "master":
+                        self._set_master(b, full[lo:hi].contiguous())
+                        if b.master_shard is not None:
+                            b.param_data[lo:hi].copy_(
+                                b.master_shard.to(b.param_data.dtype))
+                    else:
+                        dst.copy_(full[lo:hi])
+        else:
+            assert sd["world_size"] == self.world_size, (
+                "world size changed; use a gather_on_root=True checkpoint to reshard"
+            )
+            for b, bsd in zip(self.buckets, sd["buckets"]):
+                self._set_master(b, bsd["master_shard"].to(self.device, torch.float32))
+                b.exp_avg.copy_(bsd["exp_avg"])
+                b.exp_avg_sq.copy_(bsd["exp_avg_sq"])
+                if b.master_shard is not None:
+                    lo = self.rank * b.shard_size
+                    b.param_data[lo:lo + b.shard_size].copy_(
+                        b.master_shard.to(b.param_data.dtype))
         # rebroadcast params
         if self.world_size > 1:
             for b in self.buckets:

@@ -158,3 +158,97 @@ def test_fp16_optimizer_cpu():
     opt.step()
     for b, p in zip(before, model.parameters()):
         assert not torch.equal(b, p.detach())
+
+
+def _feed_synthetic_grads(opt, model, it):
+    for i, p in enumerate(model.parameters()):
+        torch.manual_seed(5000 * it + i)
+        p.grad = torch.randn(p.shape).to(p.dtype)
+    for p in model.parameters():
+        opt._grad_copy(p)
+
+
+def test_dist_adam_store_param_remainders_bitwise():
+    """(bf16 bits << 16) | int16 remainder must carry the fp32 master
+    bit-exactly: the optimizer trajectory matches the stored-master mode."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    models, opts = [], []
+    for remainders in (False, True):
+        m = _make_model().to(torch.bfloat16)
+        opt = DistributedFusedAdam(m.parameters(), lr=1e-3, weight_decay=0.01,
+                                   bucket_cap_mb=1, store_param_remainders=remainders)
+        models.append(m)
+        opts.append(opt)
+    assert opts[1].buckets[0].master_shard is None  # implicit master engaged
+    assert opts[1].buckets[0].param_remainder is not None
+    for it in range(6):
+        for m, o in zip(models, opts):
+            _feed_synthetic_grads(o, m, it)
+            o.step()
+        for b0, b1 in zip(opts[0].buckets, opts[1].buckets):
+            assert torch.equal(opts[0]._get_master(b0), opts[1]._get_master(b1))
+            assert torch.equal(b0.exp_avg, b1.exp_avg)
+
+
+def test_dist_adam_remainders_state_dict_roundtrip():
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    m = _make_model().to(torch.bfloat16)
+    opt = DistributedFusedAdam(m.parameters(), lr=1e-3, bucket_cap_mb=1,
+                               store_param_remainders=True)
+    for it in range(3):
+        _feed_synthetic_grads(opt, m, it)
+        opt.step()
+    sd = opt.state_dict()
+
+    m2 = _make_model().to(torch.bfloat16)
+    opt2 = DistributedFusedAdam(m2.parameters(), lr=1e-3, bucket_cap_mb=1,
+                                store_param_remainders=True)
+    opt2.load_state_dict(sd)
+    for b, b2 in zip(opt.buckets, opt2.buckets):
+        assert torch.equal(opt._get_master(b), opt2._get_master(b2))
+    for p, q in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(p.detach(), q.detach())
+
+
+def _gather_sd_worker(rank, world_size, path):
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-3, bucket_cap_mb=1)
+    for it in range(3):
+        _set_rank_grads(model, rank, world_size, it)
+        for p in model.parameters():
+            opt._grad_copy(p)
+        opt.step()
+    sd = opt.state_dict(gather_on_root=True)
+    assert sd.get("gathered")
+    # same-world reload from the gathered checkpoint
+    model2 = _make_model()
+    opt2 = DistributedFusedAdam(model2.parameters(), lr=1e-3, bucket_cap_mb=1)
+    opt2.load_state_dict(sd)
+    for p, q in zip(model.parameters(), model2.parameters()):
+        torch.testing.assert_close(p.detach(), q.detach())
+    for b, b2 in zip(opt.buckets, opt2.buckets):
+        torch.testing.assert_close(b.exp_avg, b2.exp_avg)
+    if rank == 0:
+        torch.save({"sd": sd,
+                    "params": [p.detach().clone() for p in model.parameters()]}, path)
+
+
+def test_dist_adam_gather_on_root_resharding(tmp_path):
+    path = str(tmp_path / "dfa_gathered.pt")
+    run_distributed(_gather_sd_worker, world_size=2, args=(path,))
+    # reshard the world_size=2 gathered checkpoint onto world_size=1
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    saved = torch.load(path, weights_only=False)
+    model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-3, bucket_cap_mb=1)
+    opt.load_state_dict(saved["sd"])
+    assert opt._step == saved["sd"]["step"]
+    for p, ref in zip(model.parameters(), saved["params"]):
+        torch.testing.assert_close(p.detach(), ref)
+    for b, bsd in zip(opt.buckets, saved["sd"]["buckets"]):
+        torch.testing.assert_close(opt._get_master(b)[:b.numel_unpadded], bsd["master"])
