@@ -35,9 +35,15 @@ def _group_kept_sum(w_abs, m=4, n=2):
 
 
 def search_for_good_permutation(weight2d, m=4, n=2, max_iters=100, escape_attempts=10,
-                                seed=0):
+                                seed=0, strategy="channel_swap"):
     """Return a permutation of the input channels (dim 1) improving the 2:4
-    kept magnitude. Greedy channel-swap with random restarts."""
+    kept magnitude.
+
+    strategy="channel_swap": greedy sampled channel-pair swaps (fast, larger
+    matrices). strategy="exhaustive": stripe-pair exhaustive repartitioning
+    (deterministic, strictly monotone — the reference's Exhaustive_Search)."""
+    if strategy == "exhaustive":
+        return exhaustive_search(weight2d, m=m, n=n)
     w = weight2d.detach().abs().float()
     rows, cols = w.shape
     assert cols % m == 0
@@ -72,6 +78,62 @@ def search_for_good_permutation(weight2d, m=4, n=2, max_iters=100, escape_attemp
             else:
                 perm[a], perm[b] = perm[b].item(), perm[a].item()  # revert
     return perm
+
+
+def _stripe_pair_partitions(m=4):
+    """All unique ways to partition 2*m columns into two unordered m-wide
+    stripes (C(2m-1, m-1) of them: column 0 pinned to the first stripe).
+    For m=4 that is 35 candidate layouts per stripe pair — the reference's
+    Exhaustive_Search unit (permutation_search_kernels/exhaustive_search)."""
+    import itertools
+
+    two_m = 2 * m
+    perms = []
+    for rest in itertools.combinations(range(1, two_m), m - 1):
+        a = (0,) + rest
+        b = tuple(c for c in range(two_m) if c not in a)
+        perms.append(a + b)
+    return torch.tensor(perms, dtype=torch.long)
+
+
+def exhaustive_search(weight2d, m=4, n=2, max_sweeps=8):
+    """Stripe-pair exhaustive permutation search.
+
+    For every pair of m-wide stripes, evaluates ALL unique repartitions of
+    their 2m columns (batched topk over the 35 candidates at m=4) and keeps
+    the best; sweeps until a full pass makes no improvement. Deterministic,
+    no randomness; strictly monotone in kept magnitude.
+
+    Returns the permutation of the input channels (dim 1).
+    """
+    w = weight2d.detach().abs().float()
+    rows, cols = w.shape
+    assert cols % m == 0
+    nstripes = cols // m
+    parts = _stripe_pair_partitions(m).to(w.device)  # [P, 2m]
+    perm = torch.arange(cols, device=w.device)
+
+    def pair_kept(cols2m):
+        # cols2m: [rows, 2m] -> kept magnitude per candidate partition [P]
+        wp = cols2m[:, parts]                       # [rows, P, 2m]
+        g = wp.reshape(rows, parts.shape[0], 2, m)  # two stripes of m
+        return g.topk(n, dim=3).values.sum(dim=(0, 2, 3))
+
+    for _ in range(max_sweeps):
+        improved = False
+        for i in range(nstripes - 1):
+            for j in range(i + 1, nstripes):
+                idx = torch.cat([perm[i * m:(i + 1) * m], perm[j * m:(j + 1) * m]])
+                kept = pair_kept(w[:, idx])
+                best = int(kept.argmax())
+                if best != 0 and float(kept[best]) > float(kept[0]) + 1e-7:
+                    new = idx[parts[best]]
+                    perm[i * m:(i + 1) * m] = new[:m]
+                    perm[j * m:(j + 1) * m] = new[m:]
+                    improved = True
+        if not improved:
+            break
+    return perm.cpu()
 
 
 def apply_permutation_in_place(module, perm):
