@@ -120,12 +120,24 @@ class _TransformerLM(nn.Module):
         nn.init.normal_(self.pos_emb.weight, std=0.02)
 
     def forward(self, tokens, mask=None):
+        from ..normalization import fused_add_norm
+
         b, s = tokens.shape
         pos = torch.arange(s, device=tokens.device).unsqueeze(0)
         x = self.tok_emb(tokens) + self.pos_emb(pos)
+        # pre-LN residual stream with every add fused into the next norm:
+        # each sublayer's output is carried as `delta` and folded into the
+        # following norm's single-read kernel (z = x + delta written once)
+        delta = None
         for layer in self.layers:
-            x = layer(x, mask)
-        x = self.final_norm(x)
+            if delta is None:
+                n1 = layer.ln1(x)
+            else:
+                n1, x = fused_add_norm(x, delta, layer.ln1)
+            delta = layer.attn(n1, mask)
+            n2, x = fused_add_norm(x, delta, layer.ln2)
+            delta = layer.mlp(n2)
+        x, _ = fused_add_norm(x, delta, self.final_norm)
         # weight-tied LM head
         return torch.matmul(x, self.tok_emb.weight.t())
 

@@ -208,6 +208,92 @@ def mixed_dtype_fused_rms_norm_affine(input, weight, normalized_shape, eps=1e-6,
     return FusedRMSNormAffineMixedDtypesFunction.apply(input, weight, normalized_shape, eps, memory_efficient)
 
 
+# ----- fused residual-add + norm -----
+# Pre-LN transformers interleave `z = x + sublayer(…); y = norm(z)` — the
+# fused kernel reads x and the sublayer output once and writes both z and y,
+# saving a full re-read of z vs. eager add followed by norm. Backward needs
+# no new kernel: d(add) is identity, so dX = dResidual = LN-backward(dy, z)
+# plus the downstream gradient that arrives on z.
+
+class FusedAddLayerNormAffineFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, input, residual, weight, bias, normalized_shape, eps):
+        ext = get_ext("fused_norm")
+        ctx.normalized_shape = normalized_shape
+        ctx.eps = eps
+        input_ = input.contiguous()
+        residual_ = residual.contiguous()
+        weight_ = weight.contiguous()
+        bias_ = bias.contiguous()
+        output, z, mean, invvar = ext.forward_add_affine(
+            input_, residual_, ctx.normalized_shape, weight_, bias_, ctx.eps)
+        ctx.save_for_backward(z, weight_, bias_, mean, invvar)
+        return output, z
+
+    @staticmethod
+    def backward(ctx, grad_output, grad_z):
+        ext = get_ext("fused_norm")
+        z, weight_, bias_, mean, invvar = ctx.saved_tensors
+        grad_input, grad_weight, grad_bias = ext.backward_affine(
+            grad_output.contiguous(), mean, invvar, z,
+            ctx.normalized_shape, weight_, bias_, ctx.eps, False,
+        )
+        if grad_z is not None:
+            grad_input = grad_input + grad_z
+        return grad_input, grad_input, grad_weight, grad_bias, None, None
+
+
+class FusedAddRMSNormAffineFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, input, residual, weight, normalized_shape, eps):
+        ext = get_ext("fused_norm")
+        ctx.normalized_shape = normalized_shape
+        ctx.eps = eps
+        input_ = input.contiguous()
+        residual_ = residual.contiguous()
+        weight_ = weight.contiguous()
+        output, z, invvar = ext.rms_forward_add_affine(
+            input_, residual_, ctx.normalized_shape, weight_, ctx.eps)
+        ctx.save_for_backward(z, weight_, invvar)
+        return output, z
+
+    @staticmethod
+    def backward(ctx, grad_output, grad_z):
+        ext = get_ext("fused_norm")
+        z, weight_, invvar = ctx.saved_tensors
+        grad_input, grad_weight = ext.rms_backward_affine(
+            grad_output.contiguous(), invvar, z,
+            ctx.normalized_shape, weight_, ctx.eps, False,
+        )
+        if grad_z is not None:
+            grad_input = grad_input + grad_z
+        return grad_input, grad_input, grad_weight, None, None
+
+
+def fused_add_layer_norm_affine(input, residual, weight, bias, normalized_shape, eps=1e-6):
+    """Returns (layer_norm(input + residual), input + residual)."""
+    return FusedAddLayerNormAffineFunction.apply(input, residual, weight, bias,
+                                                 normalized_shape, eps)
+
+
+def fused_add_rms_norm_affine(input, residual, weight, normalized_shape, eps=1e-6):
+    """Returns (rms_norm(input + residual), input + residual)."""
+    return FusedAddRMSNormAffineFunction.apply(input, residual, weight, normalized_shape, eps)
+
+
+def fused_add_norm(x, delta, norm):
+    """z = x + delta; y = norm(z) through a FusedLayerNorm / FusedRMSNorm
+    module — fused on GPU, eager composition elsewhere. Returns (y, z)."""
+    if x.is_cuda and norm.elementwise_affine:
+        if isinstance(norm, FusedRMSNorm):
+            return fused_add_rms_norm_affine(x, delta, norm.weight,
+                                             norm.normalized_shape, norm.eps)
+        return fused_add_layer_norm_affine(x, delta, norm.weight, norm.bias,
+                                           norm.normalized_shape, norm.eps)
+    z = x + delta
+    return norm(z), z
+
+
 # ----- modules -----
 
 class FusedLayerNorm(torch.nn.Module):

@@ -135,6 +135,76 @@ __global__ void __launch_bounds__(LN_BLOCK) ln_fwd_wave_kernel(
   }
 }
 
+// fused residual-add + norm forward: z = x + res is written once and the
+// norm is computed from the registers — one read of each input, one write of
+// z and y, vs. separate add (2R+1W) + norm (1R+1W) passes.
+template <typename T, typename WT, bool RMS, bool AFFINE, int NPACK>
+__global__ void __launch_bounds__(LN_BLOCK) ln_add_fwd_wave_kernel(
+    const T* __restrict__ input, const T* __restrict__ residual, T* __restrict__ sum_out,
+    T* __restrict__ output, float* __restrict__ mean_out, float* __restrict__ invvar_out,
+    const WT* __restrict__ gamma, const WT* __restrict__ beta, long n1, long n2, float eps) {
+  constexpr int W = VecPack<T>::width;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * WPB + wid; row < n1; row += (long)gridDim.x * WPB) {
+    const T* x = input + row * n2;
+    const T* r = residual + row * n2;
+    T* z = sum_out + row * n2;
+    T* y = output + row * n2;
+    VecPack<T> xs[NPACK];
+    Welford w;
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        VecPack<T> rv;
+        load_pack(xs[k], x + i);
+        load_pack(rv, r + i);
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float s = to_float(xs[k].a[j]) + to_float(rv.a[j]);
+          xs[k].a[j] = from_float<T>(s);
+          w.add(to_float(xs[k].a[j]));  // stats on the stored (rounded) sum
+        }
+        store_pack(z + i, xs[k]);
+      }
+    }
+#pragma unroll
+    for (int off = WAVE_SIZE / 2; off > 0; off >>= 1) {
+      float mb = __shfl_xor(w.mean, off);
+      float m2b = __shfl_xor(w.m2, off);
+      float nb = __shfl_xor(w.count, off);
+      w.combine(mb, m2b, nb);
+    }
+    const float mean = RMS ? 0.f : w.mean;
+    const float sumsq = w.m2 + w.count * w.mean * w.mean;
+    const float invvar = RMS ? rsqrtf(sumsq / n2 + eps) : rsqrtf(w.m2 / n2 + eps);
+    if (lane == 0) {
+      if (!RMS && mean_out) mean_out[row] = mean;
+      invvar_out[row] = invvar;
+    }
+#pragma unroll
+    for (int k = 0; k < NPACK; ++k) {
+      const long i = (long)(k * WAVE_SIZE + lane) * W;
+      if (i < n2) {
+        VecPack<T> o;
+#pragma unroll
+        for (int j = 0; j < W; ++j) {
+          float xhat = (to_float(xs[k].a[j]) - mean) * invvar;
+          float rr = xhat;
+          if (AFFINE) {
+            rr = xhat * to_float(gamma[i + j]);
+            if (!RMS) rr += to_float(beta[i + j]);
+          }
+          o.a[j] = from_float<T>(rr);
+        }
+        store_pack(y + i, o);
+      }
+    }
+  }
+}
+
 // wave-per-row backward dx: dy and x(/y) rows in registers, two wave sums.
 template <typename T, typename WT, bool RMS, bool AFFINE, bool MEMEFF, int NPACK>
 __global__ void __launch_bounds__(LN_BLOCK) ln_bwd_dx_wave_kernel(
@@ -515,6 +585,79 @@ std::vector<at::Tensor> norm_fwd(const at::Tensor& input,
   return {out, mean, invvar};
 }
 
+// fused residual-add + norm forward; falls back to an eager add + the block
+// kernel for rows too wide for the wave kernel's register budget.
+template <bool RMS>
+std::vector<at::Tensor> norm_add_fwd(const at::Tensor& input, const at::Tensor& residual,
+                                     const std::vector<long>& normalized_shape,
+                                     const c10::optional<at::Tensor>& gamma,
+                                     const c10::optional<at::Tensor>& beta, double eps) {
+  long n1, n2;
+  shape_split(input, normalized_shape, n1, n2);
+  auto in = input.contiguous();
+  auto res = residual.contiguous();
+  TORCH_CHECK(res.sizes() == in.sizes() && res.scalar_type() == in.scalar_type(),
+              "residual must match input shape/dtype");
+  const bool affine = gamma.has_value();
+
+  bool wave_possible = false;
+  APEX_DISPATCH_FLOAT_HALF_BF(in.scalar_type(), "fused_norm_add_fwd_probe", ([&] {
+    constexpr int W = VecPack<scalar_t>::width;
+    wave_possible = (n2 % W == 0) && is_pack_aligned<scalar_t>(in.data_ptr()) &&
+                    is_pack_aligned<scalar_t>(res.data_ptr()) && n2 <= (long)WAVE_SIZE * W * 4;
+  }()));
+  if (!wave_possible) {
+    auto z = in + res;
+    auto out = norm_fwd<RMS>(z, normalized_shape, gamma, beta, eps);
+    out.insert(out.begin() + 1, z);
+    return out;
+  }
+
+  auto z = at::empty_like(in);
+  auto out = at::empty_like(in);
+  auto fopts = at::TensorOptions().dtype(at::kFloat).device(in.device());
+  auto mean = RMS ? at::empty({0}, fopts) : at::empty({n1}, fopts);
+  auto invvar = at::empty({n1}, fopts);
+  auto stream = current_stream();
+
+  APEX_DISPATCH_FLOAT_HALF_BF(in.scalar_type(), "fused_norm_add_fwd", ([&] {
+    using in_t = scalar_t;
+    const auto w_type = affine ? gamma->scalar_type() : in.scalar_type();
+    APEX_DISPATCH_FLOAT_HALF_BF(w_type, "fused_norm_add_fwd", ([&] {
+      using w_t = scalar_t;
+      constexpr int W = VecPack<in_t>::width;
+      const int npack = (int)((n2 + WAVE_SIZE * W - 1) / (WAVE_SIZE * W));
+      const w_t* g_ptr = affine ? (const w_t*)gamma->data_ptr() : nullptr;
+      const w_t* b_ptr = (affine && !RMS) ? (const w_t*)beta->data_ptr() : nullptr;
+      float* mean_ptr = RMS ? nullptr : mean.data_ptr<float>();
+      constexpr int WPB = LN_BLOCK / WAVE_SIZE;
+      const int grid = (int)std::min<long>((n1 + WPB - 1) / WPB, 32768);
+      auto lw = [&](auto aff, auto np_tag) {
+        constexpr bool AFF = decltype(aff)::value;
+        constexpr int NP = decltype(np_tag)::value;
+        hipLaunchKernelGGL((ln_add_fwd_wave_kernel<in_t, w_t, RMS, AFF, NP>), dim3(grid),
+                           dim3(LN_BLOCK), 0, stream, (const in_t*)in.data_ptr(),
+                           (const in_t*)res.data_ptr(), (in_t*)z.data_ptr(),
+                           (in_t*)out.data_ptr(), mean_ptr, invvar.data_ptr<float>(), g_ptr,
+                           b_ptr, n1, n2, (float)eps);
+      };
+      auto dispatch_np = [&](auto aff) {
+        switch (npack) {
+          case 1: lw(aff, std::integral_constant<int, 1>{}); break;
+          case 2: lw(aff, std::integral_constant<int, 2>{}); break;
+          case 3: lw(aff, std::integral_constant<int, 3>{}); break;
+          default: lw(aff, std::integral_constant<int, 4>{}); break;
+        }
+      };
+      if (affine) dispatch_np(std::true_type{});
+      else dispatch_np(std::false_type{});
+      HIP_CHECK(hipGetLastError());
+    }()));
+  }()));
+  if (RMS) return {out, z, invvar};
+  return {out, z, mean, invvar};
+}
+
 template <typename in_t, typename w_t, bool RMS, bool AFF, bool MEFF>
 void launch_bwd_impl(const at::Tensor& dy, const at::Tensor& io, const float* mean_ptr,
                      const at::Tensor& invvar, const w_t* g_ptr, const w_t* b_ptr,
@@ -663,6 +806,18 @@ at::Tensor backward_plain(at::Tensor grad_out, c10::optional<at::Tensor> mean, a
                          c10::nullopt, eps, memory_efficient)[0];
 }
 
+std::vector<at::Tensor> forward_add_affine(at::Tensor input, at::Tensor residual,
+                                           std::vector<long> normalized_shape, at::Tensor weight,
+                                           at::Tensor bias, double eps) {
+  return norm_add_fwd<false>(input, residual, normalized_shape, weight, bias, eps);
+}
+
+std::vector<at::Tensor> rms_forward_add_affine(at::Tensor input, at::Tensor residual,
+                                               std::vector<long> normalized_shape,
+                                               at::Tensor weight, double eps) {
+  return norm_add_fwd<true>(input, residual, normalized_shape, weight, c10::nullopt, eps);
+}
+
 std::vector<at::Tensor> rms_forward_affine(at::Tensor input, std::vector<long> normalized_shape,
                                            at::Tensor weight, double eps) {
   return norm_fwd<true>(input, normalized_shape, weight, c10::nullopt, eps);
@@ -701,6 +856,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("forward", &forward_plain, "LayerNorm fwd (no affine)");
   m.def("backward_affine", &backward_affine, "LayerNorm bwd (affine)");
   m.def("backward", &backward_plain, "LayerNorm bwd (no affine)");
+  m.def("forward_add_affine", &forward_add_affine,
+        "fused residual-add + LayerNorm fwd: returns (y, z=x+res, mean, invvar)");
+  m.def("rms_forward_add_affine", &rms_forward_add_affine,
+        "fused residual-add + RMSNorm fwd: returns (y, z=x+res, invvar)");
   m.def("rms_forward_affine", &rms_forward_affine, "RMSNorm fwd (affine)");
   m.def("rms_forward_affine_mixed_dtypes", &rms_forward_affine_mixed_dtypes,
         "RMSNorm fwd, low-precision input with fp32 params");

@@ -74,3 +74,33 @@ def test_dist_lamb_single_gpu():
     torch.cuda.synchronize()
     for p, rp in zip(model.parameters(), ref_params):
         torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_dist_adam_store_param_remainders_gpu():
+    """Implicit (bf16 bits << 16 | int16) master must track the stored fp32
+    master bit-exactly through the multi_tensor_adam kernel path."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    def build(remainders):
+        torch.manual_seed(4)
+        m = torch.nn.Sequential(torch.nn.Linear(64, 128), torch.nn.Tanh(),
+                                torch.nn.Linear(128, 8)).cuda().bfloat16()
+        o = DistributedFusedAdam(m.parameters(), lr=1e-3, weight_decay=0.01,
+                                 bucket_cap_mb=1, store_param_remainders=remainders)
+        return m, o
+
+    m0, o0 = build(False)
+    m1, o1 = build(True)
+    assert o1.buckets[0].master_shard is None
+    for it in range(6):
+        for mm, oo in ((m0, o0), (m1, o1)):
+            torch.manual_seed(100 + it)
+            for p in mm.parameters():
+                p.grad = torch.randn_like(p)
+            for p in mm.parameters():
+                oo._grad_copy(p)
+            oo.step()
+    torch.cuda.synchronize()
+    for b0, b1 in zip(o0.buckets, o1.buckets):
+        assert torch.equal(o0._get_master(b0), o1._get_master(b1))
+        assert torch.equal(b0.exp_avg, b1.exp_avg)

@@ -131,3 +131,67 @@ def test_layer_norm_non_contiguous_input():
     y = ln(x)
     y_ref = torch.nn.functional.layer_norm(x.contiguous(), (128,), ln.weight, ln.bias, ln.eps)
     torch.testing.assert_close(y, y_ref, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("hidden", [768, 1024, 2048, 3000])
+def test_fused_add_layer_norm(hidden):
+    """Fused z = x + res; y = LN(z) vs eager fp32 composition, fwd + bwd
+    including a downstream gradient arriving on z. hidden=3000 exercises the
+    non-wave fallback path."""
+    from apex_amd.normalization import fused_add_layer_norm_affine
+
+    torch.manual_seed(7)
+    n1 = 64
+    x = torch.randn(n1, hidden, device="cuda", requires_grad=True)
+    r = torch.randn(n1, hidden, device="cuda", requires_grad=True)
+    w = torch.randn(hidden, device="cuda", requires_grad=True)
+    b = torch.randn(hidden, device="cuda", requires_grad=True)
+    y, z = fused_add_layer_norm_affine(x, r, w, b, (hidden,), 1e-5)
+
+    xr = x.detach().clone().requires_grad_(True)
+    rr = r.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    z_ref = xr + rr
+    y_ref = torch.nn.functional.layer_norm(z_ref, (hidden,), wr, br, 1e-5)
+
+    torch.testing.assert_close(z, z_ref, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(y, y_ref, rtol=1e-4, atol=1e-4)
+
+    dy = torch.randn_like(y)
+    dz = torch.randn_like(z)
+    (y * dy + z * dz).sum().backward()
+    (y_ref * dy + z_ref * dz).sum().backward()
+    torch.testing.assert_close(x.grad, xr.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(r.grad, rr.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(w.grad, wr.grad, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(b.grad, br.grad, rtol=1e-3, atol=1e-3)
+
+
+def test_fused_add_rms_norm():
+    from apex_amd.normalization import fused_add_rms_norm_affine
+
+    torch.manual_seed(8)
+    hidden = 1024
+    x = torch.randn(32, hidden, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    r = torch.randn(32, hidden, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(hidden, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y, z = fused_add_rms_norm_affine(x, r, w, (hidden,), 1e-5)
+
+    zf = (x.detach().float() + r.detach().float())
+    y_ref = zf * torch.rsqrt(zf.pow(2).mean(-1, keepdim=True) + 1e-5) * w.detach().float()
+    torch.testing.assert_close(z.float(), zf, rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(y.float(), y_ref, rtol=3e-2, atol=3e-2)
+    y.float().sum().backward()
+    assert torch.isfinite(x.grad).all() and torch.isfinite(w.grad).all()
+
+
+def test_fused_add_norm_module_dispatch():
+    from apex_amd.normalization import FusedLayerNorm, fused_add_norm
+
+    ln = FusedLayerNorm(512).cuda()
+    x = torch.randn(16, 512, device="cuda")
+    d = torch.randn(16, 512, device="cuda")
+    y, z = fused_add_norm(x, d, ln)
+    torch.testing.assert_close(z, x + d, rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(y, ln(x + d), rtol=1e-5, atol=1e-5)
