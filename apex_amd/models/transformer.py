@@ -181,12 +181,19 @@ class TransformerLargeModel(nn.Module):
         nn.init.normal_(self.pos_emb.weight, std=0.02)
 
     def forward(self, tokens):
+        from ..normalization import fused_add_norm
+
         b, s = tokens.shape
         pos = torch.arange(s, device=tokens.device).unsqueeze(0)
         x = (self.tok_emb(tokens) + self.pos_emb(pos)).transpose(0, 1)  # [s, b, h]
+        delta = None  # residual adds fused into the next norm's kernel
         for attn, l1, l2, mlp in zip(self.attns, self.ln1, self.ln2, self.mlps):
-            a, _ = attn(l1(x), attn_mask="causal")
-            x = x + a
-            x = x + mlp(l2(x))
-        x = self.final_norm(x).transpose(0, 1)
-        return torch.matmul(x, self.tok_emb.weight.t())
+            if delta is None:
+                n1 = l1(x)
+            else:
+                n1, x = fused_add_norm(x, delta, l1)
+            delta, _ = attn(n1, attn_mask="causal")
+            n2, x = fused_add_norm(x, delta, l2)
+            delta = mlp(n2)
+        x, _ = fused_add_norm(x, delta, self.final_norm)
+        return torch.matmul(x.transpose(0, 1), self.tok_emb.weight.t())
