@@ -39,7 +39,9 @@ from .scaler import LossScaler
 from .._ext import get_ext
 from ..multi_tensor_apply import multi_tensor_applier
 
-__all__ = ["initialize", "scale_loss", "master_params", "state_dict", "load_state_dict"]
+__all__ = ["initialize", "scale_loss", "master_params", "state_dict", "load_state_dict",
+           "half_function", "float_function", "promote_function",
+           "register_half_function", "register_float_function", "register_promote_function"]
 
 
 class OptProperties:
@@ -316,11 +318,13 @@ def initialize(
 
     # --- models ---
     if props["cast_model_type"] is not None:  # O2 / O3
+        _amp_state.cast_dtype = props["cast_model_type"]
         for model in model_list:
             _cast_model(model, props["cast_model_type"], props["keep_batchnorm_fp32"])
             _wrap_forward_input_cast(model, props["cast_model_type"])
     elif props["patch_torch_functions"]:  # O1
         cast_dtype = torch.float16 if cast_model_type is None else cast_model_type
+        _amp_state.cast_dtype = cast_dtype
         for model in model_list:
             _wrap_forward_autocast(model, cast_dtype)
     if cast_model_outputs is not None:
@@ -355,6 +359,87 @@ def initialize(
         model_list if models_was_list else model_list[0],
         optimizer_list if optimizers_was_list else optimizer_list[0],
     )
+
+
+# --- function-cast registry (reference amp.half_function & friends) ---
+# For user code that runs OUTSIDE the autocast-wrapped forward (custom loss
+# functions, metrics): explicitly pin a function's compute dtype. Wrappers
+# disable autocast inside so the pinned dtype is authoritative.
+
+def _cast_tree_to(x, dtype):
+    if torch.is_tensor(x) and x.is_floating_point():
+        return x.to(dtype)
+    if isinstance(x, (list, tuple)):
+        return type(x)(_cast_tree_to(v, dtype) for v in x)
+    if isinstance(x, dict):
+        return {k: _cast_tree_to(v, dtype) for k, v in x.items()}
+    return x
+
+
+def _collect_float_dtypes(x, out):
+    if torch.is_tensor(x) and x.is_floating_point():
+        out.append(x.dtype)
+    elif isinstance(x, (list, tuple)):
+        for v in x:
+            _collect_float_dtypes(v, out)
+    elif isinstance(x, dict):
+        for v in x.values():
+            _collect_float_dtypes(v, out)
+
+
+def _wrap_cast(fn, pick_dtype):
+    @functools.wraps(fn)
+    def wrapper(*args, **kwargs):
+        dtype = pick_dtype(args, kwargs)
+        with contextlib.ExitStack() as stack:
+            for dev in ("cuda", "cpu"):
+                stack.enter_context(torch.autocast(device_type=dev, enabled=False))
+            return fn(*_cast_tree_to(args, dtype), **_cast_tree_to(kwargs, dtype))
+    return wrapper
+
+
+def half_function(fn):
+    """Run ``fn`` with floating inputs cast to the amp low-precision dtype
+    (the ``cast_model_type`` passed to initialize; fp16 default)."""
+    return _wrap_cast(fn, lambda a, k: getattr(_amp_state, "cast_dtype", None) or torch.float16)
+
+
+def float_function(fn):
+    """Run ``fn`` with floating inputs cast to fp32."""
+    return _wrap_cast(fn, lambda a, k: torch.float32)
+
+
+def promote_function(fn):
+    """Run ``fn`` with floating inputs cast to the widest floating dtype
+    present among them (fp32 wins over bf16/fp16)."""
+    def pick(args, kwargs):
+        seen = []
+        _collect_float_dtypes(args, seen)
+        _collect_float_dtypes(kwargs, seen)
+        if not seen:
+            return torch.float32
+        if torch.float64 in seen:
+            return torch.float64
+        if torch.float32 in seen:
+            return torch.float32
+        if torch.bfloat16 in seen and torch.float16 in seen:
+            return torch.float32  # no safe common low dtype; promote
+        return seen[0]
+    return _wrap_cast(fn, pick)
+
+
+def register_half_function(module, name):
+    """Patch ``module.name`` in place with :func:`half_function` (effective
+    immediately — no need to call before ``initialize``)."""
+    setattr(module, name, half_function(getattr(module, name)))
+
+
+def register_float_function(module, name):
+    setattr(module, name, float_function(getattr(module, name)))
+
+
+def register_promote_function(module, name):
+    setattr(module, name, promote_function(getattr(module, name)))
 
 
 @contextlib.contextmanager

@@ -11,6 +11,7 @@ class AmpState:
         self.optimizers = []
         self.models = []
         self.verbosity = 1
+        self.cast_dtype = None  # set by initialize; used by half_function
 
     def reset(self):
         self.__init__()

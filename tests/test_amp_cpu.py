@@ -183,3 +183,38 @@ def test_dynamic_scaler_single_tick_per_step():
         s.unscale_grads(g, g)                       # main tick
         s.unscale_grads(g, g, scale_override=2.0 ** 10)  # extra set: no tick
     assert s.loss_scale() == 2.0 ** 11  # exactly one growth after 3 steps
+
+
+def test_amp_function_cast_registry():
+    model = make_model()
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    amp.initialize(model, opt, opt_level="O1", cast_model_type=torch.bfloat16, verbosity=0)
+
+    seen = {}
+
+    def my_loss(pred, target):
+        seen["dtype"] = pred.dtype
+        return (pred - target).pow(2).mean()
+
+    half_loss = amp.half_function(my_loss)
+    out = half_loss(torch.randn(4, 8), torch.randn(4, 8))
+    assert seen["dtype"] == torch.bfloat16
+    assert out.dtype == torch.bfloat16
+
+    float_loss = amp.float_function(my_loss)
+    float_loss(torch.randn(4, 8).bfloat16(), torch.randn(4, 8).bfloat16())
+    assert seen["dtype"] == torch.float32
+
+    promo = amp.promote_function(my_loss)
+    promo(torch.randn(4, 8).bfloat16(), torch.randn(4, 8))  # bf16 + fp32 -> fp32
+    assert seen["dtype"] == torch.float32
+    promo(torch.randn(4, 8).bfloat16(), torch.randn(4, 8).bfloat16())
+    assert seen["dtype"] == torch.bfloat16
+
+
+def test_amp_register_function_patches_module():
+    import types as _types
+
+    ns = _types.SimpleNamespace(f=lambda x: x.dtype)
+    amp.register_float_function(ns, "f")
+    assert ns.f(torch.randn(2).bfloat16()) == torch.float32
