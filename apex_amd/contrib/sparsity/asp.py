@@ -45,11 +45,7 @@ class ASP:
         assert cls.__model is None, "ASP has been initialized already"
         cls.__model = model
         cls.__allow_recompute = allow_recompute_mask
-        if allow_permutation:
-            warnings.warn(
-                "apex_amd ASP: channel-permutation search is not implemented yet; "
-                "pruning without permutation."
-            )
+        cls.__allow_permutation = allow_permutation
 
         if isinstance(mask_calculator, str):
             from .sparse_masklib import create_mask
@@ -94,6 +90,12 @@ class ASP:
 
     @classmethod
     def compute_sparse_masks(cls):
+        if getattr(cls, "_ASP__allow_permutation", False):
+            # re-parameterize first: permute channel groups so the 2:4 masks
+            # keep more magnitude (model function unchanged)
+            from .permutation_lib import permute_model_for_sparsity
+
+            permute_model_for_sparsity(cls.__model)
         with torch.no_grad():
             for _, mod, p_name, p, buf_name in cls.__sparse_parameters:
                 mask = cls.__calculate_mask(p)
@@ -125,3 +127,4 @@ class ASP:
         cls.__optimizer = None
         cls.__sparse_parameters = []
         cls.__calculate_mask = None
+        cls.__allow_permutation = False

@@ -177,3 +177,83 @@ def test_exhaustive_permutation_search():
     # monotone: re-running on the permuted matrix cannot reduce efficacy
     perm3 = exhaustive_search(w[:, perm])
     assert efficacy(w[:, perm][:, perm3]) >= improved - 1e-5
+
+
+def test_permutation_propagation_preserves_function():
+    """FX-walked group permutation is a pure re-parameterization: model
+    outputs are unchanged while 2:4 kept magnitude improves."""
+    from apex_amd.contrib.sparsity.permutation_lib import (
+        find_permutation_groups, permute_model_for_sparsity,
+    )
+    from apex_amd.contrib.sparsity.permutation_search import efficacy
+
+    class Res(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin1 = torch.nn.Linear(16, 32)
+            self.ln = torch.nn.LayerNorm(32)
+            self.lin2 = torch.nn.Linear(32, 32)
+            self.lin3 = torch.nn.Linear(32, 8)
+
+        def forward(self, x):
+            h = self.ln(self.lin1(x))
+            h = h + torch.relu(self.lin2(h))
+            return self.lin3(h)
+
+    torch.manual_seed(11)
+    m = Res()
+    with torch.no_grad():  # make some channels dominant so permutation helps
+        m.lin1.weight[:4] *= 10
+    groups = find_permutation_groups(m)
+    assert len(groups) == 1
+    g = groups[0]
+    assert len(g.consumers) == 2 and len(g.producers) == 2 and len(g.norms) == 1
+
+    x = torch.randn(8, 16)
+    before_out = m(x)
+    before_eff = efficacy(m.lin3.weight.detach()) + efficacy(m.lin2.weight.detach())
+    n = permute_model_for_sparsity(m)
+    assert n == 1
+    after_out = m(x)
+    after_eff = efficacy(m.lin3.weight.detach()) + efficacy(m.lin2.weight.detach())
+    torch.testing.assert_close(after_out, before_out, rtol=1e-4, atol=1e-5)
+    assert after_eff >= before_eff - 1e-4
+
+
+def test_permutation_propagation_rejects_leaky_group():
+    """If the stream feeding a consumer is also returned directly, permuting
+    it would change the model output — the group must be dropped."""
+    from apex_amd.contrib.sparsity.permutation_lib import find_permutation_groups
+
+    class Leaky(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin1 = torch.nn.Linear(16, 32)
+            self.lin2 = torch.nn.Linear(32, 8)
+
+        def forward(self, x):
+            h = self.lin1(x)
+            return self.lin2(h), h  # h escapes un-permuted
+
+    assert find_permutation_groups(Leaky()) == []
+
+
+def test_asp_allow_permutation_end_to_end():
+    from apex_amd.contrib.sparsity import ASP
+
+    torch.manual_seed(2)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 8))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    x = torch.randn(4, 16)
+    before = model(x)
+    ASP._reset()
+    ASP.init_model_for_pruning(model, mask_calculator="m4n2_1d", verbosity=0,
+                               allow_permutation=True)
+    ASP.init_optimizer_for_pruning(opt)
+    ASP.compute_sparse_masks()
+    # weights are now 2:4 masked; each 4-group along dim1 has >=2 zeros
+    w = model[2].weight.detach()
+    zeros_per_group = (w.reshape(w.shape[0], -1, 4) == 0).sum(-1)
+    assert (zeros_per_group >= 2).all()
+    ASP._reset()
