@@ -79,6 +79,12 @@ class DistributedFusedAdam(torch.optim.Optimizer):
     param is the truncated top half of the master rather than
     round-to-nearest; the optimizer trajectory (which reads the exact
     master) is unchanged.
+
+    ``with_scaled_states`` stores the Adam moments as fp16 with one fp32
+    scale per bucket shard (reference: with_scaled_states) — 2 bytes per
+    moment element instead of 4. Moments are rescaled to fp32 around each
+    step; exp_avg_sq (non-negative, huge dynamic range) is stored as
+    sqrt(v) so fp16's ~2^-24..2^15 span covers v down to ~1e-14.
     """
 
     def __init__(
@@ -98,6 +104,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         process_group=None,
         set_grad_none=True,
         store_param_remainders=False,
+        with_scaled_states=False,
     ):
         if amsgrad:
             raise RuntimeError("DistributedFusedAdam does not support AMSGrad")
@@ -114,6 +121,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
         self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
         self.store_param_remainders = store_param_remainders
+        self.with_scaled_states = with_scaled_states
         self._step = 0
 
         device = self.param_groups[0]["params"][0].device
@@ -171,8 +179,16 @@ class DistributedFusedAdam(torch.optim.Optimizer):
                                             device=self.device)
         else:
             b.master_shard = b.param_data[lo:hi].float().clone()
-        b.exp_avg = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
-        b.exp_avg_sq = torch.zeros_like(b.exp_avg)
+        if self.with_scaled_states:
+            b.exp_avg_q = torch.zeros(b.shard_size, dtype=torch.float16, device=self.device)
+            b.exp_avg_sq_q = torch.zeros_like(b.exp_avg_q)  # stores sqrt(v)
+            b.m_scale = 1.0
+            b.v_scale = 1.0
+            b.exp_avg = None
+            b.exp_avg_sq = None
+        else:
+            b.exp_avg = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
+            b.exp_avg_sq = torch.zeros_like(b.exp_avg)
         b.grad_shard = torch.zeros(b.shard_size, dtype=torch.float32, device=self.device)
         self.buckets.append(b)
 
@@ -201,6 +217,31 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         low = bits & 0xFFFF
         low = low - ((low >> 15) << 16)  # map [32768,65535] -> negative int16
         b.param_remainder.copy_(low.to(torch.int16))
+
+    _F16_HEADROOM = 60000.0  # fp16 max is 65504; leave growth headroom
+
+    def _get_moments(self, b):
+        """(exp_avg, exp_avg_sq) as fp32 — stored directly, or dequantized
+        from the per-shard-scaled fp16 representation."""
+        if not self.with_scaled_states:
+            return b.exp_avg, b.exp_avg_sq
+        m = b.exp_avg_q.float() * b.m_scale
+        s = b.exp_avg_sq_q.float() * b.v_scale
+        return m, s * s
+
+    def _set_moments(self, b, m, v):
+        if not self.with_scaled_states:
+            if m is not b.exp_avg:
+                b.exp_avg.copy_(m)
+                b.exp_avg_sq.copy_(v)
+            return
+        m_max = float(m.abs().max())
+        b.m_scale = (m_max / self._F16_HEADROOM) if m_max > 0 else 1.0
+        b.exp_avg_q.copy_((m / b.m_scale).half())
+        s = v.sqrt()
+        s_max = float(s.max())
+        b.v_scale = (s_max / self._F16_HEADROOM) if s_max > 0 else 1.0
+        b.exp_avg_sq_q.copy_((s / b.v_scale).half())
 
     def _register_hooks(self):
         self._hook_handles = []
@@ -313,16 +354,19 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             beta1, beta2 = group["betas"]
             bias_correction = 1 if group["bias_correction"] else 0
             master = self._get_master(b)
+            exp_avg, exp_avg_sq = self._get_moments(b)
             if self.device.type == "cuda":
                 amp_C = get_ext("amp_C")
                 multi_tensor_applier(
                     amp_C.multi_tensor_adam, self._noop,
-                    [[b.grad_shard], [master], [b.exp_avg], [b.exp_avg_sq]],
+                    [[b.grad_shard], [master], [exp_avg], [exp_avg_sq]],
                     group["lr"], beta1, beta2, group["eps"], self._step,
                     self.adam_w_mode, bias_correction, group["weight_decay"],
                 )
             else:
-                self._adam_ref(group, bias_correction, beta1, beta2, b, master)
+                self._adam_ref(group, bias_correction, beta1, beta2, b, master,
+                               exp_avg, exp_avg_sq)
+            self._set_moments(b, exp_avg, exp_avg_sq)
             self._set_master(b, master)
 
         # param sync: shard -> param bucket, then all_gather. On GPU the
@@ -363,7 +407,8 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             b.sync_work = None
         return loss
 
-    def _adam_ref(self, group, bias_correction, beta1, beta2, b, master):
+    def _adam_ref(self, group, bias_correction, beta1, beta2, b, master,
+                  exp_avg, exp_avg_sq):
         step = self._step
         bc1 = 1 - beta1 ** step if bias_correction else 1.0
         bc2 = 1 - beta2 ** step if bias_correction else 1.0
@@ -371,9 +416,9 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         g = b.grad_shard
         if self.adam_w_mode == 0 and wd != 0:
             g = g + wd * master
-        b.exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
-        b.exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
-        update = (b.exp_avg / bc1) / ((b.exp_avg_sq / bc2).sqrt() + eps)
+        exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+        exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        update = (exp_avg / bc1) / ((exp_avg_sq / bc2).sqrt() + eps)
         if self.adam_w_mode == 1 and wd != 0:
             update = update + wd * master
         master.add_(update, alpha=-lr)
@@ -391,10 +436,11 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         if gather_on_root:
             buckets_sd = []
             for b in self.buckets:
+                m, v = self._get_moments(b)
                 full = {}
                 for name, shard in (("master", self._get_master(b)),
-                                    ("exp_avg", b.exp_avg),
-                                    ("exp_avg_sq", b.exp_avg_sq)):
+                                    ("exp_avg", m),
+                                    ("exp_avg_sq", v)):
                     if self.world_size > 1:
                         chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
                         dist.all_gather(chunks, shard.contiguous(),
@@ -411,11 +457,8 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             "world_size": self.world_size,
             "param_groups": groups_sd,
             "buckets": [
-                {
-                    "master_shard": self._get_master(b),
-                    "exp_avg": b.exp_avg,
-                    "exp_avg_sq": b.exp_avg_sq,
-                }
+                dict(zip(("master_shard", "exp_avg", "exp_avg_sq"),
+                         (self._get_master(b),) + self._get_moments(b)))
                 for b in self.buckets
             ],
         }
@@ -431,27 +474,25 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             for b, bsd in zip(self.buckets, sd["buckets"]):
                 lo = self.rank * b.shard_size
                 hi = lo + b.shard_size
-                for name, dst in (("master", None), ("exp_avg", b.exp_avg),
-                                  ("exp_avg_sq", b.exp_avg_sq)):
+                shards = {}
+                for name in ("master", "exp_avg", "exp_avg_sq"):
                     full = bsd[name].to(self.device, torch.float32)
                     pad = b.numel - full.numel()
                     if pad:
                         full = torch.cat([full, full.new_zeros(pad)])
-                    if name == "master":
-                        self._set_master(b, full[lo:hi].contiguous())
-                        if b.master_shard is not None:
-                            b.param_data[lo:hi].copy_(
-                                b.master_shard.to(b.param_data.dtype))
-                    else:
-                        dst.copy_(full[lo:hi])
+                    shards[name] = full[lo:hi].contiguous()
+                self._set_master(b, shards["master"])
+                if b.master_shard is not None:
+                    b.param_data[lo:hi].copy_(b.master_shard.to(b.param_data.dtype))
+                self._set_moments(b, shards["exp_avg"], shards["exp_avg_sq"])
         else:
             assert sd["world_size"] == self.world_size, (
                 "world size changed; use a gather_on_root=True checkpoint to reshard"
             )
             for b, bsd in zip(self.buckets, sd["buckets"]):
                 self._set_master(b, bsd["master_shard"].to(self.device, torch.float32))
-                b.exp_avg.copy_(bsd["exp_avg"])
-                b.exp_avg_sq.copy_(bsd["exp_avg_sq"])
+                self._set_moments(b, bsd["exp_avg"].to(self.device, torch.float32),
+                                  bsd["exp_avg_sq"].to(self.device, torch.float32))
                 if b.master_shard is not None:
                     lo = self.rank * b.shard_size
                     b.param_data[lo:lo + b.shard_size].copy_(

@@ -34,8 +34,9 @@ class DistributedFusedLAMB(DistributedFusedAdam):
         clip_after_ar=True,
         **kwargs,
     ):
-        if kwargs.get("store_param_remainders"):
-            raise ValueError("DistributedFusedLAMB does not support store_param_remainders")
+        for unsupported in ("store_param_remainders", "with_scaled_states"):
+            if kwargs.get(unsupported):
+                raise ValueError(f"DistributedFusedLAMB does not support {unsupported}")
         super().__init__(
             params, lr=lr, bias_correction=bias_correction, betas=betas, eps=eps,
             adam_w_mode=adam_w_mode, weight_decay=weight_decay, **kwargs,

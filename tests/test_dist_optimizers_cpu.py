@@ -252,3 +252,55 @@ def test_dist_adam_gather_on_root_resharding(tmp_path):
         torch.testing.assert_close(p.detach(), ref)
     for b, bsd in zip(opt.buckets, saved["sd"]["buckets"]):
         torch.testing.assert_close(opt._get_master(b)[:b.numel_unpadded], bsd["master"])
+
+
+def test_dist_adam_scaled_states_tracks_fp32():
+    """fp16 scaled moments (sqrt storage for v) track the fp32-state
+    trajectory closely over many steps."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    models, opts = [], []
+    for scaled in (False, True):
+        m = _make_model()
+        opt = DistributedFusedAdam(m.parameters(), lr=1e-3, weight_decay=0.01,
+                                   bucket_cap_mb=1, with_scaled_states=scaled)
+        models.append(m)
+        opts.append(opt)
+    assert opts[1].buckets[0].exp_avg is None
+    assert opts[1].buckets[0].exp_avg_q.dtype == torch.float16
+    for it in range(10):
+        for m, o in zip(models, opts):
+            _feed_synthetic_grads(o, m, it)
+            o.step()
+    for p, q in zip(models[0].parameters(), models[1].parameters()):
+        torch.testing.assert_close(p.detach(), q.detach(), rtol=2e-3, atol=2e-5)
+    for b0, b1 in zip(opts[0].buckets, opts[1].buckets):
+        m0, v0 = opts[0]._get_moments(b0)
+        m1, v1 = opts[1]._get_moments(b1)
+        # fp16 quantization error is relative to the per-shard max (~2^-11)
+        torch.testing.assert_close(m0, m1, rtol=2e-3,
+                                   atol=float(m0.abs().max()) * 1.5e-3)
+        torch.testing.assert_close(v0, v1, rtol=4e-3,
+                                   atol=float(v0.max()) * 2e-3)
+
+
+def test_dist_adam_scaled_states_state_dict():
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    m = _make_model()
+    opt = DistributedFusedAdam(m.parameters(), lr=1e-3, bucket_cap_mb=1,
+                               with_scaled_states=True)
+    for it in range(3):
+        _feed_synthetic_grads(opt, m, it)
+        opt.step()
+    sd = opt.state_dict()
+    assert sd["buckets"][0]["exp_avg"].dtype == torch.float32  # portable ckpt
+
+    m2 = _make_model()
+    opt2 = DistributedFusedAdam(m2.parameters(), lr=1e-3, bucket_cap_mb=1,
+                                with_scaled_states=True)
+    opt2.load_state_dict(sd)
+    for b, b2 in zip(opt.buckets, opt2.buckets):
+        for a, c in zip(opt._get_moments(b), opt2._get_moments(b2)):
+            torch.testing.assert_close(a, c)
+        assert torch.equal(opt._get_master(b), opt2._get_master(b2))
