@@ -102,6 +102,8 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         average_grad_sync=True,
         grad_sync_dtype=None,
         process_group=None,
+        distributed_process_group=None,
+        redundant_process_group=None,
         set_grad_none=True,
         store_param_remainders=False,
         with_scaled_states=False,
@@ -116,9 +118,20 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self.set_grad_none = set_grad_none
         self.overlap_grad_sync = overlap_grad_sync
         self.average_grad_sync = average_grad_sync
-        self.process_group = process_group
-        self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
-        self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
+        # 2-D process grid (reference: distributed_process_group /
+        # redundant_process_group): optimizer state is SHARDED over the
+        # distributed group and REPLICATED over the redundant group (e.g.
+        # shard within a node's xGMI island, replicate across nodes —
+        # collectives that must cross the slow axis then move only the
+        # already-reduced shard). Defaults collapse to a flat 1-D grid.
+        self.process_group = distributed_process_group if distributed_process_group is not None \
+            else process_group
+        self.redundant_process_group = redundant_process_group
+        self.world_size = dist.get_world_size(self.process_group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(self.process_group) if dist.is_initialized() else 0
+        self.redundant_size = (dist.get_world_size(redundant_process_group)
+                               if (redundant_process_group is not None and dist.is_initialized())
+                               else 1)
         self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
         self.store_param_remainders = store_param_remainders
         self.with_scaled_states = with_scaled_states
@@ -270,7 +283,14 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
     # ---------- grad sync ----------
     def _start_bucket_grad_sync(self, b):
-        if self.world_size == 1:
+        if self.world_size == 1 and self.redundant_process_group is None:
+            b.grad_shard.copy_(b.grad_data.float())
+            b.synced = True
+            return
+        if self.world_size == 1:  # pure replication: reduce across replicas only
+            if self.average_grad_sync:
+                b.grad_data.div_(self.redundant_size)
+            dist.all_reduce(b.grad_data, group=self.redundant_process_group)
             b.grad_shard.copy_(b.grad_data.float())
             b.synced = True
             return
@@ -283,7 +303,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
     def _issue_grad_collective(self, b):
         if self.average_grad_sync:
-            b.grad_data.div_(self.world_size)
+            b.grad_data.div_(self.world_size * self.redundant_size)
         if _backend_supports_rs(self.process_group):
             shard = b.grad_data[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
             b.sync_work = dist.reduce_scatter_tensor(
@@ -301,6 +321,9 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             self._start_bucket_grad_sync(b)
         if b.sync_work is not None:
             b.sync_work.wait()
+            if self.redundant_process_group is not None:
+                # cross-replica reduction of the (small) local shard only
+                dist.all_reduce(b._sync_shard, group=self.redundant_process_group)
             if self._use_stream:
                 torch.cuda.current_stream().wait_stream(self._comm_stream)
             b.grad_shard.copy_(b._sync_shard.float())

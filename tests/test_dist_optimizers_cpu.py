@@ -304,3 +304,51 @@ def test_dist_adam_scaled_states_state_dict():
         for a, c in zip(opt._get_moments(b), opt2._get_moments(b2)):
             torch.testing.assert_close(a, c)
         assert torch.equal(opt._get_master(b), opt2._get_master(b2))
+
+
+def _grid_worker(rank, world_size):
+    """2x2 grid: shard over {0,1}/{2,3}, replicate over {0,2}/{1,3}. Params
+    must follow AdamW on the 4-rank mean gradient, and replicas must agree."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    shard_groups = [dist.new_group([0, 1]), dist.new_group([2, 3])]
+    red_groups = [dist.new_group([0, 2]), dist.new_group([1, 3])]
+    my_shard = shard_groups[rank // 2]
+    my_red = red_groups[rank % 2]
+
+    model = _make_model()
+    ref_params = [p.detach().clone().requires_grad_(True) for p in model.parameters()]
+    opt = DistributedFusedAdam(
+        model.parameters(), lr=1e-3, weight_decay=0.01, bucket_cap_mb=1,
+        distributed_process_group=my_shard, redundant_process_group=my_red,
+    )
+    assert opt.world_size == 2 and opt.redundant_size == 2
+    ref_opt = torch.optim.AdamW(ref_params, lr=1e-3, weight_decay=0.01)
+
+    for it in range(5):
+        mean_grads = []
+        for i, p in enumerate(model.parameters()):
+            contribs = []
+            for r in range(world_size):
+                torch.manual_seed(7000 * it + 13 * i + r)
+                contribs.append(torch.randn(p.shape))
+            p.grad = contribs[rank].clone()
+            mean_grads.append(torch.stack(contribs).mean(0))
+        for p in model.parameters():
+            opt._grad_copy(p)
+        opt.step()
+        for p, g in zip(ref_params, mean_grads):
+            p.grad = g
+        ref_opt.step()
+        for p, rp in zip(model.parameters(), ref_params):
+            torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-5, atol=1e-6)
+    # replicas hold identical params
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    flats = [torch.empty_like(flat) for _ in range(world_size)]
+    dist.all_gather(flats, flat)
+    for f in flats[1:]:
+        torch.testing.assert_close(flats[0], f)
+
+
+def test_dist_adam_2d_process_grid():
+    run_distributed(_grid_worker, world_size=4)
