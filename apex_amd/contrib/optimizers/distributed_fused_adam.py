@@ -107,6 +107,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         set_grad_none=True,
         store_param_remainders=False,
         with_scaled_states=False,
+        nccl_ub=False,
     ):
         if amsgrad:
             raise RuntimeError("DistributedFusedAdam does not support AMSGrad")
@@ -143,7 +144,19 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self._comm_stream = torch.cuda.Stream() if self._use_stream else None
         self._noop = torch.zeros(1, dtype=torch.int32, device=device)
 
-        self._build_buckets(grad_sync_dtype)
+        # nccl_ub: place the flat buckets in an ncclMemAlloc-backed pool so
+        # RCCL registers them as user buffers (zero-copy collectives over
+        # xGMI; reference: distributed_fused_adam nccl_ub)
+        self.nccl_ub = nccl_ub
+        self._mem_pool = None
+        if nccl_ub and self.device.type == "cuda":
+            from ..nccl_allocator import create_nccl_mem_pool, nccl_mem
+
+            self._mem_pool = create_nccl_mem_pool()
+            with nccl_mem(self._mem_pool):
+                self._build_buckets(grad_sync_dtype)
+        else:
+            self._build_buckets(grad_sync_dtype)
         self._register_hooks()
 
     # ---------- setup ----------

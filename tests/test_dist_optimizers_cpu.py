@@ -352,3 +352,15 @@ def _grid_worker(rank, world_size):
 
 def test_dist_adam_2d_process_grid():
     run_distributed(_grid_worker, world_size=4)
+
+
+def test_dist_adam_nccl_ub_flag_cpu_noop():
+    """nccl_ub routes bucket allocation through the RCCL mem pool on GPU;
+    on CPU it must be a clean no-op (pool only exists with a device)."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    m = _make_model()
+    opt = DistributedFusedAdam(m.parameters(), lr=1e-3, bucket_cap_mb=1, nccl_ub=True)
+    assert opt._mem_pool is None
+    _feed_synthetic_grads(opt, m, 0)
+    opt.step()
