@@ -99,6 +99,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         amsgrad=False,
         bucket_cap_mb=64,
         overlap_grad_sync=True,
+        overlap_param_sync=False,
         average_grad_sync=True,
         grad_sync_dtype=None,
         process_group=None,
@@ -118,6 +119,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self.adam_w_mode = 1 if adam_w_mode else 0
         self.set_grad_none = set_grad_none
         self.overlap_grad_sync = overlap_grad_sync
+        self.overlap_param_sync = overlap_param_sync
         self.average_grad_sync = average_grad_sync
         # 2-D process grid (reference: distributed_process_group /
         # redundant_process_group): optimizer state is SHARDED over the
@@ -348,6 +350,42 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         for b in self.buckets:
             self._finish_bucket_grad_sync(b)
 
+    # ---------- lazy param sync (overlap_param_sync) ----------
+    def _finish_param_sync_bucket(self, b):
+        work = getattr(b, "param_sync_work", None)
+        if work is None:
+            return
+        work.wait()
+        chunks = getattr(b, "_param_chunks", None)
+        if chunks is not None:  # gloo fallback gathered into a list
+            for r, c in enumerate(chunks):
+                b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
+            b._param_chunks = None
+        if self._use_stream:
+            torch.cuda.current_stream().wait_stream(self._comm_stream)
+        b.param_sync_work = None
+
+    def register_model_for_param_sync(self, model):
+        """With ``overlap_param_sync=True``: hook each module's pre-forward
+        to wait only for the all-gathers of the buckets holding its own
+        params — the next forward's early layers run while later buckets are
+        still in flight (reference: overlap_param_sync + param-sync hooks)."""
+        for module in model.modules():
+            buckets = {}
+            for p in module.parameters(recurse=False):
+                if p in self.param_to_bucket:
+                    b = self.param_to_bucket[p][0]
+                    buckets[id(b)] = b
+            if buckets:
+                module.register_forward_pre_hook(
+                    self._make_param_sync_hook(list(buckets.values())))
+
+    def _make_param_sync_hook(self, buckets):
+        def hook(module, inputs):
+            for b in buckets:
+                self._finish_param_sync_bucket(b)
+        return hook
+
     # ---------- norms / clipping ----------
     def grad_norm(self):
         """Global L2 norm over the sharded (already reduced) gradients."""
@@ -376,6 +414,11 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+
+        # safety net: any param gathers still outstanding from the previous
+        # step (overlap_param_sync with partial module coverage) finish here
+        for b in self.buckets:
+            self._finish_param_sync_bucket(b)
 
         self.grad_sync()
         self._step += 1
@@ -426,15 +469,19 @@ class DistributedFusedAdam(torch.optim.Optimizer):
                     lo = self.rank * b.shard_size
                     shard = b.param_data[lo:lo + b.shard_size]
                     if _backend_supports_rs(self.process_group):
-                        dist.all_gather_into_tensor(b.param_data, shard,
-                                                    group=self.process_group)
+                        b.param_sync_work = dist.all_gather_into_tensor(
+                            b.param_data, shard, group=self.process_group, async_op=True)
                     else:
                         chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
-                        dist.all_gather(chunks, shard, group=self.process_group)
-                        for r, c in enumerate(chunks):
-                            b.param_data[r * b.shard_size:(r + 1) * b.shard_size].copy_(c)
-            if use_comm_stream:
-                torch.cuda.current_stream().wait_stream(self._comm_stream)
+                        b.param_sync_work = dist.all_gather(
+                            chunks, shard.contiguous(), group=self.process_group,
+                            async_op=True)
+                        b._param_chunks = chunks
+            if not self.overlap_param_sync:
+                for b in self.buckets:
+                    self._finish_param_sync_bucket(b)
+                if use_comm_stream:
+                    torch.cuda.current_stream().wait_stream(self._comm_stream)
         for b in self.buckets:
             # reset for the next iteration
             b.grad_data.zero_()
