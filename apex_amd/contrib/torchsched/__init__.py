@@ -1,20 +1,26 @@
-"""torchsched — torch.compile backend registration (reference:
+"""torchsched — multi-stream torch.compile backend (reference:
 apex/contrib/torchsched — an inductor-based multi-stream graph scheduler).
 
-Round-1 scope: the backend registry surface (``get_backend`` /
-``set_default_backend`` / the ``torchsched`` @register_backend entry) is in
-place and compiles through inductor; the multi-stream (dwb) event/wrapper
-codegen that overlaps independent graph partitions on side HIP streams is a
-documented later-round item — on MI355X it will map partitions onto HIP
-streams with event-based cross-stream ordering.
+The ``torchsched`` backend partitions the captured FX graph into chains
+(scheduler.py) and, when the graph has ≥2 concurrently-runnable partitions,
+executes them on side HIP streams with event-based ordering; graphs with no
+exploitable parallelism compile straight through inductor.
 """
 
 import torch
+
+from .scheduler import MultiStreamGraphModule, partition_graph, max_parallel_width
 
 _default_backend = "inductor"
 
 
 def _torchsched_backend(gm, example_inputs):
+    try:
+        parts, _ = partition_graph(gm)
+        if torch.cuda.is_available() and max_parallel_width(parts) >= 2:
+            return MultiStreamGraphModule(gm)
+    except Exception:
+        pass
     from torch._inductor.compile_fx import compile_fx
 
     return compile_fx(gm, example_inputs)
