@@ -81,7 +81,7 @@ def main():
     cast_dtype = torch.bfloat16 if args.cast_dtype == "bf16" else torch.float16
     model, optimizer = amp.initialize(
         model, optimizer, opt_level=args.opt_level,
-        cast_model_type=None if args.opt_level in ("O0", "O1") else cast_dtype,
+        cast_model_type=None if args.opt_level == "O0" else cast_dtype,
         keep_batchnorm_fp32=args.keep_batchnorm_fp32,
         loss_scale=loss_scale, verbosity=1 if rank == 0 else 0,
     )
