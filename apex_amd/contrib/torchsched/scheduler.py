@@ -86,6 +86,34 @@ def max_parallel_width(partitions):
     return max(width.values(), default=1)
 
 
+def extract_partition_module(gm, partition):
+    """Lift one partition into its own GraphModule. Returns
+    (submodule, external_input_nodes, output_nodes): external inputs are the
+    original graph's nodes whose values must be fed in at call time (other
+    partitions' outputs, placeholders, get_attrs); output nodes are this
+    partition's nodes consumed outside it."""
+    g = torch.fx.Graph()
+    env = {}
+    ext_inputs = []
+    in_part = set(partition.nodes)
+
+    def lookup(n):
+        if n not in env:
+            env[n] = g.placeholder(n.name.replace(".", "_"))
+            ext_inputs.append(n)
+        return env[n]
+
+    for node in partition.nodes:
+        env[node] = g.node_copy(node, lookup)
+    outputs = [n for n in partition.nodes
+               if any(u not in in_part for u in n.users)]
+    if not outputs:
+        outputs = [partition.nodes[-1]]
+    g.output(tuple(env[n] for n in outputs))
+    sub = torch.fx.GraphModule(gm, g)
+    return sub, ext_inputs, outputs
+
+
 class MultiStreamGraphModule:
     """Executes an FX graph with each partition on an assigned HIP stream.
 
@@ -93,13 +121,25 @@ class MultiStreamGraphModule:
     S reuses S (chain continuation after a join); additional concurrent
     partitions round-robin over the side-stream pool. Cross-stream deps are
     ordered with one event each.
+
+    ``compile_partitions=True`` lifts each partition into its own
+    GraphModule and runs it through ``torch.compile`` (inductor) — the
+    reference's partition-wise codegen — while this class keeps doing the
+    stream/event orchestration around the compiled callables.
     """
 
-    def __init__(self, gm, num_streams=4):
+    def __init__(self, gm, num_streams=4, compile_partitions=False):
         self.gm = gm
         self.partitions, self.part_of = partition_graph(gm)
         self.num_streams = num_streams
         self._streams = None
+        self.compile_partitions = compile_partitions
+        if compile_partitions:
+            self._compiled = []
+            for p in self.partitions:
+                sub, ext, outs = extract_partition_module(gm, p)
+                self._compiled.append((torch.compile(sub, backend="inductor"),
+                                       ext, outs))
 
     def _ensure_streams(self):
         if self._streams is None:
@@ -107,7 +147,82 @@ class MultiStreamGraphModule:
             # work, so no false dependencies against host-run-ahead enqueues
             self._streams = [torch.cuda.Stream() for _ in range(self.num_streams)]
 
+    def _assign_streams(self):
+        rr = 0
+        for p in self.partitions:
+            deps = [self.partitions[d] for d in sorted(p.deps)]
+            if deps and deps[0].stream is not None:
+                p.stream = deps[0].stream
+            else:
+                p.stream = self._streams[rr % len(self._streams)]
+                rr += 1
+
+    def _seed_env(self, args):
+        """placeholder/get_attr values + the output spec."""
+        env = {}
+        interp = torch.fx.Interpreter(self.gm)
+        args_iter = iter(args)
+        out_node = None
+        for node in self.gm.graph.nodes:
+            if node.op == "placeholder":
+                env[node] = next(args_iter)
+            elif node.op == "get_attr":
+                env[node] = interp.fetch_attr(node.target)
+            elif node.op == "output":
+                out_node = node
+        return env, out_node
+
+    def _run_compiled(self, *args):
+        env, out_node = self._seed_env(args)
+        use_streams = torch.cuda.is_available() and any(
+            isinstance(a, torch.Tensor) and a.is_cuda for a in args)
+        if not use_streams:
+            for p, (fn, ext, outs) in zip(self.partitions, self._compiled):
+                res = fn(*[env[n] for n in ext])
+                for n, v in zip(outs, res):
+                    env[n] = v
+            return torch.fx.map_arg(out_node.args[0], lambda n: env[n])
+
+        self._ensure_streams()
+        for p in self.partitions:
+            p.stream = None
+            p.event = None
+        self._assign_streams()
+        current = torch.cuda.current_stream()
+        inputs_ready = torch.cuda.Event()
+        inputs_ready.record(current)
+        gated = set()
+        prod_stream = {}
+        for p, (fn, ext, outs) in zip(self.partitions, self._compiled):
+            if id(p.stream) not in gated:
+                p.stream.wait_event(inputs_ready)
+                gated.add(id(p.stream))
+            for d in sorted(p.deps):
+                dp = self.partitions[d]
+                if dp.stream is not p.stream and dp.event is not None:
+                    p.stream.wait_event(dp.event)
+            vals = []
+            for n in ext:
+                v = env[n]
+                if (isinstance(v, torch.Tensor) and v.is_cuda
+                        and prod_stream.get(n) is not p.stream):
+                    v.record_stream(p.stream)
+                vals.append(v)
+            with torch.cuda.stream(p.stream):
+                res = fn(*vals)
+            for n, v in zip(outs, res):
+                env[n] = v
+                prod_stream[n] = p.stream
+            p.event = torch.cuda.Event()
+            p.event.record(p.stream)
+        for p in self.partitions:
+            if p.event is not None:
+                current.wait_event(p.event)
+        return torch.fx.map_arg(out_node.args[0], lambda n: env[n])
+
     def __call__(self, *args):
+        if self.compile_partitions:
+            return self._run_compiled(*args)
         if not (torch.cuda.is_available() and any(
                 isinstance(a, torch.Tensor) and a.is_cuda for a in args)):
             return self.gm(*args)  # CPU: plain sequential execution

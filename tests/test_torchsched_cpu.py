@@ -61,3 +61,15 @@ def test_backend_registered():
 
     assert callable(torchsched.get_backend())
     assert "torchsched" in dynamo.list_backends()
+
+
+def test_compiled_partitions_cpu_matches_eager():
+    torch.manual_seed(1)
+    m = ForkJoin()
+    gm = torch.fx.symbolic_trace(m)
+    ms = MultiStreamGraphModule(gm, compile_partitions=True)
+    x = torch.randn(4, 16)
+    torch.testing.assert_close(ms(x), m(x), rtol=1e-5, atol=1e-6)
+    # second call reuses the compiled callables
+    x2 = torch.randn(4, 16)
+    torch.testing.assert_close(ms(x2), m(x2), rtol=1e-5, atol=1e-6)

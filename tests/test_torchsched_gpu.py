@@ -51,3 +51,26 @@ def test_torchsched_compile_backend_gpu():
     cm = torch.compile(m, backend="torchsched")
     x = torch.randn(32, 128, device="cuda")
     torch.testing.assert_close(cm(x), m(x), rtol=1e-5, atol=1e-5)
+
+
+def test_compiled_partitions_gpu():
+    from apex_amd.contrib.torchsched.scheduler import MultiStreamGraphModule
+
+    class FJ(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(128, 128)
+            self.b = torch.nn.Linear(128, 128)
+
+        def forward(self, x):
+            h = torch.relu(x)
+            return torch.tanh(self.a(h)) + torch.sigmoid(self.b(h))
+
+    torch.manual_seed(2)
+    m = FJ().cuda()
+    gm = torch.fx.symbolic_trace(m)
+    ms = MultiStreamGraphModule(gm, compile_partitions=True)
+    x = torch.randn(16, 128, device="cuda")
+    out = ms(x)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out, m(x), rtol=1e-4, atol=1e-4)
