@@ -4,18 +4,25 @@ API parity with the reference ``apex.contrib.bottleneck``
 (apex/contrib/bottleneck/bottleneck.py: Bottleneck:154,
 SpatialBottleneckFunction:304): frozen-BN folded into per-channel
 scale/bias, conv+scale+bias+ReLU composed per branch, and a spatial variant
-that splits H across a rank group and halo-exchanges ``dilation`` edge rows
-before the 3x3 convolution (``spatial_method=1``; the output-halo variants
-2/3 are a later round).
+that splits H across a rank group: ``spatial_method=1`` splices exchanged
+input halos and runs one conv; methods 2/3 run the interior conv on the
+unpadded local slab and recompute only the two edge output rows from the
+received halos (3 overlaps those small convs on a side stream).
 
 The reference drives cuDNN-frontend fusion; on MI355X the convs run through
 MIOpen (torch conv2d) with the scale/bias/ReLU epilogue composed around them
 (see conv_bias_relu.py for the fusion note).
 """
 
+import contextlib
+
 import torch
 
 from ..conv_bias_relu import ConvFrozenScaleBiasReLU
+
+
+def _null_ctx():
+    return contextlib.nullcontext()
 
 
 class FrozenBatchNorm2d(torch.nn.Module):
@@ -107,22 +114,12 @@ class SpatialBottleneck(Bottleneck):
             self.spatial_args = spatial_parallel_args
         (self.spatial_group_size, self.spatial_group_rank, self.spatial_communicator,
          self.spatial_halo_exchanger, self.spatial_method, self.use_delay_kernel) = self.spatial_args
-        if self.spatial_method not in (0, 1):
-            raise NotImplementedError(
-                "spatial_method 2/3 (remote output-halo compute) lands in a later round"
-            )
+        if self.spatial_method not in (0, 1, 2, 3):
+            raise ValueError(f"unknown spatial_method {self.spatial_method}")
 
-    def forward(self, x):
-        if self.spatial_group_size <= 1:
-            return super().forward(x)
-
-        s1, b1 = self.bn1.get_scale_bias()
-        s2, b2 = self.bn2.get_scale_bias()
-        s3, b3 = self.bn3.get_scale_bias()
-
-        out = ConvFrozenScaleBiasReLU(x, self.conv1.weight, s1, b1, 0, 1)
-
-        # halo exchange: 1 row each side for the padded 3x3 conv
+    def _conv2_input_halo(self, out, s2, b2):
+        """spatial_method 1: splice exchanged input halos onto the local
+        slab, run one conv over the padded slab, crop."""
         top_halo = out[:, :, :1, :]
         btm_halo = out[:, :, -1:, :]
         left_in, right_in = self.spatial_halo_exchanger.left_right_halo_exchange(
@@ -141,7 +138,61 @@ class SpatialBottleneck(Bottleneck):
         top_crop = 1 if self.spatial_group_rank > 0 else 0
         h_local = out.shape[2] // self.stride if self.stride > 1 else out.shape[2]
         mid = mid[:, :, top_crop // max(self.stride, 1):, :]
-        mid = mid[:, :, :h_local, :].contiguous()
+        return mid[:, :, :h_local, :].contiguous()
+
+    def _conv2_edge_correction(self, out, s2, b2):
+        """spatial_method 2/3: the interior conv runs on the local slab only
+        (zero-padded) while halos are in flight; the two edge output rows are
+        then recomputed from the received halo rows and spliced in
+        (reference: output-halo compute, bottleneck.py spatial_method 2/3).
+        Method 3 additionally runs the small edge convs on the exchanger's
+        side stream so they overlap the 1x1 conv3 launch on the main stream.
+        """
+        gr, gs = self.spatial_group_rank, self.spatial_group_size
+        top_halo = out[:, :, :1, :]
+        btm_halo = out[:, :, -1:, :]
+        left_in, right_in = self.spatial_halo_exchanger.left_right_halo_exchange(
+            top_halo.contiguous(), btm_halo.contiguous()
+        )
+        # interior: all output rows whose 3x3 receptive field is local
+        mid = ConvFrozenScaleBiasReLU(out, self.conv2.weight, s2, b2, 1, 1)
+
+        def edge(rows3):
+            z = torch.nn.functional.conv2d(rows3, self.conv2.weight, padding=(0, 1))
+            return torch.relu(z * s2 + b2)
+
+        stream = getattr(self.spatial_halo_exchanger, "stream1", None)
+        use_side = self.spatial_method == 3 and stream is not None and out.is_cuda
+        ctx = torch.cuda.stream(stream) if use_side else _null_ctx()
+        if use_side:
+            stream.wait_stream(torch.cuda.current_stream())
+        with ctx:
+            top_row = edge(torch.cat([left_in, out[:, :, :2, :]], dim=2)) if gr > 0 else None
+            btm_row = edge(torch.cat([out[:, :, -2:, :], right_in], dim=2)) if gr < gs - 1 else None
+        if use_side:
+            torch.cuda.current_stream().wait_stream(stream)
+        rows = [mid[:, :, 1:, :] if gr > 0 else mid]
+        if gr > 0:
+            rows.insert(0, top_row)
+        if gr < gs - 1:
+            rows[-1] = rows[-1][:, :, :-1, :]
+            rows.append(btm_row)
+        return torch.cat(rows, dim=2)
+
+    def forward(self, x):
+        if self.spatial_group_size <= 1:
+            return super().forward(x)
+
+        s1, b1 = self.bn1.get_scale_bias()
+        s2, b2 = self.bn2.get_scale_bias()
+        s3, b3 = self.bn3.get_scale_bias()
+
+        out = ConvFrozenScaleBiasReLU(x, self.conv1.weight, s1, b1, 0, 1)
+
+        if self.spatial_method in (2, 3) and self.stride == 1:
+            mid = self._conv2_edge_correction(out, s2, b2)
+        else:  # methods 0/1, and the strided fallback for 2/3
+            mid = self._conv2_input_halo(out, s2, b2)
 
         out3 = torch.nn.functional.conv2d(mid, self.conv3.weight) * s3 + b3
         if self.use_downsample:

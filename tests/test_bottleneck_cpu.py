@@ -89,7 +89,7 @@ def test_halo_exchangers_gloo(kind):
     run_distributed(_halo_worker, world_size=2, args=(kind,))
 
 
-def _spatial_worker(rank, world_size):
+def _spatial_worker(rank, world_size, method=1):
     from apex_amd.contrib.bottleneck import (
         Bottleneck, SpatialBottleneck, HaloExchangerAllGather,
     )
@@ -106,7 +106,7 @@ def _spatial_worker(rank, world_size):
 
     hx = HaloExchangerAllGather(list(range(world_size)), rank)
     sp = SpatialBottleneck(8, 4, 8, stride=1,
-                           spatial_parallel_args=(world_size, rank, None, hx, 1, False))
+                           spatial_parallel_args=(world_size, rank, None, hx, method, False))
     sp.load_state_dict(full.state_dict())
     h_local = H // world_size
     x_local = x_full[:, :, rank * h_local:(rank + 1) * h_local, :].contiguous()
@@ -115,8 +115,9 @@ def _spatial_worker(rank, world_size):
     torch.testing.assert_close(y_local, expected, rtol=1e-4, atol=1e-5)
 
 
-def test_spatial_bottleneck_matches_full():
-    run_distributed(_spatial_worker, world_size=2)
+@pytest.mark.parametrize("method", [1, 2, 3])
+def test_spatial_bottleneck_matches_full(method):
+    run_distributed(_spatial_worker, world_size=2, args=(method,))
 
 
 def test_gbn_single_rank_matches_bn():
