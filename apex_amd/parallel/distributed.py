@@ -248,3 +248,31 @@ class _nullcontext:
 
     def __exit__(self, *a):
         return False
+
+
+class Reducer:
+    """Manual gradient/state reducer (reconstructed removed-apex
+    ``apex.parallel.Reducer``): no hooks, no overlap — the user calls
+    ``reduce()`` once per accumulation window. Useful when backward timing
+    must stay untouched (e.g. gradient accumulation loops)."""
+
+    def __init__(self, module_or_grads_list):
+        if isinstance(module_or_grads_list, torch.nn.Module):
+            self.module = module_or_grads_list
+            flat_dist_call([p.data for p in self.module.parameters()],
+                           dist.broadcast, (0,))
+        else:
+            self.module = None
+            self.grads = module_or_grads_list
+
+    def reduce(self):
+        if self.module is not None:
+            grads = [p.grad.data for p in self.module.parameters()
+                     if p.grad is not None]
+        else:
+            grads = self.grads
+        if grads:
+            flat_dist_call(grads, dist.all_reduce)
+            world = dist.get_world_size()
+            for g in grads:
+                g.div_(world)

@@ -95,3 +95,37 @@ def test_ddp_multiple_allreduce_pgs():
     # the multi-PG rotation is the same code path as on GPU.
     run_distributed(_ddp_worker, world_size=2,
                     args=({"num_allreduce_streams": 2, "message_size": 1},))
+
+def _reducer_worker(rank, world_size):
+    from apex_amd.parallel import Reducer
+
+    torch.manual_seed(rank)
+    model = torch.nn.Linear(8, 4)
+    red = Reducer(model)  # broadcasts initial params from rank 0
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    flats = [torch.empty_like(flat) for _ in range(world_size)]
+    dist.all_gather(flats, flat)
+    torch.testing.assert_close(flats[0], flats[1])
+
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    model(x).sum().backward()
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x)
+    red.reduce()
+    # expected: mean of per-rank grads
+    ref = torch.nn.Linear(8, 4)
+    ref.load_state_dict(model.state_dict())
+    grads = None
+    for xi in xs:
+        m = torch.nn.Linear(8, 4)
+        m.load_state_dict(model.state_dict())
+        m(xi).sum().backward()
+        g = [p.grad.clone() for p in m.parameters()]
+        grads = g if grads is None else [a + b for a, b in zip(grads, g)]
+    for p, g in zip(model.parameters(), grads):
+        torch.testing.assert_close(p.grad, g / world_size, rtol=1e-6, atol=1e-7)
+
+
+def test_reducer():
+    run_distributed(_reducer_worker, world_size=2)
