@@ -272,7 +272,5 @@ class Reducer:
         else:
             grads = self.grads
         if grads:
+            # flat_dist_call divides by world size after the all_reduce
             flat_dist_call(grads, dist.all_reduce)
-            world = dist.get_world_size()
-            for g in grads:
-                g.div_(world)
