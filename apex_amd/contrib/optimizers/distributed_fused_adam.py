@@ -120,6 +120,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self.set_grad_none = set_grad_none
         self.overlap_grad_sync = overlap_grad_sync
         self.overlap_param_sync = overlap_param_sync
+        self._hooks_registered = False
         self.average_grad_sync = average_grad_sync
         # 2-D process grid (reference: distributed_process_group /
         # redundant_process_group): optimizer state is SHARDED over the
@@ -365,11 +366,14 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             torch.cuda.current_stream().wait_stream(self._comm_stream)
         b.param_sync_work = None
 
+    _warned_no_hooks = False
+
     def register_model_for_param_sync(self, model):
         """With ``overlap_param_sync=True``: hook each module's pre-forward
         to wait only for the all-gathers of the buckets holding its own
         params — the next forward's early layers run while later buckets are
         still in flight (reference: overlap_param_sync + param-sync hooks)."""
+        self._hooks_registered = True
         for module in model.modules():
             buckets = {}
             for p in module.parameters(recurse=False):
@@ -417,6 +421,16 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
         # safety net: any param gathers still outstanding from the previous
         # step (overlap_param_sync with partial module coverage) finish here
+        if (self.overlap_param_sync and self.world_size > 1 and self._step > 0
+                and not self._hooks_registered and not type(self)._warned_no_hooks):
+            import warnings
+
+            warnings.warn(
+                "overlap_param_sync=True but register_model_for_param_sync() was "
+                "never called — the previous step's param all-gathers were only "
+                "completed now, at step time, NOT before the forward that used "
+                "them. Register the model or disable overlap_param_sync.")
+            type(self)._warned_no_hooks = True
         for b in self.buckets:
             self._finish_param_sync_bucket(b)
 
