@@ -65,3 +65,30 @@ def test_torchsched_compiles_a_function():
     compiled = torch.compile(f, backend="torchsched")
     a, b = torch.randn(4, 8), torch.randn(8, 4)
     torch.testing.assert_close(compiled(a, b), f(a, b))
+
+
+def test_fused_adam_swa_from_optim():
+    from apex_amd.contrib.openfold import FusedAdamSWA
+
+    torch.manual_seed(0)
+    fp32 = [torch.randn(8, 4, requires_grad=True), torch.randn(6, requires_grad=True)]
+    adam = torch.optim.Adam(fp32, lr=1e-2, weight_decay=0.01)
+    for _ in range(3):  # build up real Adam state
+        for p in fp32:
+            p.grad = torch.randn_like(p)
+        adam.step()
+    bf16 = [p.detach().to(torch.bfloat16) for p in fp32]
+    swa = [p.detach().clone() for p in fp32]
+    opt = FusedAdamSWA.from_optim(adam, fp32, bf16, swa, swa_decay_rate=0.9)
+    assert opt.param_groups[0]["step"] == 3
+    for src, dst in zip(fp32, opt.param_groups[0]["params"]):
+        torch.testing.assert_close(opt.state[dst]["exp_avg"],
+                                   adam.state[src]["exp_avg"])
+    for p in fp32:
+        p.grad = torch.randn_like(p)
+    before = [s.clone() for s in swa]
+    opt.step()
+    for s, b, p in zip(swa, before, fp32):
+        torch.testing.assert_close(s, 0.9 * b + 0.1 * p.detach(), rtol=1e-5, atol=1e-6)
+    for c, p in zip(bf16, fp32):  # compute copy refreshed
+        torch.testing.assert_close(c, p.detach().to(torch.bfloat16))
