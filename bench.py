@@ -18,6 +18,13 @@ import json
 import os
 import time
 
+# Cold-box guard: cudnn.benchmark=True would trigger MIOpen's exhaustive
+# per-shape search (minutes for ResNet-50's ~50 conv configs on a fresh
+# machine with an empty user find-db). FAST find consults the shipped gfx950
+# perf db and returns near-tuned solutions immediately. Honored only if the
+# caller hasn't set their own find mode.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 import torch.distributed as dist
 

@@ -14,6 +14,8 @@ import json
 import os
 import time
 
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")  # cold-box guard (see bench.py)
+
 import os
 import sys
 
