@@ -118,27 +118,27 @@ class SpatialBottleneck(Bottleneck):
             raise ValueError(f"unknown spatial_method {self.spatial_method}")
 
     def _conv2_input_halo(self, out, s2, b2):
-        """spatial_method 1: splice exchanged input halos onto the local
-        slab, run one conv over the padded slab, crop."""
+        """spatial_method 1: splice exchanged input halos (or the global
+        zero-padding rows at the outer boundaries) onto the local slab, run
+        one conv with explicit rows instead of vertical padding, crop.
+
+        Using explicit boundary rows keeps the stride phase right: the
+        unpadded 3x3 conv's first output centers on slab row 1, which is
+        always the rank's first global center (r * h_local is a multiple of
+        the stride because h_local is)."""
+        gr, gs = self.spatial_group_rank, self.spatial_group_size
         top_halo = out[:, :, :1, :]
         btm_halo = out[:, :, -1:, :]
         left_in, right_in = self.spatial_halo_exchanger.left_right_halo_exchange(
             top_halo.contiguous(), btm_halo.contiguous()
         )
-        parts = []
-        if self.spatial_group_rank > 0:
-            parts.append(left_in)
-        parts.append(out)
-        if self.spatial_group_rank < self.spatial_group_size - 1:
-            parts.append(right_in)
-        padded = torch.cat(parts, dim=2)
-
-        mid = ConvFrozenScaleBiasReLU(padded, self.conv2.weight, s2, b2, 1, self.stride)
-        # crop the rows produced by the halo padding
-        top_crop = 1 if self.spatial_group_rank > 0 else 0
-        h_local = out.shape[2] // self.stride if self.stride > 1 else out.shape[2]
-        mid = mid[:, :, top_crop // max(self.stride, 1):, :]
-        return mid[:, :, :h_local, :].contiguous()
+        zero_row = torch.zeros_like(out[:, :, :1, :])
+        parts = [left_in if gr > 0 else zero_row, out,
+                 right_in if gr < gs - 1 else zero_row]
+        slab = torch.cat(parts, dim=2)
+        mid = ConvFrozenScaleBiasReLU(slab, self.conv2.weight, s2, b2, (0, 1), self.stride)
+        h_out_local = out.shape[2] // self.stride
+        return mid[:, :, :h_out_local, :].contiguous()
 
     def _conv2_edge_correction(self, out, s2, b2):
         """spatial_method 2/3: the interior conv runs on the local slab only

@@ -133,3 +133,35 @@ def test_gbn_single_rank_matches_bn():
         bn.bias.copy_(gbn.bias)
     x = torch.randn(4, 8, 5, 5)
     torch.testing.assert_close(gbn(x), bn(x), rtol=1e-5, atol=1e-6)
+
+
+def _spatial_stride2_worker(rank, world_size):
+    # methods 2/3 fall back to the input-halo path when stride>1 — results
+    # must still match the single-rank reference
+    from apex_amd.contrib.bottleneck import (
+        Bottleneck, SpatialBottleneck, HaloExchangerAllGather,
+    )
+
+    torch.manual_seed(5)
+    H = 16
+    full = Bottleneck(8, 4, 16, stride=2)
+    x_full = torch.randn(2, 8, H, 8)
+    for p in full.state_dict().values():
+        dist.broadcast(p, 0)
+    dist.broadcast(x_full, 0)
+    y_full = full(x_full)
+
+    hx = HaloExchangerAllGather(list(range(world_size)), rank)
+    sp = SpatialBottleneck(8, 4, 16, stride=2,
+                           spatial_parallel_args=(world_size, rank, None, hx, 2, False))
+    sp.load_state_dict(full.state_dict())
+    h_local = H // world_size
+    x_local = x_full[:, :, rank * h_local:(rank + 1) * h_local, :].contiguous()
+    y_local = sp(x_local)
+    h_out = y_full.shape[2] // world_size
+    expected = y_full[:, :, rank * h_out:(rank + 1) * h_out, :]
+    torch.testing.assert_close(y_local, expected, rtol=1e-4, atol=1e-5)
+
+
+def test_spatial_bottleneck_stride2_fallback():
+    run_distributed(_spatial_stride2_worker, world_size=2)

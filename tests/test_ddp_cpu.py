@@ -129,3 +129,35 @@ def _reducer_worker(rank, world_size):
 
 def test_reducer():
     run_distributed(_reducer_worker, world_size=2)
+
+
+def _no_sync_worker(rank, world_size):
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _model_ctor()
+    ddp = DDP(model, message_size=1)
+    torch.manual_seed(31 + rank)
+    x1, x2 = torch.randn(4, 8), torch.randn(4, 8)
+    with ddp.no_sync():  # grads accumulate locally, no collective
+        ddp(x1).sum().backward()
+    local = [p.grad.clone() for p in ddp.module.parameters()]
+    ddp(x2).sum().backward()  # second backward DOES reduce (accumulated grads)
+    # compare vs expected: mean over ranks of (g(x1)+g(x2))
+    xs = [torch.empty_like(x1) for _ in range(world_size)]
+    ys = [torch.empty_like(x2) for _ in range(world_size)]
+    dist.all_gather(xs, x1)
+    dist.all_gather(ys, x2)
+    grads = None
+    for a, b in zip(xs, ys):
+        m = _model_ctor()
+        (m(a).sum() + m(b).sum()).backward()
+        g = [p.grad.clone() for p in m.parameters()]
+        grads = g if grads is None else [u + v for u, v in zip(grads, g)]
+    for p, e in zip(ddp.module.parameters(), grads):
+        torch.testing.assert_close(p.grad, e / world_size, rtol=1e-5, atol=1e-6)
+    assert any(not torch.equal(l, p.grad) for l, p in
+               zip(local, ddp.module.parameters()))  # reduction did change them
+
+
+def test_ddp_no_sync():
+    run_distributed(_no_sync_worker, world_size=2)
