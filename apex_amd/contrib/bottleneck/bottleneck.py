@@ -148,33 +148,41 @@ class SpatialBottleneck(Bottleneck):
         Method 3 additionally runs the small edge convs on the exchanger's
         side stream so they overlap the 1x1 conv3 launch on the main stream.
         """
-        gr, gs = self.spatial_group_rank, self.spatial_group_size
+        gr, gs, s = self.spatial_group_rank, self.spatial_group_size, self.stride
         top_halo = out[:, :, :1, :]
         btm_halo = out[:, :, -1:, :]
         left_in, right_in = self.spatial_halo_exchanger.left_right_halo_exchange(
             top_halo.contiguous(), btm_halo.contiguous()
         )
-        # interior: all output rows whose 3x3 receptive field is local
-        mid = ConvFrozenScaleBiasReLU(out, self.conv2.weight, s2, b2, 1, 1)
+        # interior: the zero-padded local conv; output centers land on local
+        # rows 0, s, 2s, … which are globally phase-aligned (h_local % s == 0)
+        mid = ConvFrozenScaleBiasReLU(out, self.conv2.weight, s2, b2, 1, s)
 
         def edge(rows3):
-            z = torch.nn.functional.conv2d(rows3, self.conv2.weight, padding=(0, 1))
+            # 3 input rows -> 1 output row; W keeps the conv's stride
+            z = torch.nn.functional.conv2d(rows3, self.conv2.weight,
+                                           stride=(1, s), padding=(0, 1))
             return torch.relu(z * s2 + b2)
 
+        # which output rows have a receptive field reaching a neighbour:
+        # the first (center local row 0, needs the left halo) for every
+        # stride; the last (center h_local-1) only when s == 1
+        need_top = gr > 0
+        need_btm = s == 1 and gr < gs - 1
         stream = getattr(self.spatial_halo_exchanger, "stream1", None)
         use_side = self.spatial_method == 3 and stream is not None and out.is_cuda
         ctx = torch.cuda.stream(stream) if use_side else _null_ctx()
         if use_side:
             stream.wait_stream(torch.cuda.current_stream())
         with ctx:
-            top_row = edge(torch.cat([left_in, out[:, :, :2, :]], dim=2)) if gr > 0 else None
-            btm_row = edge(torch.cat([out[:, :, -2:, :], right_in], dim=2)) if gr < gs - 1 else None
+            top_row = edge(torch.cat([left_in, out[:, :, :2, :]], dim=2)) if need_top else None
+            btm_row = edge(torch.cat([out[:, :, -2:, :], right_in], dim=2)) if need_btm else None
         if use_side:
             torch.cuda.current_stream().wait_stream(stream)
-        rows = [mid[:, :, 1:, :] if gr > 0 else mid]
-        if gr > 0:
+        rows = [mid[:, :, 1:, :] if need_top else mid]
+        if need_top:
             rows.insert(0, top_row)
-        if gr < gs - 1:
+        if need_btm:
             rows[-1] = rows[-1][:, :, :-1, :]
             rows.append(btm_row)
         return torch.cat(rows, dim=2)
@@ -189,9 +197,9 @@ class SpatialBottleneck(Bottleneck):
 
         out = ConvFrozenScaleBiasReLU(x, self.conv1.weight, s1, b1, 0, 1)
 
-        if self.spatial_method in (2, 3) and self.stride == 1:
+        if self.spatial_method in (2, 3):
             mid = self._conv2_edge_correction(out, s2, b2)
-        else:  # methods 0/1, and the strided fallback for 2/3
+        else:  # methods 0/1
             mid = self._conv2_input_halo(out, s2, b2)
 
         out3 = torch.nn.functional.conv2d(mid, self.conv3.weight) * s3 + b3

@@ -135,9 +135,9 @@ def test_gbn_single_rank_matches_bn():
     torch.testing.assert_close(gbn(x), bn(x), rtol=1e-5, atol=1e-6)
 
 
-def _spatial_stride2_worker(rank, world_size):
-    # methods 2/3 fall back to the input-halo path when stride>1 — results
-    # must still match the single-rank reference
+def _spatial_stride2_worker(rank, world_size, method):
+    # stride 2: method 1 splices halos with explicit boundary rows; methods
+    # 2/3 correct only the top edge row (the bottom field is fully local)
     from apex_amd.contrib.bottleneck import (
         Bottleneck, SpatialBottleneck, HaloExchangerAllGather,
     )
@@ -153,7 +153,7 @@ def _spatial_stride2_worker(rank, world_size):
 
     hx = HaloExchangerAllGather(list(range(world_size)), rank)
     sp = SpatialBottleneck(8, 4, 16, stride=2,
-                           spatial_parallel_args=(world_size, rank, None, hx, 2, False))
+                           spatial_parallel_args=(world_size, rank, None, hx, method, False))
     sp.load_state_dict(full.state_dict())
     h_local = H // world_size
     x_local = x_full[:, :, rank * h_local:(rank + 1) * h_local, :].contiguous()
@@ -163,5 +163,6 @@ def _spatial_stride2_worker(rank, world_size):
     torch.testing.assert_close(y_local, expected, rtol=1e-4, atol=1e-5)
 
 
-def test_spatial_bottleneck_stride2_fallback():
-    run_distributed(_spatial_stride2_worker, world_size=2)
+@pytest.mark.parametrize("method", [1, 2, 3])
+def test_spatial_bottleneck_stride2(method):
+    run_distributed(_spatial_stride2_worker, world_size=2, args=(method,))
