@@ -15,6 +15,7 @@ from .rope import (
     fused_apply_rotary_pos_emb_2d,
 )
 from .wgrad import wgrad_gemm_accum_fp32, wgrad_gemm_accum_fp16
+from .fmha import flash_attention_forward
 
 __all__ = [
     "ScaledMaskedSoftmax",
@@ -31,4 +32,5 @@ __all__ = [
     "fused_apply_rotary_pos_emb_2d",
     "wgrad_gemm_accum_fp32",
     "wgrad_gemm_accum_fp16",
+    "flash_attention_forward",
 ]

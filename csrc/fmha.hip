@@ -1,0 +1,218 @@
+// apex_amd._mfma fmha_fwd — flash-attention forward for gfx950 (EXPERIMENTAL:
+// compile-checked + unit-tested against the eager composition; not yet wired
+// into the model hot path — see ROADMAP round-2 perf plan).
+//
+// One 256-thread workgroup per (batch*head, 64 query rows); each wave64 owns
+// 16 query rows independently (no cross-wave barriers — causal waves finish
+// their shorter KV loops early). Per 32-wide KV tile:
+//   S = Q K^T via mfma_f32_16x16x32_bf16 (Q held as A-fragments in registers
+//       for the whole row block; K loaded straight from HBM in B-fragment
+//       order — both are 16-byte vector loads, no LDS staging),
+//   online softmax (row max/sum via 4-step shfl_xor over the 16 column
+//       lanes; m/l replicated across those lanes),
+//   P V via one MFMA per 16 head-dim columns, with P transposed from the
+//       D-fragment to the A-fragment layout through a 1 KB per-wave LDS
+//       bounce (wave-local: s_waitcnt instead of a block barrier).
+//
+// Fragment layouts (verified on-device by mfma_tile_probe):
+//   A[r][k]: lane l -> r = l%16, k = (l/16)*8 + j
+//   B[k][c]: lane l -> c = l%16, k = (l/16)*8 + j
+//   D[r][c]: lane l reg q -> r = (l>>4)*4 + q, c = l&15
+#include "common.h"
+
+#include <vector>
+
+namespace {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+constexpr int FM_ROWS = 16;   // q rows per wave
+constexpr int FM_WAVES = 4;   // waves per workgroup
+constexpr int FM_BN = 32;     // kv tile (one MFMA K step for P@V)
+constexpr int FM_MAXD = 128;
+
+__device__ __forceinline__ float row_reduce_max16(float v) {
+#pragma unroll
+  for (int m = 1; m < 16; m <<= 1) v = fmaxf(v, __shfl_xor(v, m));
+  return v;
+}
+
+__device__ __forceinline__ float row_reduce_sum16(float v) {
+#pragma unroll
+  for (int m = 1; m < 16; m <<= 1) v += __shfl_xor(v, m);
+  return v;
+}
+
+template <bool CAUSAL, int D>
+__global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
+    short* __restrict__ O, float* __restrict__ LSE, int S, float scale) {
+  constexpr int NK = D / 32;  // K chunks for Q@K^T
+  constexpr int ND = D / 16;  // 16-col output groups for P@V
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
+  if (q0 >= S) return;
+
+  const short* q_ptr = Q + (long)bh * S * D;
+  const short* k_ptr = K + (long)bh * S * D;
+  const short* v_ptr = V + (long)bh * S * D;
+
+  __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
+  short* pbuf = lds_p[wave];
+
+  // Q rows for this wave, as A-fragments, resident for the whole pass
+  bf16x8 aq[NK];
+  const int a_row = q0 + (lane & 15);
+#pragma unroll
+  for (int c = 0; c < NK; ++c)
+    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+
+  f32x4 acc[ND];
+#pragma unroll
+  for (int d = 0; d < ND; ++d) acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    m_run[q] = -1e30f;
+    l_run[q] = 0.f;
+  }
+
+  // rows this lane's D-fragments correspond to (replicated over col lanes)
+  const int my_r0 = (lane >> 4) * 4;  // + q
+  const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
+    // ---- S = scale * Q K^T for this 16 x 32 tile (two 16x16 halves) ----
+    f32x4 s_half[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      f32x4 s = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int k_row = kv0 + j * 16 + (lane & 15);
+#pragma unroll
+      for (int c = 0; c < NK; ++c) {
+        const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
+            k_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[c], bk, s, 0, 0, 0);
+      }
+      s_half[j] = s;
+    }
+
+    // ---- scale + causal mask + online softmax update ----
+    float p_val[2][4];
+    float alpha[4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int row_g = q0 + my_r0 + q;
+      float mx = -1e30f;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        float v = s_half[j][q] * scale;
+        if (CAUSAL) {
+          const int col_g = kv0 + j * 16 + (lane & 15);
+          if (col_g > row_g) v = -1e30f;
+        }
+        s_half[j][q] = v;
+        mx = fmaxf(mx, v);
+      }
+      mx = row_reduce_max16(mx);
+      const float m_new = fmaxf(m_run[q], mx);
+      alpha[q] = __expf(m_run[q] - m_new);
+      float sum = 0.f;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const float p = __expf(s_half[j][q] - m_new);
+        p_val[j][q] = p;
+        sum += p;
+      }
+      sum = row_reduce_sum16(sum);
+      l_run[q] = l_run[q] * alpha[q] + sum;
+      m_run[q] = m_new;
+    }
+
+    // ---- rescale the running output ----
+#pragma unroll
+    for (int d = 0; d < ND; ++d)
+#pragma unroll
+      for (int q = 0; q < 4; ++q) acc[d][q] *= alpha[q];
+
+    // ---- P: D-fragment -> A-fragment via the wave-local LDS bounce ----
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const __hip_bfloat16 pb = __float2bfloat16(p_val[j][q]);
+        pbuf[(my_r0 + q) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&pb);
+      }
+    __builtin_amdgcn_s_waitcnt(0);  // wave-local: drain LDS stores before reads
+    const bf16x8 ap = *reinterpret_cast<const bf16x8*>(
+        pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
+
+    // ---- acc += P V (one MFMA per 16 head-dim columns) ----
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      // B[k=kv][c=dim]: lane -> c = lane%16, k = (lane/16)*8 + j
+      bf16x8 bv;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        const int kvr = kv0 + (lane >> 4) * 8 + jj;
+        bv[jj] = v_ptr[(long)kvr * D + d * 16 + (lane & 15)];
+      }
+      acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc[d], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // LDS reads done before next tile's stores
+  }
+
+  // ---- epilogue: O = acc / l, LSE = m + log(l) ----
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const long row_g = q0 + my_r0 + q;
+    const float inv_l = 1.f / l_run[q];
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      const __hip_bfloat16 o = __float2bfloat16(acc[d][q] * inv_l);
+      O[((long)bh * S + row_g) * D + d * 16 + (lane & 15)] =
+          *reinterpret_cast<const short*>(&o);
+    }
+    if ((lane & 15) == 0 && LSE)
+      LSE[(long)bh * S + row_g] = m_run[q] + logf(l_run[q]);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
+                                 double scale) {
+  TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_fwd: bf16 only");
+  TORCH_CHECK(q.dim() == 4, "fmha_fwd: [B, H, S, D]");
+  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
+  TORCH_CHECK(kc.sizes() == qc.sizes() && vc.sizes() == qc.sizes(),
+              "fmha_fwd: q/k/v shapes must match (no MQA yet)");
+  TORCH_CHECK(D == 64 || D == 128, "fmha_fwd: head_dim must be 64 or 128");
+  TORCH_CHECK(S % 32 == 0, "fmha_fwd: seq_len must be a multiple of 32");
+  auto out = at::empty_like(qc);
+  auto lse = at::empty({B, H, S}, qc.options().dtype(at::kFloat));
+  auto stream = current_stream();
+  dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
+  dim3 block(FM_WAVES * 64);
+  const float sc = (float)scale;
+
+#define FMHA_LAUNCH(CAUSAL, DD)                                                            \
+  hipLaunchKernelGGL((fmha_fwd_kernel<CAUSAL, DD>), grid, block, 0, stream,                \
+                     (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),             \
+                     (const short*)vc.data_ptr(), (short*)out.data_ptr(),                  \
+                     lse.data_ptr<float>(), S, sc)
+  if (causal) {
+    if (D == 64) FMHA_LAUNCH(true, 64);
+    else FMHA_LAUNCH(true, 128);
+  } else {
+    if (D == 64) FMHA_LAUNCH(false, 64);
+    else FMHA_LAUNCH(false, 128);
+  }
+#undef FMHA_LAUNCH
+  HIP_CHECK(hipGetLastError());
+  return {out, lse};
+}

@@ -65,7 +65,7 @@ ext_modules = [
     hip_ext("apex_amd._peer_memory", ["csrc/peer_memory.hip"]),
     hip_ext("apex_amd._rccl_p2p", ["csrc/rccl_p2p.hip"], libraries=["rccl"]),
     hip_ext("apex_amd._rccl_allocator", ["csrc/rccl_allocator.cpp"], libraries=["rccl"]),
-    hip_ext("apex_amd._mfma", ["csrc/mfma_gemm.hip", "csrc/mfma_probe.hip"]),
+    hip_ext("apex_amd._mfma", ["csrc/mfma_gemm.hip", "csrc/mfma_probe.hip", "csrc/fmha.hip"]),
     hip_ext("apex_amd._tune", ["csrc/stream_tune.hip"]),
 ]
 
