@@ -39,3 +39,61 @@ def flash_attention_forward(q, k, v, causal=False, scale=None):
         return out, lse
     out, lse = eager_attention_reference(q, k, v, causal, scale)
     return out.to(q.dtype), lse
+
+
+class FlashAttentionFunction(torch.autograd.Function):
+    """Training wrapper: fused forward + chunked-recompute backward.
+
+    Backward recomputes P tile-by-tile from (q, k, lse) — the standard
+    flash-attention identities with D_i = rowsum(dO * O):
+        dV = P^T dO,  dP = dO V^T,  dS = P * (dP - D_i),
+        dQ = dS K * scale,  dK = dS^T Q * scale.
+    Only q-tile-sized intermediates are live (no S x S matrix), and each
+    tile's math is plain GEMMs (hipBLASLt on GPU). A fused MFMA backward is
+    the round-2 follow-up.
+    """
+
+    CHUNK = 256  # q rows recomputed per tile
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        out, lse = flash_attention_forward(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, dlse_unused):
+        q, k, v, out, lse = ctx.saved_tensors
+        causal, scale = ctx.causal, ctx.scale
+        S = q.shape[-2]
+        qf, kf, vf = q.float(), k.float(), v.float()
+        dof = dout.float()
+        delta = (dof * out.float()).sum(-1, keepdim=True)  # [B,H,S,1]
+        dq = torch.zeros_like(qf)
+        dk = torch.zeros_like(kf)
+        dv = torch.zeros_like(vf)
+        for i0 in range(0, S, FlashAttentionFunction.CHUNK):
+            i1 = min(S, i0 + FlashAttentionFunction.CHUNK)
+            qi = qf[..., i0:i1, :]
+            si = torch.matmul(qi, kf.transpose(-1, -2)) * scale
+            if causal:
+                rows = torch.arange(i0, i1, device=q.device).unsqueeze(-1)
+                cols = torch.arange(S, device=q.device)
+                si = si.masked_fill(cols > rows, float("-inf"))
+            p = torch.exp(si - lse[..., i0:i1].unsqueeze(-1))
+            doi = dof[..., i0:i1, :]
+            dv += torch.matmul(p.transpose(-1, -2), doi)
+            dp = torch.matmul(doi, vf.transpose(-1, -2))
+            ds = p * (dp - delta[..., i0:i1, :])
+            dq[..., i0:i1, :] = torch.matmul(ds, kf) * scale
+            dk += torch.matmul(ds.transpose(-1, -2), qi) * scale
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+
+
+def flash_attention(q, k, v, causal=False, scale=None):
+    """Differentiable flash attention; returns the attention output."""
+    out, _ = FlashAttentionFunction.apply(q, k, v, causal, scale)
+    return out
