@@ -17,7 +17,11 @@ import torch.nn as nn
 
 from ..normalization import FusedLayerNorm, FusedRMSNorm
 from ..fused_dense import fused_dense_function, fused_dense_gelu_dense_function
-from ..transformer import scaled_masked_softmax, scaled_upper_triang_masked_softmax
+from ..transformer import (
+    fused_apply_rotary_pos_emb,
+    scaled_masked_softmax,
+    scaled_upper_triang_masked_softmax,
+)
 
 
 @dataclass
@@ -197,3 +201,100 @@ class TransformerLargeModel(nn.Module):
             delta = mlp(n2)
         x, _ = fused_add_norm(x, delta, self.final_norm)
         return torch.matmul(x.transpose(0, 1), self.tok_emb.weight.t())
+
+
+# ---------------- LLaMA-style decoder (RMSNorm + RoPE + SwiGLU) ----------------
+
+def llama_small_config(seq_len=512):
+    """~400M-class LLaMA-shape config for the bench harness."""
+    return TransformerLMConfig(vocab_size=32000, hidden=1024, layers=16, heads=16,
+                               seq_len=seq_len, ffn_hidden=2816, causal=True,
+                               norm="rmsnorm")
+
+
+class LlamaAttention(nn.Module):
+    """Causal self-attention with fused RoPE on q/k (no biases)."""
+
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        self.nh = cfg.heads
+        self.hd = cfg.hidden // cfg.heads
+        self.qkv_w = nn.Parameter(torch.empty(3 * cfg.hidden, cfg.hidden))
+        self.proj_w = nn.Parameter(torch.empty(cfg.hidden, cfg.hidden))
+        nn.init.normal_(self.qkv_w, std=0.02)
+        nn.init.normal_(self.proj_w, std=0.02)
+
+    def forward(self, x, freqs):
+        b, s, h = x.shape
+        qkv = torch.nn.functional.linear(x, self.qkv_w)
+        qkv = qkv.view(b, s, 3, self.nh, self.hd)
+        # fused RoPE expects [s, b, h, d]
+        q = fused_apply_rotary_pos_emb(qkv[:, :, 0].transpose(0, 1).contiguous(), freqs)
+        k = fused_apply_rotary_pos_emb(qkv[:, :, 1].transpose(0, 1).contiguous(), freqs)
+        q = q.permute(1, 2, 0, 3)                       # [b, nh, s, hd]
+        k = k.permute(1, 2, 0, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)            # [b, nh, s, hd]
+        scores = torch.matmul(q, k.transpose(-2, -1))
+        probs = scaled_upper_triang_masked_softmax(
+            scores.reshape(b * self.nh, s, s), 1.0 / math.sqrt(self.hd))
+        ctx = torch.matmul(probs.view(b, self.nh, s, s), v)
+        ctx = ctx.transpose(1, 2).reshape(b, s, h)
+        return torch.nn.functional.linear(ctx, self.proj_w)
+
+
+class LlamaMLP(nn.Module):
+    """SwiGLU: down( silu(gate(x)) * up(x) ), no biases."""
+
+    def __init__(self, cfg: TransformerLMConfig):
+        super().__init__()
+        self.gate = nn.Parameter(torch.empty(cfg.ffn_hidden, cfg.hidden))
+        self.up = nn.Parameter(torch.empty(cfg.ffn_hidden, cfg.hidden))
+        self.down = nn.Parameter(torch.empty(cfg.hidden, cfg.ffn_hidden))
+        for w in (self.gate, self.up, self.down):
+            nn.init.normal_(w, std=0.02)
+
+    def forward(self, x):
+        f = torch.nn.functional
+        return f.linear(f.silu(f.linear(x, self.gate)) * f.linear(x, self.up), self.down)
+
+
+class LlamaModel(nn.Module):
+    """Decoder-only LM: FusedRMSNorm (adds fused into the next norm), fused
+    RoPE, causal wave64 softmax, SwiGLU MLP, tied LM head."""
+
+    def __init__(self, cfg=None):
+        super().__init__()
+        from ..normalization import fused_add_norm
+
+        self._fused_add_norm = fused_add_norm
+        cfg = cfg or llama_small_config()
+        self.cfg = cfg
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.attns = nn.ModuleList([LlamaAttention(cfg) for _ in range(cfg.layers)])
+        self.mlps = nn.ModuleList([LlamaMLP(cfg) for _ in range(cfg.layers)])
+        self.ln1 = nn.ModuleList([FusedRMSNorm(cfg.hidden) for _ in range(cfg.layers)])
+        self.ln2 = nn.ModuleList([FusedRMSNorm(cfg.hidden) for _ in range(cfg.layers)])
+        self.final_norm = FusedRMSNorm(cfg.hidden)
+        nn.init.normal_(self.tok_emb.weight, std=0.02)
+        hd = cfg.hidden // cfg.heads
+        inv = 1.0 / (10000.0 ** (torch.arange(0, hd, 2).float() / hd))
+        ang = torch.outer(torch.arange(cfg.seq_len).float(), inv)  # [s, hd/2]
+        self.register_buffer("rope_freqs",
+                             torch.cat([ang, ang], dim=-1).view(cfg.seq_len, 1, 1, hd),
+                             persistent=False)
+
+    def forward(self, tokens):
+        b, s = tokens.shape
+        freqs = self.rope_freqs[:s]
+        x = self.tok_emb(tokens)
+        delta = None
+        for attn, mlp, l1, l2 in zip(self.attns, self.mlps, self.ln1, self.ln2):
+            if delta is None:
+                n1 = l1(x)
+            else:
+                n1, x = self._fused_add_norm(x, delta, l1)
+            delta = attn(n1, freqs)
+            n2, x = self._fused_add_norm(x, delta, l2)
+            delta = mlp(n2)
+        x, _ = self._fused_add_norm(x, delta, self.final_norm)
+        return torch.matmul(x, self.tok_emb.weight.t())

@@ -63,7 +63,7 @@ def main():
     ap.add_argument("--image-size", type=int, default=224)
     ap.add_argument("--skip-adam-bench", action="store_true")
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "bert", "gpt2", "transformer_lg"],
+                    choices=["resnet50", "bert", "gpt2", "transformer_lg", "llama"],
                     help="resnet50 = headline config #2; bert/gpt2 = BASELINE configs #3/#4")
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--device", default="cuda", choices=["cuda", "cpu"],
@@ -131,8 +131,9 @@ def main():
         # scaled_masked_softmax) / GPT-2 345M (fused_dense GEMM+bias+GELU +
         # FusedRMSNorm + FusedLAMB + causal softmax), token throughput.
         from apex_amd.models.transformer import (
-            BertModel, GPTModel, TransformerLargeModel,
-            bert_base_config, gpt2_345m_config, transformer_large_config,
+            BertModel, GPTModel, LlamaModel, TransformerLargeModel,
+            bert_base_config, gpt2_345m_config, llama_small_config,
+            transformer_large_config,
         )
         from apex_amd.optimizers import FusedAdam, FusedLAMB
         from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
@@ -142,6 +143,11 @@ def main():
             model = BertModel(cfg).to(device)
             opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
             config_model = "bert-base"
+        elif args.model == "llama":
+            cfg = llama_small_config(seq_len=args.seq_len)
+            model = LlamaModel(cfg).to(device)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            config_model = "llama-small(rope+swiglu+rmsnorm)"
         elif args.model == "transformer_lg":
             cfg = transformer_large_config(seq_len=args.seq_len)
             model = TransformerLargeModel(cfg).to(device)
@@ -229,7 +235,7 @@ def main():
                 "seq_len": None if args.model == "resnet50" else args.seq_len,
                 "amp": "O1-bf16" if args.model == "resnet50" else "O2-bf16",
                 "optimizer": {"resnet50": "FusedSGD(momentum=0.9)", "bert": "FusedAdam",
-                              "gpt2": "FusedLAMB",
+                              "gpt2": "FusedLAMB", "llama": "FusedAdam",
                               "transformer_lg": "FusedAdam"}[args.model],
                 "syncbn": use_syncbn,
                 "parallelism": f"dp{world}",

@@ -31,7 +31,7 @@ def train_steps(model, opt, steps=5, seed=1):
         with amp.scale_loss(loss, opt) as scaled:
             scaled.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     return losses
 
 
