@@ -1,4 +1,11 @@
 from .resnet import ResNet, resnet50
-from .transformer import BertModel, GPTModel, TransformerLMConfig
+from .transformer import (
+    BertModel,
+    GPTModel,
+    LlamaModel,
+    TransformerLargeModel,
+    TransformerLMConfig,
+)
 
-__all__ = ["ResNet", "resnet50", "BertModel", "GPTModel", "TransformerLMConfig"]
+__all__ = ["ResNet", "resnet50", "BertModel", "GPTModel", "LlamaModel",
+           "TransformerLargeModel", "TransformerLMConfig"]

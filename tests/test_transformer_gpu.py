@@ -277,3 +277,29 @@ def test_wgrad_gemm_accum_fp32_gpu(dtype):
     wgrad_gemm_accum_fp32(x, dy, main)
     torch.cuda.synchronize()
     torch.testing.assert_close(main, expected, rtol=1e-2, atol=1e-2)
+
+
+def test_llama_model_gpu_step():
+    """LLaMA-style model: fused RoPE + RMSNorm(+add) + causal softmax — all
+    individually-validated kernels composed into one training step."""
+    from apex_amd.models.transformer import LlamaModel, TransformerLMConfig
+    from apex_amd.optimizers import FusedAdam
+
+    cfg = TransformerLMConfig(vocab_size=512, hidden=256, layers=2, heads=4,
+                              seq_len=64, ffn_hidden=384, causal=True, norm="rmsnorm")
+    torch.manual_seed(0)
+    m = LlamaModel(cfg).cuda().bfloat16()
+    m.rope_freqs = m.rope_freqs.float()  # rope tables stay fp32
+    opt = FusedAdam(m.parameters(), lr=1e-3)
+    tokens = torch.randint(0, 512, (2, 64), device="cuda")
+    losses = []
+    for _ in range(8):
+        opt.zero_grad()
+        logits = m(tokens)
+        loss = torch.nn.functional.cross_entropy(
+            logits.float().reshape(-1, 512), tokens.reshape(-1))
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0]
+    assert all(l == l for l in losses)  # no NaNs
