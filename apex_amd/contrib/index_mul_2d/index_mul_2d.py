@@ -31,6 +31,10 @@ class IndexMul2d_(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         in1, in2, idx1 = ctx.saved_tensors
+        if torch.is_grad_enabled():  # create_graph: differentiable composition
+            grad_in1 = torch.zeros_like(in1).index_add(0, idx1, grad_out * in2)
+            grad_in2 = in1.index_select(0, idx1) * grad_out
+            return grad_in1, grad_in2, None
         ext = get_ext("index_mul_2d")
         grad_in1, grad_in2 = ext.backward(in1, in2, idx1, grad_out.contiguous())
         return grad_in1, grad_in2, None

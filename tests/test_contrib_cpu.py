@@ -257,3 +257,16 @@ def test_asp_allow_permutation_end_to_end():
     zeros_per_group = (w.reshape(w.shape[0], -1, 4) == 0).sum(-1)
     assert (zeros_per_group >= 2).all()
     ASP._reset()
+
+
+def test_index_mul_2d_double_backward_cpu():
+    from apex_amd.contrib.index_mul_2d import index_mul_2d
+
+    torch.manual_seed(0)
+    in1 = torch.randn(6, 8, requires_grad=True)
+    in2 = torch.randn(4, 8, requires_grad=True)
+    idx = torch.tensor([0, 2, 2, 5])
+    out = index_mul_2d(in1, in2, idx)
+    g1, = torch.autograd.grad(out.sum(), in1, create_graph=True)
+    gg, = torch.autograd.grad(g1.sum(), in2)
+    torch.testing.assert_close(gg, torch.ones_like(in2))

@@ -106,3 +106,19 @@ def test_clip_grad_fused_gpu():
     torch.testing.assert_close(n1, n2, rtol=1e-5, atol=1e-6)
     for p1, p2 in zip(ps1, ps2):
         torch.testing.assert_close(p1.grad, p2.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_index_mul_2d_double_backward_gpu():
+    from apex_amd.contrib.index_mul_2d import index_mul_2d
+
+    torch.manual_seed(0)
+    in1 = torch.randn(6, 8, device="cuda", requires_grad=True)
+    in2 = torch.randn(4, 8, device="cuda", requires_grad=True)
+    idx = torch.tensor([0, 2, 2, 5], device="cuda")
+    out = index_mul_2d(in1, in2, idx)
+    g1, = torch.autograd.grad(out.sum(), in1, create_graph=True)
+    gg, = torch.autograd.grad(g1.sum(), in2)
+    torch.testing.assert_close(gg, torch.ones_like(in2))
+    # first-order grads still match the eager composition
+    o_ref = in1.index_select(0, idx) * in2
+    torch.testing.assert_close(out, o_ref)
