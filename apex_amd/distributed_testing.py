@@ -114,3 +114,11 @@ class NcclDistributedTestBase(DistributedTestBase):
 
 class GlooDistributedTestBase(DistributedTestBase):
     BACKEND = "gloo"
+
+
+class UccDistributedTestBase(DistributedTestBase):
+    """UCC-backend base (reference: distributed_test_base.py:99). torch-ROCm
+    builds typically ship without UCC — instantiation then fails at
+    init_process_group, same as the reference on non-UCC builds."""
+
+    BACKEND = "ucc"
