@@ -1,0 +1,29 @@
+"""Bitwise determinism on hardware: the deterministic two-stage kernel
+reductions (no fp32 atomics) must give identical loss records across runs."""
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+from test_determinism_cpu import run_harness  # noqa: E402
+
+# apex_amd's own kernels are deterministic by construction (fixed-order
+# two-stage reductions, no fp32 atomics); the remaining variable is MIOpen's
+# deterministic-conv selection, which this pool hasn't confirmed yet —
+# strict=False keeps the check visible without gating the suite on it.
+_soft = pytest.mark.xfail(strict=False,
+                          reason="MIOpen conv determinism unverified on this pool")
+
+
+@_soft
+def test_harness_bitwise_deterministic_gpu_o1():
+    a = run_harness("O1", iters=6, batch=8, image=64)
+    b = run_harness("O1", iters=6, batch=8, image=64)
+    assert a == b
+
+
+@_soft
+def test_harness_bitwise_deterministic_gpu_o2():
+    a = run_harness("O2", iters=6, batch=8, image=64)
+    b = run_harness("O2", iters=6, batch=8, image=64)
+    assert a == b
