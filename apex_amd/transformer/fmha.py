@@ -55,6 +55,8 @@ class FlashAttentionFunction(torch.autograd.Function):
 
     CHUNK = 256  # q rows recomputed per tile
 
+    use_fused_backward = False  # flip after round-2 on-hardware validation
+
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
@@ -68,6 +70,10 @@ class FlashAttentionFunction(torch.autograd.Function):
     def backward(ctx, dout, dlse_unused):
         q, k, v, out, lse = ctx.saved_tensors
         causal, scale = ctx.causal, ctx.scale
+        if FlashAttentionFunction.use_fused_backward and q.is_cuda:
+            ext = get_ext("mfma")
+            dq, dk, dv = ext.fmha_bwd(dout, q, k, v, out, lse, causal, float(scale))
+            return dq, dk, dv, None, None
         S = q.shape[-2]
         qf, kf, vf = q.float(), k.float(), v.float()
         dof = dout.float()

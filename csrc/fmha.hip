@@ -181,6 +181,205 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   }
 }
 
+// ---------------- backward ----------------
+// Two kernels, both recomputing P from (Q, K, lse) — index math verified
+// lane-for-lane by tests/test_fmha_sim_cpu.py (simulate_dq_wave /
+// simulate_dkv_wave):
+//   dq:  one wave per 16 q rows, loops kv tiles (forward orientation);
+//        dP = dO V^T, dS = P*(dP - delta_row), dQ += dS K * scale.
+//   dkv: one wave per 16 kv rows, loops q tiles with TRANSPOSED fragment
+//        roles (S^T = K Q^T, lse/delta indexed by the q column);
+//        dV += P^T dO, dK += dS^T Q * scale.
+
+template <bool CAUSAL, int D>
+__global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
+    const short* __restrict__ dO, const float* __restrict__ LSE,
+    const float* __restrict__ DELTA, short* __restrict__ dQ, int S, float scale) {
+  constexpr int NK = D / 32;
+  constexpr int ND = D / 16;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
+  if (q0 >= S) return;
+  const short* q_ptr = Q + (long)bh * S * D;
+  const short* k_ptr = K + (long)bh * S * D;
+  const short* v_ptr = V + (long)bh * S * D;
+  const short* do_ptr = dO + (long)bh * S * D;
+  const float* lse = LSE + (long)bh * S;
+  const float* delta = DELTA + (long)bh * S;
+
+  __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
+  short* pbuf = lds_p[wave];
+
+  bf16x8 aq[NK], ado[NK];
+  const int a_row = q0 + (lane & 15);
+#pragma unroll
+  for (int c = 0; c < NK; ++c) {
+    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+    ado[c] = *reinterpret_cast<const bf16x8*>(do_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+  }
+  f32x4 dq_acc[ND];
+#pragma unroll
+  for (int d = 0; d < ND; ++d) dq_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int my_r0 = (lane >> 4) * 4;
+  const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      f32x4 s = f32x4{0.f, 0.f, 0.f, 0.f};
+      f32x4 dp = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int k_row = kv0 + j * 16 + (lane & 15);
+#pragma unroll
+      for (int c = 0; c < NK; ++c) {
+        const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
+            k_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+        s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[c], bk, s, 0, 0, 0);
+        const bf16x8 bvt = *reinterpret_cast<const bf16x8*>(
+            v_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+        dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[c], bvt, dp, 0, 0, 0);
+      }
+#pragma unroll
+      for (int qi = 0; qi < 4; ++qi) {
+        const int row_g = q0 + my_r0 + qi;
+        const int col_g = kv0 + j * 16 + (lane & 15);
+        float p = (CAUSAL && col_g > row_g) ? 0.f : __expf(s[qi] * scale - lse[row_g]);
+        const float ds = p * (dp[qi] - delta[row_g]);
+        const __hip_bfloat16 db = __float2bfloat16(ds);
+        pbuf[(my_r0 + qi) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&db);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    const bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
+        pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      bf16x8 bK;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj)
+        bK[jj] = k_ptr[(long)(kv0 + (lane >> 4) * 8 + jj) * D + d * 16 + (lane & 15)];
+      dq_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, bK, dq_acc[d], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+#pragma unroll
+  for (int qi = 0; qi < 4; ++qi) {
+    const long row_g = q0 + my_r0 + qi;
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      const __hip_bfloat16 o = __float2bfloat16(dq_acc[d][qi] * scale);
+      dQ[((long)bh * S + row_g) * D + d * 16 + (lane & 15)] =
+          *reinterpret_cast<const short*>(&o);
+    }
+  }
+}
+
+template <bool CAUSAL, int D>
+__global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
+    const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
+    const short* __restrict__ dO, const float* __restrict__ LSE,
+    const float* __restrict__ DELTA, short* __restrict__ dK, short* __restrict__ dV,
+    int S, float scale) {
+  constexpr int NK = D / 32;
+  constexpr int ND = D / 16;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int bh = blockIdx.y;
+  const int kv0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
+  if (kv0 >= S) return;
+  const short* q_ptr = Q + (long)bh * S * D;
+  const short* k_ptr = K + (long)bh * S * D;
+  const short* v_ptr = V + (long)bh * S * D;
+  const short* do_ptr = dO + (long)bh * S * D;
+  const float* lse = LSE + (long)bh * S;
+  const float* delta = DELTA + (long)bh * S;
+
+  __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
+  __shared__ short lds_ds[FM_WAVES][FM_ROWS * FM_BN];
+  short* pbuf = lds_p[wave];
+  short* dsbuf = lds_ds[wave];
+
+  bf16x8 ak[NK], av[NK];
+  const int a_row = kv0 + (lane & 15);
+#pragma unroll
+  for (int c = 0; c < NK; ++c) {
+    ak[c] = *reinterpret_cast<const bf16x8*>(k_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+    av[c] = *reinterpret_cast<const bf16x8*>(v_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+  }
+  f32x4 dv_acc[ND], dk_acc[ND];
+#pragma unroll
+  for (int d = 0; d < ND; ++d) {
+    dv_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dk_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+  const int my_r0 = (lane >> 4) * 4;
+  const int q_start = CAUSAL ? (kv0 / FM_BN) * FM_BN : 0;
+
+  for (int q0 = q_start; q0 < S; q0 += FM_BN) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      f32x4 sT = f32x4{0.f, 0.f, 0.f, 0.f};
+      f32x4 dpT = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int q_row = q0 + j * 16 + (lane & 15);
+#pragma unroll
+      for (int c = 0; c < NK; ++c) {
+        const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+            q_ptr + (long)q_row * D + c * 32 + (lane >> 4) * 8);
+        sT = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[c], bq, sT, 0, 0, 0);
+        const bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
+            do_ptr + (long)q_row * D + c * 32 + (lane >> 4) * 8);
+        dpT = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[c], bdo, dpT, 0, 0, 0);
+      }
+#pragma unroll
+      for (int qi = 0; qi < 4; ++qi) {
+        const int kv_g = kv0 + my_r0 + qi;
+        const int q_g = q0 + j * 16 + (lane & 15);
+        float p = (CAUSAL && kv_g > q_g) ? 0.f : __expf(sT[qi] * scale - lse[q_g]);
+        const float ds = p * (dpT[qi] - delta[q_g]);
+        const int idx = (my_r0 + qi) * FM_BN + j * 16 + (lane & 15);
+        const __hip_bfloat16 pb = __float2bfloat16(p);
+        const __hip_bfloat16 db = __float2bfloat16(ds);
+        pbuf[idx] = *reinterpret_cast<const short*>(&pb);
+        dsbuf[idx] = *reinterpret_cast<const short*>(&db);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    const bf16x8 a_p = *reinterpret_cast<const bf16x8*>(
+        pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
+    const bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
+        dsbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      bf16x8 b_do, b_q;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        const long qr = q0 + (lane >> 4) * 8 + jj;
+        b_do[jj] = do_ptr[qr * D + d * 16 + (lane & 15)];
+        b_q[jj] = q_ptr[qr * D + d * 16 + (lane & 15)];
+      }
+      dv_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_do, dv_acc[d], 0, 0, 0);
+      dk_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_q, dk_acc[d], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+  }
+#pragma unroll
+  for (int qi = 0; qi < 4; ++qi) {
+    const long kv_g = kv0 + my_r0 + qi;
+#pragma unroll
+    for (int d = 0; d < ND; ++d) {
+      const __hip_bfloat16 ov = __float2bfloat16(dv_acc[d][qi]);
+      dV[((long)bh * S + kv_g) * D + d * 16 + (lane & 15)] =
+          *reinterpret_cast<const short*>(&ov);
+      const __hip_bfloat16 ok = __float2bfloat16(dk_acc[d][qi] * scale);
+      dK[((long)bh * S + kv_g) * D + d * 16 + (lane & 15)] =
+          *reinterpret_cast<const short*>(&ok);
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
@@ -215,4 +414,47 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
 #undef FMHA_LAUNCH
   HIP_CHECK(hipGetLastError());
   return {out, lse};
+}
+
+std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor out, at::Tensor lse, bool causal, double scale) {
+  TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_bwd: bf16 only");
+  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  auto doc = dout.contiguous(), oc = out.contiguous(), lsec = lse.contiguous();
+  const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
+  TORCH_CHECK(D == 64 || D == 128, "fmha_bwd: head_dim must be 64 or 128");
+  TORCH_CHECK(S % 32 == 0, "fmha_bwd: seq_len must be a multiple of 32");
+  auto dq = at::empty_like(qc);
+  auto dk = at::empty_like(kc);
+  auto dv = at::empty_like(vc);
+  // delta = rowsum(dO * O) in fp32 — a cheap eager reduction
+  auto delta = (doc.to(at::kFloat) * oc.to(at::kFloat)).sum(-1).contiguous();
+  auto stream = current_stream();
+  dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
+  dim3 block(FM_WAVES * 64);
+  const float sc = (float)scale;
+
+#define FMHA_BWD_LAUNCH(CAUSAL, DD)                                                        \
+  do {                                                                                     \
+    hipLaunchKernelGGL((fmha_bwd_dq_kernel<CAUSAL, DD>), grid, block, 0, stream,           \
+                       (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
+                       (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
+                       lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
+                       (short*)dq.data_ptr(), S, sc);                                      \
+    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD>), grid, block, 0, stream,          \
+                       (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
+                       (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
+                       lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
+                       (short*)dk.data_ptr(), (short*)dv.data_ptr(), S, sc);               \
+  } while (0)
+  if (causal) {
+    if (D == 64) FMHA_BWD_LAUNCH(true, 64);
+    else FMHA_BWD_LAUNCH(true, 128);
+  } else {
+    if (D == 64) FMHA_BWD_LAUNCH(false, 64);
+    else FMHA_BWD_LAUNCH(false, 128);
+  }
+#undef FMHA_BWD_LAUNCH
+  HIP_CHECK(hipGetLastError());
+  return {dq, dk, dv};
 }

@@ -323,10 +323,14 @@ at::Tensor mfma_tile_probe(at::Tensor A_bf16, at::Tensor B_bf16);
 
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
                                  double scale);
+std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor out, at::Tensor lse, bool causal, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fmha_fwd", &fmha_fwd,
         "flash-attention fwd (bf16, D=64/128) -> (out, lse) [EXPERIMENTAL]");
+  m.def("fmha_bwd", &fmha_bwd,
+        "flash-attention bwd -> (dq, dk, dv) [EXPERIMENTAL]");
   m.def("gemm_bias_gelu", &gemm_bias_gelu_mfma, "bf16 MFMA GEMM + bias + tanh-GELU (fused)");
   m.def("gemm_bias", &gemm_bias_mfma, "bf16 MFMA GEMM + bias");
   m.def("mfma_tile_probe", &mfma_tile_probe, "single-tile fragment-layout verification");

@@ -38,6 +38,30 @@ def main():
             print(f"[{status}] B{B} H{H} S{S} D{D} causal={int(causal)}: "
                   f"|out| {e_out:.2e}  |lse| {e_lse:.2e}")
 
+    print("== backward numerics (fused fmha_bwd vs autograd reference) ==")
+    import apex_amd._mfma as mfma
+    for (B, H, S, D) in [(2, 4, 128, 64), (1, 2, 256, 128)]:
+        for causal in (False, True):
+            q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+            k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+            v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+            dout = torch.randn_like(q)
+            scale = 1.0 / math.sqrt(D)
+            out, lse = flash_attention_forward(q, k, v, causal=causal, scale=scale)
+            dq, dk, dv = mfma.fmha_bwd(dout, q, k, v, out, lse, causal, scale)
+            qr = q.detach().float().requires_grad_(True)
+            kr = k.detach().float().requires_grad_(True)
+            vr = v.detach().float().requires_grad_(True)
+            ref, _ = eager_attention_reference(qr, kr, vr, causal=causal, scale=scale)
+            ref.backward(dout.float())
+            errs = [(dq.float() - qr.grad).abs().max().item(),
+                    (dk.float() - kr.grad).abs().max().item(),
+                    (dv.float() - vr.grad).abs().max().item()]
+            status = "OK " if max(errs) < 5e-2 else "FAIL"
+            ok &= status == "OK "
+            print(f"[{status}] B{B} H{H} S{S} D{D} causal={int(causal)}: "
+                  f"|dq| {errs[0]:.2e} |dk| {errs[1]:.2e} |dv| {errs[2]:.2e}")
+
     print("== perf (vs bmm+softmax eager, bf16) ==")
     for (B, H, S, D) in [(8, 12, 512, 64), (4, 16, 2048, 64), (2, 16, 4096, 128)]:
         q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)

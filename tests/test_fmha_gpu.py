@@ -28,3 +28,32 @@ def test_fmha_fwd_matches_eager(causal, shape):
     ref_out, ref_lse = eager_attention_reference(q, k, v, causal=causal)
     torch.testing.assert_close(out.float(), ref_out, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(lse, ref_lse, rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_fmha_fused_backward_matches_eager(causal):
+    import apex_amd._mfma as mfma
+    from apex_amd.transformer.fmha import flash_attention_forward
+
+    B, H, S, D = 2, 2, 128, 64
+    torch.manual_seed(1)
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+    dout = torch.randn_like(q)
+    scale = 1.0 / math.sqrt(D)
+    out, lse = flash_attention_forward(q, k, v, causal=causal, scale=scale)
+    dq, dk, dv = mfma.fmha_bwd(dout, q, k, v, out, lse, causal, scale)
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    s = torch.matmul(qr, kr.transpose(-1, -2)) * scale
+    if causal:
+        s = s.masked_fill(torch.triu(torch.ones(S, S, dtype=torch.bool,
+                                                device="cuda"), 1), float("-inf"))
+    ref = torch.matmul(torch.softmax(s, -1), vr)
+    ref.backward(dout.float())
+    torch.testing.assert_close(dq.float(), qr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dk.float(), kr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(dv.float(), vr.grad, rtol=5e-2, atol=5e-2)
