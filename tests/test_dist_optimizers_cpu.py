@@ -396,3 +396,25 @@ def _overlap_param_sync_worker(rank, world_size):
 
 def test_dist_adam_overlap_param_sync():
     run_distributed(_overlap_param_sync_worker, world_size=2)
+
+
+def test_param_remainder_bit_roundtrip_all_patterns():
+    """(bf16 << 16 | int16) split/reconstruct must be bit-exact for ANY fp32
+    pattern — including negatives, denormals, inf and NaN payloads."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    m = torch.nn.Linear(64, 64, bias=False).to(torch.bfloat16)
+    opt = DistributedFusedAdam(m.parameters(), lr=1e-3, bucket_cap_mb=1,
+                               store_param_remainders=True)
+    b = opt.buckets[0]
+    torch.manual_seed(0)
+    bits = torch.randint(-2**63, 2**63 - 1, (b.shard_size,), dtype=torch.int64)
+    master = bits.to(torch.int32).view(torch.float32).clone()
+    # add targeted edge patterns
+    edge = torch.tensor([0x7F800000, 0xFF800000, 0x7FC00001, 0x00000001,
+                         0x80000001, 0x00008000, 0x0000FFFF, 0xFFFFFFFF],
+                        dtype=torch.int64).to(torch.int32).view(torch.float32)
+    master[:edge.numel()] = edge
+    opt._set_master(b, master.clone())
+    back = opt._get_master(b)
+    assert torch.equal(master.view(torch.int32), back.view(torch.int32))
