@@ -134,6 +134,17 @@ class MultiStreamGraphModule:
         self.num_streams = num_streams
         self._streams = None
         self.compile_partitions = compile_partitions
+        # intermediate lifetimes: free env entries after their last consumer
+        # so peak memory matches eager execution (record_stream at the
+        # consumption site keeps the caching allocator cross-stream safe)
+        last_user = {}
+        for node in gm.graph.nodes:
+            torch.fx.map_arg((node.args, node.kwargs),
+                             lambda n, node=node: last_user.__setitem__(n, node))
+        self._dies_after = {}
+        for n, u in last_user.items():
+            if u.op != "output":  # outputs must survive the call
+                self._dies_after.setdefault(u, []).append(n)
         if compile_partitions:
             self._compiled = []
             for p in self.partitions:
@@ -181,6 +192,9 @@ class MultiStreamGraphModule:
                 res = fn(*[env[n] for n in ext])
                 for n, v in zip(outs, res):
                     env[n] = v
+                for node in p.nodes:
+                    for dead in self._dies_after.get(node, ()):
+                        env.pop(dead, None)
             return torch.fx.map_arg(out_node.args[0], lambda n: env[n])
 
         self._ensure_streams()
@@ -213,6 +227,9 @@ class MultiStreamGraphModule:
             for n, v in zip(outs, res):
                 env[n] = v
                 prod_stream[n] = p.stream
+            for node in p.nodes:
+                for dead in self._dies_after.get(node, ()):
+                    env.pop(dead, None)
             p.event = torch.cuda.Event()
             p.event.record(p.stream)
         for p in self.partitions:
@@ -291,6 +308,8 @@ class MultiStreamGraphModule:
                 elif node.op == "call_module":
                     env[node] = interp.fetch_attr(node.target)(*a, **kw)
             prod_stream[node] = p.stream
+            for dead in self._dies_after.get(node, ()):
+                env.pop(dead, None)
             if node is p.nodes[-1]:
                 p.event = torch.cuda.Event()
                 p.event.record(p.stream)
