@@ -30,7 +30,6 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 constexpr int FM_ROWS = 16;   // q rows per wave
 constexpr int FM_WAVES = 4;   // waves per workgroup
 constexpr int FM_BN = 32;     // kv tile (one MFMA K step for P@V)
-constexpr int FM_MAXD = 128;
 
 __device__ __forceinline__ float row_reduce_max16(float v) {
 #pragma unroll
