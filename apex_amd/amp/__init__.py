@@ -456,11 +456,17 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
 
     if loss_scale == 1.0 and not scaler.dynamic:
         yield loss
+        if delay_unscale:
+            return
         # still materialize masters for O2 with static scale 1.0
         if _amp_state.opt_properties.master_weights:
             for opt in opt_list:
                 model_grads, master_grads = _materialize_master_grads(opt, 1.0)
-                scaler.unscale_grads(model_grads, master_grads)
+                ov = scaler.unscale_grads(model_grads, master_grads)
+                for p in opt._amp_stash.all_fp16_params:
+                    p.grad = None  # consumed into masters (see dynamic path)
+                if ov:
+                    opt._amp_skip_next_step = True
         return
 
     yield loss.float() * loss_scale

@@ -218,3 +218,27 @@ def test_amp_register_function_patches_module():
     ns = _types.SimpleNamespace(f=lambda x: x.dtype)
     amp.register_float_function(ns, "f")
     assert ns.f(torch.randn(2).bfloat16()) == torch.float32
+
+
+def test_o2_static_scale1_clears_model_grads_and_skips_on_overflow():
+    model = make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O2",
+                                cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
+    x = torch.randn(4, 16).bfloat16()
+    loss = model(x).float().sum()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    # model grads consumed into masters even on the scale==1 fast path
+    assert all(p.grad is None for p in opt._amp_stash.all_fp16_params)
+    opt.step()
+
+    # overflow with static scale must still skip the step
+    before = [p.detach().clone() for p in amp.master_params(opt)]
+    loss = (model(torch.full((4, 16), 1e30, dtype=torch.bfloat16)).float().sum() * 1e30)
+    opt.zero_grad()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    opt.step()
+    for b, a in zip(before, amp.master_params(opt)):
+        torch.testing.assert_close(b, a)
