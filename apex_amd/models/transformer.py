@@ -213,7 +213,14 @@ def llama_small_config(seq_len=512):
 
 
 class LlamaAttention(nn.Module):
-    """Causal self-attention with fused RoPE on q/k (no biases)."""
+    """Causal self-attention with fused RoPE on q/k (no biases).
+
+    ``use_flash=True`` routes the attention core through
+    ``transformer.flash_attention`` (no S x S matrix) instead of
+    bmm + causal wave64 softmax — flip once the experimental fmha kernels
+    are validated on hardware (ROADMAP round 2)."""
+
+    use_flash = False
 
     def __init__(self, cfg: TransformerLMConfig):
         super().__init__()
@@ -234,10 +241,16 @@ class LlamaAttention(nn.Module):
         q = q.permute(1, 2, 0, 3)                       # [b, nh, s, hd]
         k = k.permute(1, 2, 0, 3)
         v = qkv[:, :, 2].permute(0, 2, 1, 3)            # [b, nh, s, hd]
-        scores = torch.matmul(q, k.transpose(-2, -1))
-        probs = scaled_upper_triang_masked_softmax(
-            scores.reshape(b * self.nh, s, s), 1.0 / math.sqrt(self.hd))
-        ctx = torch.matmul(probs.view(b, self.nh, s, s), v)
+        if LlamaAttention.use_flash:
+            from ..transformer import flash_attention
+
+            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  causal=True, scale=1.0 / math.sqrt(self.hd))
+        else:
+            scores = torch.matmul(q, k.transpose(-2, -1))
+            probs = scaled_upper_triang_masked_softmax(
+                scores.reshape(b * self.nh, s, s), 1.0 / math.sqrt(self.hd))
+            ctx = torch.matmul(probs.view(b, self.nh, s, s), v)
         ctx = ctx.transpose(1, 2).reshape(b, s, h)
         return torch.nn.functional.linear(ctx, self.proj_w)
 

@@ -207,3 +207,24 @@ def test_llama_model_cpu_train_step():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0]  # memorizes the fixed batch
+
+
+def test_llama_flash_path_matches_softmax_path_cpu():
+    """The flash-attention route must be output-equivalent to the bmm+causal
+    softmax route (CPU: flash uses its eager-composition fallback)."""
+    from apex_amd.models.transformer import (
+        LlamaAttention, LlamaModel, TransformerLMConfig,
+    )
+
+    cfg = TransformerLMConfig(vocab_size=128, hidden=64, layers=2, heads=4, seq_len=16,
+                              ffn_hidden=96, causal=True, norm="rmsnorm")
+    torch.manual_seed(3)
+    m = LlamaModel(cfg)
+    tokens = torch.randint(0, 128, (2, 16))
+    base = m(tokens)
+    try:
+        LlamaAttention.use_flash = True
+        flash = m(tokens)
+    finally:
+        LlamaAttention.use_flash = False
+    torch.testing.assert_close(flash, base, rtol=1e-5, atol=1e-5)
