@@ -270,3 +270,19 @@ def test_index_mul_2d_double_backward_cpu():
     g1, = torch.autograd.grad(out.sum(), in1, create_graph=True)
     gg, = torch.autograd.grad(g1.sum(), in2)
     torch.testing.assert_close(gg, torch.ones_like(in2))
+
+
+def test_exhaustive_search_monotone_random():
+    """Property: the found permutation never reduces 2:4 kept magnitude, on
+    arbitrary random matrices (not just crafted ones)."""
+    from apex_amd.contrib.sparsity.permutation_search import efficacy, exhaustive_search
+
+    for seed in range(5):
+        torch.manual_seed(seed)
+        rows = 8 + seed * 3
+        cols = 16 + 8 * (seed % 3)
+        w = torch.randn(rows, cols) * torch.rand(1, cols).exp()
+        base = efficacy(w)
+        perm = exhaustive_search(w)
+        assert sorted(perm.tolist()) == list(range(cols))
+        assert efficacy(w[:, perm]) >= base - 1e-5
