@@ -73,3 +73,25 @@ def test_compiled_partitions_cpu_matches_eager():
     # second call reuses the compiled callables
     x2 = torch.randn(4, 16)
     torch.testing.assert_close(ms(x2), m(x2), rtol=1e-5, atol=1e-6)
+
+
+def test_lifetime_freeing_preserves_shared_values():
+    """A value consumed by several partitions AND returned as an output must
+    survive until its last use; earlier frees must not corrupt results."""
+
+    class Shared(torch.nn.Module):
+        def forward(self, x):
+            y = torch.relu(x)          # shared producer, also returned
+            a = torch.tanh(y)
+            b = torch.sigmoid(y)
+            return a + b, y
+
+    torch.manual_seed(2)
+    m = Shared()
+    gm = torch.fx.symbolic_trace(m)
+    ms = MultiStreamGraphModule(gm)
+    x = torch.randn(4, 8)
+    out = ms(x)
+    ref = m(x)
+    torch.testing.assert_close(out[0], ref[0])
+    torch.testing.assert_close(out[1], ref[1])
