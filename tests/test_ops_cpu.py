@@ -228,3 +228,44 @@ def test_llama_flash_path_matches_softmax_path_cpu():
     finally:
         LlamaAttention.use_flash = False
     torch.testing.assert_close(flash, base, rtol=1e-5, atol=1e-5)
+
+
+def test_rope_thd_cpu_matches_sbhd_per_segment():
+    from apex_amd.transformer import (
+        fused_apply_rotary_pos_emb, fused_apply_rotary_pos_emb_thd,
+    )
+
+    torch.manual_seed(0)
+    h, d = 4, 32
+    cu = torch.tensor([0, 5, 12, 20], dtype=torch.int32)
+    total = int(cu[-1])
+    t = torch.randn(total, h, d, requires_grad=True)
+    freqs = torch.randn(20, 1, 1, d)
+    out = fused_apply_rotary_pos_emb_thd(t, cu, freqs)
+    for i in range(cu.numel() - 1):
+        s0, s1 = int(cu[i]), int(cu[i + 1])
+        seg = t.detach()[s0:s1].unsqueeze(1)  # [s, 1, h, d]
+        ref = fused_apply_rotary_pos_emb(seg, freqs[: s1 - s0]).squeeze(1)
+        torch.testing.assert_close(out[s0:s1], ref)
+    out.sum().backward()
+    assert t.grad is not None
+
+
+def test_rope_2d_cpu_invariants():
+    from apex_amd.transformer import fused_apply_rotary_pos_emb_2d
+
+    torch.manual_seed(1)
+    b, H, W, h, d = 2, 4, 6, 3, 16
+    t = torch.randn(b, H * W, h, d)
+    # zero angles -> identity
+    zc = torch.ones(1, 8, 1, d // 2)
+    zs = torch.zeros(1, 8, 1, d // 2)
+    out = fused_apply_rotary_pos_emb_2d(t.view(b, H, W, h, d), H, W, zc, zs, zc, zs)
+    torch.testing.assert_close(out, t)
+    # arbitrary angles preserve rowwise L2 norm (pure rotation)
+    ang_h = torch.randn(1, 8, 1, d // 4).repeat(1, 1, 1, 2)
+    ang_w = torch.randn(1, 8, 1, d // 4).repeat(1, 1, 1, 2)
+    out2 = fused_apply_rotary_pos_emb_2d(t.view(b, H, W, h, d), H, W,
+                                         ang_h.cos(), ang_h.sin(),
+                                         ang_w.cos(), ang_w.sin())
+    torch.testing.assert_close(out2.norm(dim=-1), t.norm(dim=-1), rtol=1e-5, atol=1e-5)
