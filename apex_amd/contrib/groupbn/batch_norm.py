@@ -25,8 +25,6 @@ class BatchNorm2d_NHWC(SyncBatchNorm):
                          process_group=process_group, channel_last=True, fuse_relu=fuse_relu)
 
     def forward(self, x, z=None):
-        if z is not None:
-            # bn_add_relu: relu(bn(x) + z)
-            out = SyncBatchNorm.forward(self, x)
-            return torch.relu(out + z)
-        return SyncBatchNorm.forward(self, x)
+        # bn_add_relu semantics relu(bn(x) + z) are handled inside
+        # SyncBatchnormFunction (z added before the fused ReLU, grad_z in bwd)
+        return SyncBatchNorm.forward(self, x, z)

@@ -77,7 +77,7 @@ def _to_channels_second(x, channel_last):
 
 class SyncBatchnormFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, input, weight, bias, running_mean, running_var, eps, track_running_stats,
+    def forward(ctx, input, weight, bias, z, running_mean, running_var, eps, track_running_stats,
                 momentum, process_group, channel_last, fuse_relu):
         input = input.contiguous(
             memory_format=torch.contiguous_format
@@ -128,19 +128,28 @@ class SyncBatchnormFunction(torch.autograd.Function):
                 running_mean.mul_(1 - momentum).add_(mean.to(running_mean.dtype), alpha=momentum)
                 running_var.mul_(1 - momentum).add_(unbiased.to(running_var.dtype), alpha=momentum)
 
-        ctx.save_for_backward(input, weight, mean, inv_std)
         ctx.process_group = process_group
         ctx.channel_last = channel_last
         ctx.world_size = world_size
         ctx.total_count = total_count
         ctx.fuse_relu = fuse_relu
+        ctx.has_z = z is not None
 
         if use_kernels:
             syncbn = get_ext("syncbn")
             if channel_last:
-                out = syncbn.batchnorm_forward_c_last(input, mean, inv_std, weight, bias, fuse_relu)
+                # in-kernel ReLU only when there is no residual add: with z the
+                # order is relu(bn(x) + z), so add z before the ReLU here.
+                out = syncbn.batchnorm_forward_c_last(
+                    input, mean, inv_std, weight, bias, fuse_relu and z is None)
+                if z is not None:
+                    out = out.add_(z)
+                    if fuse_relu:
+                        out = out.relu_()
             else:
                 out = syncbn.batchnorm_forward(input, mean, inv_std, weight, bias)
+                if z is not None:
+                    out = out.add_(z)
                 if fuse_relu:
                     out = out.relu_()
         else:
@@ -154,13 +163,26 @@ class SyncBatchnormFunction(torch.autograd.Function):
             if bias is not None:
                 out = out + bias.float().view(shape)
             out = out.to(input.dtype)
+            if z is not None:
+                out = out + z
             if fuse_relu:
                 out = out.relu()
+        if fuse_relu:
+            # the ReLU mask is needed to gate grad_output in backward
+            ctx.save_for_backward(input, weight, mean, inv_std, out)
+        else:
+            ctx.save_for_backward(input, weight, mean, inv_std)
         return out
 
     @staticmethod
     def backward(ctx, grad_output):
-        input, weight, mean, inv_std = ctx.saved_tensors
+        if ctx.fuse_relu:
+            input, weight, mean, inv_std, out = ctx.saved_tensors
+            # gate by the ReLU: units clipped to 0 in forward get zero grad
+            grad_output = grad_output * (out > 0).to(grad_output.dtype)
+        else:
+            input, weight, mean, inv_std = ctx.saved_tensors
+        grad_z = grad_output if ctx.has_z else None
         grad_output = grad_output.contiguous()
         channel_last = ctx.channel_last
         c = input.shape[-1] if channel_last else input.shape[1]
@@ -219,7 +241,8 @@ class SyncBatchnormFunction(torch.autograd.Function):
             grad_weight = grad_weight.to(weight.dtype)
             grad_bias = grad_bias.to(weight.dtype)
 
-        return grad_input, grad_weight, grad_bias, None, None, None, None, None, None, None, None
+        return (grad_input, grad_weight, grad_bias, grad_z,
+                None, None, None, None, None, None, None, None)
 
 
 class SyncBatchNorm(_BatchNorm):
@@ -245,10 +268,11 @@ class SyncBatchNorm(_BatchNorm):
 
         if not self.training and self.track_running_stats and not self.channel_last and z is None:
             # eval mode: plain affine transform with running stats
-            return torch.nn.functional.batch_norm(
+            out = torch.nn.functional.batch_norm(
                 input, self.running_mean, self.running_var, self.weight, self.bias,
                 False, 0.0, self.eps,
             )
+            return out.relu_() if self.fuse_relu else out
         exponential_average_factor = 0.0
         if self.training and self.track_running_stats:
             self.num_batches_tracked += 1
@@ -266,12 +290,14 @@ class SyncBatchNorm(_BatchNorm):
             out = (input - self.running_mean.view(shape)) * inv_std
             if self.affine:
                 out = out * self.weight.view(shape) + self.bias.view(shape)
+            if z is not None:
+                out = out + z
             if self.fuse_relu:
                 out = out.relu()
             return out
 
         return SyncBatchnormFunction.apply(
-            input, self.weight, self.bias, self.running_mean, self.running_var,
+            input, self.weight, self.bias, z, self.running_mean, self.running_var,
             self.eps, self.track_running_stats, exponential_average_factor,
             self.process_group, channel_last, self.fuse_relu,
         )

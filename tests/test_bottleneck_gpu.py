@@ -12,7 +12,9 @@ def test_conv_bias_relu_gpu():
 
     torch.manual_seed(0)
     x = torch.randn(4, 8, 16, 16, device="cuda", requires_grad=True)
-    w = torch.randn(16, 8, 3, 3, device="cuda", requires_grad=True) * 0.1
+    # w/b must be LEAF tensors or .grad stays None (round-1 bug: the `* 0.1`
+    # made w a non-leaf and aborted the driver's --maxfail=1 GPU tier).
+    w = (torch.randn(16, 8, 3, 3, device="cuda") * 0.1).requires_grad_()
     b = torch.randn(1, 16, 1, 1, device="cuda", requires_grad=True)
     y = ConvBiasReLU(x, w, b, 1, 1)
     ref = torch.relu(torch.nn.functional.conv2d(

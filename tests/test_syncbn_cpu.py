@@ -106,3 +106,63 @@ def test_syncbn_single_process_matches_bn():
     torch.testing.assert_close(x.grad, x2.grad, rtol=1e-5, atol=1e-6)
     torch.testing.assert_close(sbn.running_mean, bn.running_mean)
     torch.testing.assert_close(sbn.running_var, bn.running_var)
+
+
+def test_syncbn_fuse_relu_backward():
+    """fuse_relu=True must gate backward grads by the ReLU mask (round-1
+    advisor finding: backward ignored ctx.fuse_relu)."""
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(3)
+    x = torch.randn(6, 4, 3, 3, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    sbn = SyncBatchNorm(4, fuse_relu=True)
+    bn = torch.nn.BatchNorm2d(4)
+    sbn.train()
+    bn.train()
+    with torch.no_grad():
+        bn.weight.copy_(sbn.weight)
+        bn.bias.copy_(sbn.bias)
+    y1 = sbn(x)
+    y2 = torch.relu(bn(x2))
+    torch.testing.assert_close(y1, y2, rtol=1e-5, atol=1e-6)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.bias.grad, bn.bias.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_groupbn_bn_add_relu_semantics():
+    """bn_add_relu computes relu(bn(x) + z) — z added BEFORE the ReLU — and
+    produces grads for x, z, weight, bias (round-1 advisor finding: double
+    ReLU + silently ignored z)."""
+    from apex_amd.contrib.groupbn import BatchNorm2d_NHWC
+
+    torch.manual_seed(4)
+    C = 4
+    x = torch.randn(3, 5, 5, C, requires_grad=True)
+    z = torch.randn(3, 5, 5, C, requires_grad=True)
+    m = BatchNorm2d_NHWC(C, fuse_relu=True)
+    m.train()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    z2 = z.detach().clone().requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(C)
+    bn.train()
+    with torch.no_grad():
+        bn.weight.copy_(m.weight)
+        bn.bias.copy_(m.bias)
+
+    y1 = m(x, z)
+    xc = x2.permute(0, 3, 1, 2)
+    y2 = torch.relu(bn(xc).permute(0, 2, 3, 1) + z2)
+    torch.testing.assert_close(y1, y2, rtol=1e-4, atol=1e-5)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(z.grad, z2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(m.weight.grad, bn.weight.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(m.bias.grad, bn.bias.grad, rtol=1e-4, atol=1e-5)

@@ -101,3 +101,54 @@ def test_syncbn_channel_last_gpu():
     y.backward(g)
     y_ref.backward(g)
     torch.testing.assert_close(x.grad, x2.grad.permute(0, 2, 3, 1), rtol=1e-4, atol=1e-4)
+
+
+def test_syncbn_fuse_relu_backward_gpu():
+    """fuse_relu backward must gate grads by the ReLU mask (kernel path)."""
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(7)
+    C = 16
+    x = torch.randn(4, C, 6, 6, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    sbn = SyncBatchNorm(C, fuse_relu=True).cuda()
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    sbn.train(); bn.train()
+    with torch.no_grad():
+        bn.weight.copy_(sbn.weight)
+        bn.bias.copy_(sbn.bias)
+    y1 = sbn(x)
+    y2 = torch.relu(bn(x2))
+    torch.testing.assert_close(y1, y2, rtol=1e-4, atol=1e-5)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, rtol=1e-4, atol=1e-4)
+
+
+def test_groupbn_add_relu_gpu():
+    """relu(bn(x) + z) NHWC path on the welford kernels."""
+    from apex_amd.contrib.groupbn import BatchNorm2d_NHWC
+
+    torch.manual_seed(8)
+    C = 16
+    x = torch.randn(3, 6, 6, C, device="cuda", requires_grad=True)
+    z = torch.randn(3, 6, 6, C, device="cuda", requires_grad=True)
+    m = BatchNorm2d_NHWC(C, fuse_relu=True).cuda()
+    m.train()
+    x2 = x.detach().clone().requires_grad_(True)
+    z2 = z.detach().clone().requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    bn.train()
+    with torch.no_grad():
+        bn.weight.copy_(m.weight)
+        bn.bias.copy_(m.bias)
+    y1 = m(x, z)
+    y2 = torch.relu(bn(x2.permute(0, 3, 1, 2)).permute(0, 2, 3, 1) + z2)
+    torch.testing.assert_close(y1, y2, rtol=1e-4, atol=1e-5)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(z.grad, z2.grad, rtol=1e-4, atol=1e-5)
