@@ -474,11 +474,14 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
     if delay_unscale:
         return
 
-    overflow = False
+    # bracket: all unscales below share one overflow flag; the dynamic scale
+    # ticks exactly once per iteration in finish_unscale (regardless of how
+    # many optimizers / grad sets were unscaled)
+    scaler.begin_unscale()
     for opt in opt_list:
         if hasattr(opt, "_amp_stash"):  # O2: unscale fp16 grads into masters
             model_grads, master_grads = _materialize_master_grads(opt, loss_scale)
-            ov = scaler.unscale_grads(model_grads, master_grads)
+            scaler.unscale_grads(model_grads, master_grads)
             # model grads are consumed into the masters here; clear them so
             # the next backward doesn't accumulate stale gradients (the
             # optimizer's zero_grad only sees the master params)
@@ -487,12 +490,12 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
             # fp32 params' grads unscaled in place
             stash = opt._amp_stash
             fp32_grads = [p.grad for p in stash.all_fp32_from_fp32_params if p.grad is not None]
-            ov2 = scaler.unscale_grads(fp32_grads, fp32_grads, scale_override=loss_scale) if fp32_grads else False
-            overflow = overflow or ov or ov2
+            if fp32_grads:
+                scaler.unscale_grads(fp32_grads, fp32_grads, scale_override=loss_scale)
         else:  # O0/O1: unscale in place
             grads = [p.grad for p in itertools.chain(*[g["params"] for g in opt.param_groups]) if p.grad is not None]
-            ov = scaler.unscale_grads(grads, grads)
-            overflow = overflow or ov
+            scaler.unscale_grads(grads, grads)
+    overflow = scaler.finish_unscale()
 
     if overflow:
         for opt in opt_list:

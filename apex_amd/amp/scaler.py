@@ -71,33 +71,73 @@ class LossScaler:
             self._hysteresis_t = torch.tensor([self._hysteresis], dtype=torch.int32, device=device)
             self._overflow_buf = torch.zeros(1, dtype=torch.int32, device=device)
 
+    # When several grad sets are unscaled in one optimizer iteration
+    # (multiple optimizers, O2 masters + fp32 group), the dynamic-scale state
+    # must tick exactly ONCE per iteration (round-1 advisor finding).
+    # scale_loss brackets the calls with begin_unscale()/finish_unscale():
+    # between them unscale_grads only accumulates into the shared overflow
+    # flag; finish_unscale runs the hysteresis update once and does the one
+    # device→host read. A bare unscale_grads call (no bracket) keeps the old
+    # self-contained behavior.
+    _in_iteration = False
+
+    def begin_unscale(self):
+        self._in_iteration = True
+        self._iter_overflow_cpu = False
+        if self._overflow_buf is not None:
+            self._overflow_buf.zero_()
+
+    def finish_unscale(self):
+        """Tick the dynamic scale once and return the iteration's overflow."""
+        self._in_iteration = False
+        if self._overflow_buf is not None and self._overflow_buf.is_cuda:
+            if self.dynamic:
+                amp_C = get_ext("amp_C")
+                amp_C.update_scale_hysteresis(
+                    self._scale_t, self._growth_tracker_t, self._hysteresis_t,
+                    self._overflow_buf, self._scale_factor,
+                    1.0 / self._scale_factor, self._scale_window, self._hysteresis,
+                )
+            self._has_overflow = bool(self._overflow_buf.item()) or self._iter_overflow_cpu
+            if self.dynamic:
+                self._loss_scale = float(self._scale_t.item())
+                self._loss_scale = min(self._max_loss_scale, max(self._min_loss_scale, self._loss_scale))
+                self._scale_t.fill_(self._loss_scale)
+        else:
+            self._has_overflow = self._iter_overflow_cpu
+            if self.dynamic:
+                self.update_scale()
+        return self._has_overflow
+
     def unscale_grads(self, grads_in, grads_out, scale_override=None):
         """out = in * (1/scale), with isfinite check setting the overflow flag.
 
-        Returns True if an overflow was detected (host-synchronizing on GPU —
-        one int read per iteration, matching the reference contract).
-
-        The dynamic-scale state ticks ONCE per optimizer step: callers that
-        unscale several grad sets in one step (amp O2 does masters then the
-        fp32 group) pass ``scale_override`` on the extra calls, which skips
-        the update.
+        Standalone call: returns True if an overflow was detected
+        (host-synchronizing on GPU) and ticks the dynamic scale iff
+        ``scale_override`` is None. Inside a begin_unscale()/finish_unscale()
+        bracket: only accumulates the overflow flag; no tick, no host sync.
         """
-        update_state = scale_override is None
+        bracketed = self._in_iteration
+        update_state = (scale_override is None) and not bracketed
         scale = self._loss_scale if scale_override is None else scale_override
         if len(grads_in) == 0:
-            self._has_overflow = False
+            if not bracketed:
+                self._has_overflow = False
             return False
         device = grads_in[0].device
         if device.type == "cuda":
             amp_C = get_ext("amp_C")
             self._ensure_device_state(device)
-            self._overflow_buf.zero_()
+            if not bracketed:
+                self._overflow_buf.zero_()
             multi_tensor_applier(
                 amp_C.multi_tensor_scale,
                 self._overflow_buf,
                 [grads_in, grads_out],
                 1.0 / scale,
             )
+            if bracketed:
+                return False  # decision deferred to finish_unscale
             if self.dynamic and update_state:
                 # on-device scale update; host reads only the skip decision
                 amp_C.update_scale_hysteresis(
@@ -123,6 +163,9 @@ class LossScaler:
                 if not torch.isfinite(gf).all():
                     overflow = True
                 go.copy_(gf.to(go.dtype))
+            if bracketed:
+                self._iter_overflow_cpu = self._iter_overflow_cpu or overflow
+                return False
             self._has_overflow = overflow
             if update_state:
                 self.update_scale()

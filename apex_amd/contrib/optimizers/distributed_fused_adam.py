@@ -21,6 +21,7 @@ differences from a line-by-line port, for MI355X:
 """
 
 from collections import defaultdict
+from contextlib import contextmanager
 
 import torch
 import torch.distributed as dist
@@ -86,6 +87,10 @@ class DistributedFusedAdam(torch.optim.Optimizer):
     step; exp_avg_sq (non-negative, huge dynamic range) is stored as
     sqrt(v) so fp16's ~2^-24..2^15 span covers v down to ~1e-14.
     """
+
+    # torch.amp.GradScaler: pass grad_scaler into step() instead of unscaling;
+    # step() checks found_inf itself and skips the update on overflow
+    _step_supports_amp_scaling = True
 
     def __init__(
         self,
@@ -282,16 +287,43 @@ class DistributedFusedAdam(torch.optim.Optimizer):
     def _make_hook(self, p):
         def hook(param):
             self._grad_copy(p)
-            if self.overlap_grad_sync:
+            if self.overlap_grad_sync and not self._in_no_sync:
                 b, _ = self.param_to_bucket[p]
                 if len(b.ready_params) == len(b.params) and not b.synced and b.sync_work is None:
                     self._start_bucket_grad_sync(b)
 
         return hook
 
+    _in_no_sync = False
+
+    @contextmanager
+    def no_sync(self):
+        """Gradient accumulation: inside this context backward passes only
+        accumulate into the flat grad buffers — no collectives are issued and
+        buckets are not marked synced, so a later (outside) backward + step
+        reduces the accumulated sum exactly once (reference:
+        distributed_fused_adam.py no_sync / greedy-grad-copy handling)."""
+        prev = self._in_no_sync
+        self._in_no_sync = True
+        try:
+            yield
+        finally:
+            self._in_no_sync = prev
+            if not prev:
+                # re-arm the bucket-full triggers for the next (syncing) backward
+                for b in self.buckets:
+                    b.ready_params.clear()
+
     def _grad_copy(self, p):
         b, offset = self.param_to_bucket[p]
         if p.grad is not None:
+            if b.synced or b.sync_work is not None:
+                raise RuntimeError(
+                    "DistributedFusedAdam: a gradient for a bucket arrived after "
+                    "its reduction was already issued — it would be silently "
+                    "dropped. Wrap accumulation backward passes in "
+                    "optimizer.no_sync(), or call step()/zero_grad() between "
+                    "backward passes.")
             n = p.numel()
             b.grad_data[offset:offset + n].add_(p.grad.detach().reshape(-1).to(b.grad_data.dtype))
             p.grad = None
@@ -435,12 +467,33 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             self._finish_param_sync_bucket(b)
 
         self.grad_sync()
-        self._step += 1
 
         if grad_scaler is not None:
             inv_scale = float(grad_scaler._get_scale_async().double().reciprocal())
+            found = torch.zeros(1, dtype=torch.float32, device=self.device)
             for b in self.buckets:
                 b.grad_shard.mul_(inv_scale)
+                found += (~torch.isfinite(b.grad_shard)).sum()
+            if self.world_size > 1:
+                dist.all_reduce(found, group=self.process_group)
+            if self.redundant_process_group is not None:
+                dist.all_reduce(found, group=self.redundant_process_group)
+            found = (found > 0).to(torch.float32)
+            # report into the scaler's per-optimizer state so update() backs
+            # the scale off (torch GradScaler contract for optimizers with
+            # _step_supports_amp_scaling)
+            try:
+                state = grad_scaler._per_optimizer_states[id(self)]
+                state["found_inf_per_device"] = {self.device: found}
+            except (AttributeError, KeyError):
+                pass
+            if bool(found):
+                # overflow: skip the update entirely — optimizer state and
+                # params must stay untouched (round-1 advisor finding)
+                self._reset_buckets_after_step()
+                return loss
+
+        self._step += 1
 
         for b in self.buckets:
             group = b.group
@@ -496,13 +549,15 @@ class DistributedFusedAdam(torch.optim.Optimizer):
                     self._finish_param_sync_bucket(b)
                 if use_comm_stream:
                     torch.cuda.current_stream().wait_stream(self._comm_stream)
+        self._reset_buckets_after_step()
+        return loss
+
+    def _reset_buckets_after_step(self):
         for b in self.buckets:
-            # reset for the next iteration
             b.grad_data.zero_()
             b.ready_params.clear()
             b.synced = False
             b.sync_work = None
-        return loss
 
     def _adam_ref(self, group, bias_correction, beta1, beta2, b, master,
                   exp_avg, exp_avg_sq):

@@ -418,3 +418,87 @@ def test_param_remainder_bit_roundtrip_all_patterns():
     opt._set_master(b, master.clone())
     back = opt._get_master(b)
     assert torch.equal(master.view(torch.int32), back.view(torch.int32))
+
+
+def _no_sync_worker(rank, world_size):
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    ref_model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-2, weight_decay=0.0,
+                               bucket_cap_mb=1, overlap_grad_sync=True)
+    ref_opt = torch.optim.AdamW(ref_model.parameters(), lr=1e-2, weight_decay=0.0)
+    accum = 3
+    for it in range(3):
+        xs_all = []
+        for micro in range(accum):
+            torch.manual_seed(100 + rank + it * 17 + micro * 31)
+            x = torch.randn(4, 32)
+            gathered = [torch.empty_like(x) for _ in range(world_size)]
+            dist.all_gather(gathered, x)
+            xs_all.extend(gathered)
+            if micro < accum - 1:
+                with opt.no_sync():
+                    model(x).pow(2).mean().backward()
+            else:
+                model(x).pow(2).mean().backward()
+        opt.step()
+        ref_opt.zero_grad()
+        # reference: SUM of micro-batch losses averaged over ranks (DFA
+        # accumulates micro-grads by summation, averages across ranks)
+        loss = sum(ref_model(xi).pow(2).mean() for xi in xs_all) / world_size
+        loss.backward()
+        ref_opt.step()
+        for p, rp in zip(model.parameters(), ref_model.parameters()):
+            torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_dist_adam_no_sync_grad_accumulation():
+    run_distributed(_no_sync_worker, world_size=2)
+
+
+def test_dist_adam_late_grad_raises():
+    """A gradient arriving after the bucket's reduction was issued must error
+    loudly (not be silently dropped) — world_size 1, no dist init needed."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-2, bucket_cap_mb=1,
+                               overlap_grad_sync=True)
+    model(torch.randn(4, 32)).pow(2).mean().backward()
+    with pytest.raises(RuntimeError, match="no_sync"):
+        model(torch.randn(4, 32)).pow(2).mean().backward()
+
+
+def test_dist_adam_found_inf_skips_step():
+    """Overflowed grads under a GradScaler must not touch params or moments."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-2, bucket_cap_mb=1,
+                               overlap_grad_sync=False)
+
+    class FakeScaler:
+        _per_optimizer_states = {}
+
+        def _get_scale_async(self):
+            return torch.tensor(2.0)
+
+    before = [p.detach().clone() for p in model.parameters()]
+    for i, p in enumerate(model.parameters()):
+        p.grad = torch.full_like(p, float("inf") if i == 0 else 1.0)
+        opt._grad_copy(p)
+    opt.step(grad_scaler=FakeScaler())
+    for p, b in zip(model.parameters(), before):
+        torch.testing.assert_close(p.detach(), b)
+    assert opt._step == 0
+    for b in opt.buckets:
+        assert float(b.exp_avg.abs().sum()) == 0.0
+    # next clean step must proceed
+    for p in model.parameters():
+        p.grad = torch.ones_like(p)
+        opt._grad_copy(p)
+    opt.step(grad_scaler=FakeScaler())
+    assert opt._step == 1
+    changed = any(not torch.equal(p.detach(), b) for p, b in zip(model.parameters(), before))
+    assert changed
