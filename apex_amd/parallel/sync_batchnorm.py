@@ -79,9 +79,23 @@ class SyncBatchnormFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, input, weight, bias, z, running_mean, running_var, eps, track_running_stats,
                 momentum, process_group, channel_last, fuse_relu):
-        input = input.contiguous(
-            memory_format=torch.contiguous_format
+        # torch channels_last memory format ([N,C,H,W] shape with NHWC
+        # strides): permute to an [N,H,W,C]-shaped contiguous VIEW (zero
+        # copy) and run the NHWC kernels — the round-1 path re-packed to
+        # NCHW here, forfeiting the layout MIOpen bf16 convs prefer.
+        nhwc_mem = (
+            not channel_last and input.dim() == 4
+            and input.is_contiguous(memory_format=torch.channels_last)
+            and not input.is_contiguous()
         )
+        if nhwc_mem:
+            input = input.permute(0, 2, 3, 1)
+            if z is not None:
+                z = z.permute(0, 2, 3, 1)
+            channel_last = True
+        else:
+            input = input.contiguous(memory_format=torch.contiguous_format)
+        ctx.nhwc_mem = nhwc_mem
         world_size = dist.get_world_size(process_group) if (dist.is_available() and dist.is_initialized()) else 1
 
         use_kernels = input.is_cuda
@@ -172,10 +186,18 @@ class SyncBatchnormFunction(torch.autograd.Function):
             ctx.save_for_backward(input, weight, mean, inv_std, out)
         else:
             ctx.save_for_backward(input, weight, mean, inv_std)
+        if nhwc_mem:
+            # back to the [N,C,H,W] shape the caller sees; the data stays in
+            # NHWC order, so this returns a channels_last-contiguous tensor
+            out = out.permute(0, 3, 1, 2)
         return out
 
     @staticmethod
     def backward(ctx, grad_output):
+        if ctx.nhwc_mem:
+            # saved tensors are [N,H,W,C] views; bring the upstream grad to
+            # the same layout (zero-copy when it is channels_last already)
+            grad_output = grad_output.permute(0, 2, 3, 1)
         if ctx.fuse_relu:
             input, weight, mean, inv_std, out = ctx.saved_tensors
             # gate by the ReLU: units clipped to 0 in forward get zero grad
@@ -241,6 +263,10 @@ class SyncBatchnormFunction(torch.autograd.Function):
             grad_weight = grad_weight.to(weight.dtype)
             grad_bias = grad_bias.to(weight.dtype)
 
+        if ctx.nhwc_mem:
+            grad_input = grad_input.permute(0, 3, 1, 2)
+            if grad_z is not None:
+                grad_z = grad_z.permute(0, 3, 1, 2)
         return (grad_input, grad_weight, grad_bias, grad_z,
                 None, None, None, None, None, None, None, None)
 

@@ -113,10 +113,16 @@ def main():
         if use_syncbn:
             model = convert_syncbn_model(model)
         model = model.to(device)
+        if not use_cpu:
+            # NHWC end to end: MIOpen bf16 convs and the welford_*_c_last
+            # SyncBN kernels both prefer channels_last
+            model = model.to(memory_format=torch.channels_last)
         opt = FusedSGD(model.parameters(), lr=0.1, momentum=0.9, weight_decay=1e-4)
         model, opt = amp.initialize(model, opt, opt_level="O1",
                                     cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
         x = torch.randn(args.batch, 3, args.image_size, args.image_size, device=device)
+        if not use_cpu:
+            x = x.contiguous(memory_format=torch.channels_last)
         y = torch.randint(0, 1000, (args.batch,), device=device)
         criterion = torch.nn.CrossEntropyLoss()
 

@@ -166,3 +166,33 @@ def test_groupbn_bn_add_relu_semantics():
     torch.testing.assert_close(z.grad, z2.grad, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(m.weight.grad, bn.weight.grad, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(m.bias.grad, bn.bias.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_syncbn_channels_last_memory_format():
+    """torch channels_last tensors ([N,C,H,W] shape, NHWC strides) must run
+    the NHWC path zero-copy and return channels_last output with matching
+    numerics + grads."""
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(5)
+    C = 8
+    base = torch.randn(4, C, 5, 5)
+    x = base.clone().to(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = base.clone().requires_grad_(True)
+    sbn = SyncBatchNorm(C)
+    bn = torch.nn.BatchNorm2d(C)
+    sbn.train(); bn.train()
+    with torch.no_grad():
+        bn.weight.copy_(sbn.weight)
+        bn.bias.copy_(sbn.bias)
+    y1 = sbn(x)
+    assert y1.is_contiguous(memory_format=torch.channels_last)
+    y2 = bn(x2)
+    torch.testing.assert_close(y1, y2, rtol=1e-5, atol=1e-6)
+    g = torch.randn_like(y2)
+    y1.backward(g.to(memory_format=torch.channels_last))
+    y2.backward(g)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.running_mean, bn.running_mean, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(sbn.running_var, bn.running_var, rtol=1e-5, atol=1e-6)
