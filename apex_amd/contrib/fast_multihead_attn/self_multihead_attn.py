@@ -56,6 +56,22 @@ class SelfMultiheadAttn(torch.nn.Module):
         k = qkv[:, :, 1].permute(1, 2, 0, 3)
         v = qkv[:, :, 2].permute(1, 2, 0, 3)
 
+        from ...transformer import flash_attention, flash_attention_supported
+
+        dropout_active = self.dropout if (is_training and self.training) else 0.0
+        if (not need_weights and key_padding_mask is None
+                and (attn_mask is None or attn_mask == "causal")
+                and flash_attention_supported(q, dropout=dropout_active)):
+            # MFMA flash path: hardware-validated round 2, no S x S matrix
+            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  causal=(attn_mask == "causal"), scale=self.scaling)
+            ctx = ctx.permute(2, 0, 1, 3).reshape(s * b, h)
+            out = fused_dense_function(ctx, self.out_proj_weight, self.out_proj_bias)
+            out = out.reshape(s, b, h)
+            if self.include_norm_add:
+                out = out + residual
+            return out, None
+
         scores = torch.matmul(q, k.transpose(-2, -1))
         if attn_mask == "causal":
             probs = scaled_upper_triang_masked_softmax(

@@ -20,6 +20,7 @@ from ..fused_dense import fused_dense_function, fused_dense_gelu_dense_function
 from ..transformer import (
     fused_apply_rotary_pos_emb,
     scaled_masked_softmax,
+    scaled_softmax,
     scaled_upper_triang_masked_softmax,
 )
 
@@ -65,19 +66,28 @@ class FusedSelfAttention(nn.Module):
         nn.init.normal_(self.proj_w, std=0.02)
 
     def forward(self, x, mask=None):
+        from ..transformer import flash_attention, flash_attention_supported
+
         # x: [b, s, h]
         b, s, h = x.shape
         qkv = fused_dense_function(x, self.qkv_w, self.qkv_b)  # [b, s, 3h]
         qkv = qkv.view(b, s, 3, self.nh, self.hd).permute(2, 0, 3, 1, 4)  # [3, b, nh, s, hd]
         q, k, v = qkv[0], qkv[1], qkv[2]
-        scores = torch.matmul(q, k.transpose(-2, -1))  # [b, nh, s, s]
         scale = 1.0 / math.sqrt(self.hd)
-        if self.causal:
-            probs = scaled_upper_triang_masked_softmax(scores.view(b * self.nh, s, s), scale)
-            probs = probs.view(b, self.nh, s, s)
+        if mask is None and flash_attention_supported(q):
+            # MFMA flash kernel: no S x S matrix, no separate softmax pass
+            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  causal=self.causal, scale=scale)
         else:
-            probs = scaled_masked_softmax(scores, mask, scale)
-        ctx = torch.matmul(probs, v)  # [b, nh, s, hd]
+            scores = torch.matmul(q, k.transpose(-2, -1))  # [b, nh, s, s]
+            if self.causal:
+                probs = scaled_upper_triang_masked_softmax(scores.view(b * self.nh, s, s), scale)
+                probs = probs.view(b, self.nh, s, s)
+            elif mask is not None:
+                probs = scaled_masked_softmax(scores, mask, scale)
+            else:
+                probs = scaled_softmax(scores.contiguous(), scale)
+            ctx = torch.matmul(probs, v)  # [b, nh, s, hd]
         ctx = ctx.transpose(1, 2).reshape(b, s, h)
         return fused_dense_function(ctx, self.proj_w, self.proj_b)
 
@@ -215,12 +225,12 @@ def llama_small_config(seq_len=512):
 class LlamaAttention(nn.Module):
     """Causal self-attention with fused RoPE on q/k (no biases).
 
-    ``use_flash=True`` routes the attention core through
-    ``transformer.flash_attention`` (no S x S matrix) instead of
-    bmm + causal wave64 softmax — flip once the experimental fmha kernels
-    are validated on hardware (ROADMAP round 2)."""
+    ``use_flash=True`` (default; hardware-validated round 2) routes the
+    attention core through ``transformer.flash_attention`` (no S x S matrix)
+    instead of bmm + causal wave64 softmax whenever the kernel supports the
+    shape/dtype; unsupported cases fall back automatically."""
 
-    use_flash = False
+    use_flash = True
 
     def __init__(self, cfg: TransformerLMConfig):
         super().__init__()
@@ -241,9 +251,9 @@ class LlamaAttention(nn.Module):
         q = q.permute(1, 2, 0, 3)                       # [b, nh, s, hd]
         k = k.permute(1, 2, 0, 3)
         v = qkv[:, :, 2].permute(0, 2, 1, 3)            # [b, nh, s, hd]
-        if LlamaAttention.use_flash:
-            from ..transformer import flash_attention
+        from ..transformer import flash_attention, flash_attention_supported
 
+        if LlamaAttention.use_flash and flash_attention_supported(q):
             ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
                                   causal=True, scale=1.0 / math.sqrt(self.hd))
         else:

@@ -15,7 +15,7 @@ from .rope import (
     fused_apply_rotary_pos_emb_2d,
 )
 from .wgrad import wgrad_gemm_accum_fp32, wgrad_gemm_accum_fp16
-from .fmha import flash_attention, flash_attention_forward
+from .fmha import flash_attention, flash_attention_forward, flash_attention_supported
 
 __all__ = [
     "ScaledMaskedSoftmax",
@@ -34,4 +34,5 @@ __all__ = [
     "wgrad_gemm_accum_fp16",
     "flash_attention_forward",
     "flash_attention",
+    "flash_attention_supported",
 ]

@@ -1,14 +1,15 @@
-"""Flash-attention forward (EXPERIMENTAL — round-2 perf item).
+"""Flash attention on hand-written MFMA kernels (csrc/fmha.hip).
 
-``flash_attention_forward(q, k, v)`` runs the hand-written MFMA kernel
-(csrc/fmha.hip) on GPU: Q K^T, online softmax, and P V in one pass — no
-S x S attention matrix is ever materialized. Returns ``(out, lse)`` with
-``lse = logsumexp`` rows for a later backward. bf16, head_dim 64/128,
-seq_len % 32 == 0, layout [B, H, S, D].
+``flash_attention_forward(q, k, v)`` runs Q K^T, online softmax, and P V in
+one pass — no S x S attention matrix is ever materialized. Returns
+``(out, lse)`` with ``lse = logsumexp`` rows for the backward. bf16,
+head_dim 64/128, seq_len % 32 == 0, layout [B, H, S, D].
 
-Status: compile-checked and unit-testable against the eager composition
-(tests/test_fmha_gpu.py — currently skipped pending on-hardware validation);
-not wired into the bundled models yet.
+Hardware-validated round 2 (profiles/validate_fmha_r2.log): forward and the
+fused MFMA backward match the fp32 eager composition across the shape sweep;
+forward is 2.1-2.6x faster than bmm + wave64-softmax at D64. Wired into the
+bundled transformer models and contrib.fast_multihead_attn via
+``flash_attention_supported``.
 """
 
 import math
@@ -16,6 +17,18 @@ import math
 import torch
 
 from .._ext import get_ext
+
+
+def flash_attention_supported(q, dropout=0.0):
+    """True when the MFMA flash kernel can take this tensor: CUDA bf16,
+    head_dim 64/128, seq_len % 32 == 0, no attention dropout."""
+    return (
+        q.is_cuda
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[-2] % 32 == 0
+        and dropout == 0.0
+    )
 
 
 def eager_attention_reference(q, k, v, causal=False, scale=None):
@@ -55,7 +68,10 @@ class FlashAttentionFunction(torch.autograd.Function):
 
     CHUNK = 256  # q rows recomputed per tile
 
-    use_fused_backward = False  # flip after round-2 on-hardware validation
+    # fused MFMA backward validated on hardware round 2
+    # (profiles/validate_fmha_r2.log); the chunked-recompute composition
+    # below remains the CPU / debug fallback
+    use_fused_backward = True
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):

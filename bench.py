@@ -168,10 +168,11 @@ def main():
                                     cast_model_type=torch.bfloat16, loss_scale=1.0,
                                     keep_batchnorm_fp32=False, verbosity=0)
         tokens = torch.randint(0, cfg.vocab_size, (args.batch, cfg.seq_len), device=device)
+        # BERT trains unpadded/full attention here (synthetic fixed-length
+        # batches): mask=None routes through the MFMA flash kernel; padded
+        # batches would pass a bool mask and take the scaled_masked_softmax
+        # path (both covered by tests)
         mask = None
-        if not cfg.causal and args.model == "bert":
-            mask = torch.zeros(args.batch, 1, cfg.seq_len, cfg.seq_len, dtype=torch.bool,
-                               device=device)
 
         def fwd_loss():
             logits = model(tokens, mask) if mask is not None else model(tokens)

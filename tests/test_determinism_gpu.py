@@ -8,21 +8,17 @@ pytestmark = pytest.mark.gpu
 from test_determinism_cpu import run_harness  # noqa: E402
 
 # apex_amd's own kernels are deterministic by construction (fixed-order
-# two-stage reductions, no fp32 atomics); the remaining variable is MIOpen's
-# deterministic-conv selection, which this pool hasn't confirmed yet —
-# strict=False keeps the check visible without gating the suite on it.
-_soft = pytest.mark.xfail(strict=False,
-                          reason="MIOpen conv determinism unverified on this pool")
+# two-stage reductions, no fp32 atomics). MIOpen conv determinism was
+# confirmed on this pool in round 2 (both runs xpassed on hardware), so the
+# former soft-xfail markers are gone — these are hard gates now.
 
 
-@_soft
 def test_harness_bitwise_deterministic_gpu_o1():
     a = run_harness("O1", iters=6, batch=8, image=64)
     b = run_harness("O1", iters=6, batch=8, image=64)
     assert a == b
 
 
-@_soft
 def test_harness_bitwise_deterministic_gpu_o2():
     a = run_harness("O2", iters=6, batch=8, image=64)
     b = run_harness("O2", iters=6, batch=8, image=64)

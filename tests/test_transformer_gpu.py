@@ -303,3 +303,68 @@ def test_llama_model_gpu_step():
         losses.append(float(loss))
     assert losses[-1] < losses[0]
     assert all(l == l for l in losses)  # no NaNs
+
+
+def test_model_attention_flash_route_matches_composed():
+    """FusedSelfAttention's flash route (mask=None, bf16, D64) must match
+    the composed bmm+softmax route, forward and backward."""
+    from apex_amd.models.transformer import FusedSelfAttention, TransformerLMConfig
+    from apex_amd.transformer import fmha
+
+    cfg = TransformerLMConfig(vocab_size=128, hidden=256, layers=1, heads=4,
+                              seq_len=64, causal=False)
+    torch.manual_seed(0)
+    attn = FusedSelfAttention(cfg).cuda().bfloat16()
+    x = torch.randn(2, 64, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+
+    y_flash = attn(x)  # mask=None → flash (supported shape)
+    g = torch.randn_like(y_flash)
+    y_flash.backward(g)
+    gx_flash = x.grad.clone()
+    gw_flash = attn.qkv_w.grad.clone()
+
+    x.grad = None
+    attn.zero_grad()
+    import apex_amd.transformer as tr
+    orig = fmha.flash_attention_supported
+    try:
+        fmha.flash_attention_supported = lambda q, dropout=0.0: False
+        tr.flash_attention_supported = fmha.flash_attention_supported
+        y_comp = attn(x)
+        y_comp.backward(g)
+    finally:
+        fmha.flash_attention_supported = orig
+        tr.flash_attention_supported = orig
+    torch.testing.assert_close(y_flash.float(), y_comp.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gx_flash.float(), x.grad.float(), rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(gw_flash.float(), attn.qkv_w.grad.float(),
+                               rtol=5e-2, atol=5e-2)
+
+
+def test_self_mha_flash_route_matches_composed_gpu():
+    """contrib SelfMultiheadAttn causal flash route vs composed softmax."""
+    from apex_amd.contrib.fast_multihead_attn import SelfMultiheadAttn
+    from apex_amd.transformer import fmha
+    import apex_amd.contrib.fast_multihead_attn.self_multihead_attn  # noqa
+
+    torch.manual_seed(1)
+    mha = SelfMultiheadAttn(256, 4, dropout=0.0).cuda().bfloat16()
+    x = torch.randn(64, 2, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y1, _ = mha(x, attn_mask="causal")
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    gx1 = x.grad.clone()
+    x.grad = None
+    mha.zero_grad()
+    import apex_amd.transformer as tr
+    orig = fmha.flash_attention_supported
+    try:
+        tr.flash_attention_supported = lambda q, dropout=0.0: False
+        y2, _ = mha(x, attn_mask="causal")
+        y2.backward(g)
+    finally:
+        tr.flash_attention_supported = orig
+    torch.testing.assert_close(y1.float(), y2.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gx1.float(), x.grad.float(), rtol=5e-2, atol=5e-2)
