@@ -66,6 +66,16 @@ def convert_syncbn_model(module, process_group=None, channel_last=False):
     return mod
 
 
+def _nhwc_route(input, channel_last):
+    """True for torch channels_last tensors ([N,C,H,W] shape, NHWC strides)
+    that should run the NHWC kernels via a zero-copy permute."""
+    return (
+        not channel_last and input.dim() == 4
+        and input.is_contiguous(memory_format=torch.channels_last)
+        and not input.is_contiguous()
+    )
+
+
 def _to_channels_second(x, channel_last):
     """View input as [N*, C, M] reduction layout helpers for the torch path."""
     if channel_last:
@@ -83,11 +93,11 @@ class SyncBatchnormFunction(torch.autograd.Function):
         # strides): permute to an [N,H,W,C]-shaped contiguous VIEW (zero
         # copy) and run the NHWC kernels — the round-1 path re-packed to
         # NCHW here, forfeiting the layout MIOpen bf16 convs prefer.
-        nhwc_mem = (
-            not channel_last and input.dim() == 4
-            and input.is_contiguous(memory_format=torch.channels_last)
-            and not input.is_contiguous()
-        )
+        # The OUTPUT stays [N,H,W,C]: the module wrapper permutes it back
+        # outside this Function (a custom Function must not return a view of
+        # its own output, or downstream inplace ops like ReLU(inplace=True)
+        # are rejected by autograd).
+        nhwc_mem = _nhwc_route(input, channel_last)
         if nhwc_mem:
             input = input.permute(0, 2, 3, 1)
             if z is not None:
@@ -186,18 +196,12 @@ class SyncBatchnormFunction(torch.autograd.Function):
             ctx.save_for_backward(input, weight, mean, inv_std, out)
         else:
             ctx.save_for_backward(input, weight, mean, inv_std)
-        if nhwc_mem:
-            # back to the [N,C,H,W] shape the caller sees; the data stays in
-            # NHWC order, so this returns a channels_last-contiguous tensor
-            out = out.permute(0, 3, 1, 2)
         return out
 
     @staticmethod
     def backward(ctx, grad_output):
-        if ctx.nhwc_mem:
-            # saved tensors are [N,H,W,C] views; bring the upstream grad to
-            # the same layout (zero-copy when it is channels_last already)
-            grad_output = grad_output.permute(0, 2, 3, 1)
+        # nhwc_mem: forward returned [N,H,W,C], so grad_output arrives in
+        # that same shape (the module-level permute is autograd-tracked)
         if ctx.fuse_relu:
             input, weight, mean, inv_std, out = ctx.saved_tensors
             # gate by the ReLU: units clipped to 0 in forward get zero grad
@@ -322,8 +326,14 @@ class SyncBatchNorm(_BatchNorm):
                 out = out.relu()
             return out
 
-        return SyncBatchnormFunction.apply(
+        out = SyncBatchnormFunction.apply(
             input, self.weight, self.bias, z, self.running_mean, self.running_var,
             self.eps, self.track_running_stats, exponential_average_factor,
             self.process_group, channel_last, self.fuse_relu,
         )
+        if _nhwc_route(input, channel_last):
+            # function returned [N,H,W,C]; present the [N,C,H,W]-shaped
+            # channels_last view the caller expects (plain autograd view op,
+            # safe for downstream inplace use)
+            out = out.permute(0, 3, 1, 2)
+        return out

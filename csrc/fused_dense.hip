@@ -110,6 +110,25 @@ at::Tensor linear_bias_forward(at::Tensor input, at::Tensor weight, at::Tensor b
   return out.reshape(sizes);
 }
 
+// out = relu(X @ W^T + b) in ONE hipBLASLt launch (RELU_BIAS epilogue) —
+// the fused 1x1-conv/bias/ReLU building block (reference op surface:
+// apex/contrib/csrc/conv_bias_relu/conv_bias_relu.cpp — cuDNN runtime
+// fusion there; the CDNA4-native answer is a GEMM epilogue since a 1x1
+// conv over NHWC IS a GEMM).
+at::Tensor linear_bias_relu_forward(at::Tensor input, at::Tensor weight, at::Tensor bias) {
+  auto x = flat2d(input);
+  auto w = weight.contiguous();
+  auto b = bias.contiguous();
+  auto out = at::empty({x.size(0), w.size(0)}, x.options());
+  if (!lt_linear(x, w, out, &b, HIPBLASLT_EPILOGUE_RELU_BIAS, nullptr, /*allow_fail=*/true)) {
+    lt_linear(x, w, out, &b, HIPBLASLT_EPILOGUE_BIAS, nullptr);
+    out.relu_();
+  }
+  auto sizes = input.sizes().vec();
+  sizes.back() = w.size(0);
+  return out.reshape(sizes);
+}
+
 at::Tensor linear_forward(at::Tensor input, at::Tensor weight) {
   auto x = flat2d(input);
   auto w = weight.contiguous();
@@ -232,6 +251,7 @@ void wgrad_gemm_accum_fp16(at::Tensor input, at::Tensor grad_output, at::Tensor 
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("linear_bias_relu_forward", &linear_bias_relu_forward);
   m.def("linear_bias_forward", &linear_bias_forward);
   m.def("linear_forward", &linear_forward);
   m.def("linear_bias_backward", &linear_bias_backward);
