@@ -114,9 +114,14 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         store_param_remainders=False,
         with_scaled_states=False,
         nccl_ub=False,
+        capturable=False,
     ):
         if amsgrad:
             raise RuntimeError("DistributedFusedAdam does not support AMSGrad")
+        if capturable and (store_param_remainders or with_scaled_states):
+            raise RuntimeError(
+                "capturable is incompatible with store_param_remainders / "
+                "with_scaled_states (their re-quantization is host-driven)")
         defaults = dict(lr=lr, bias_correction=bias_correction, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
@@ -151,6 +156,18 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         self._use_stream = device.type == "cuda"
         self._comm_stream = torch.cuda.Stream() if self._use_stream else None
         self._noop = torch.zeros(1, dtype=torch.int32, device=device)
+
+        # capturable: every step-time decision lives on device so the whole
+        # step (unscale check + Adam + param copy/gather) records into a
+        # hipGraph and replays with updated lr/step/scale (reference:
+        # distributed_fused_adam.py:2368-2586 CUDA-graph step)
+        self.capturable = capturable
+        if capturable:
+            for group in self.param_groups:
+                group["lr_t"] = torch.full((1,), float(group["lr"]),
+                                           dtype=torch.float32, device=device)
+            self._step_t = torch.zeros(1, dtype=torch.int32, device=device)
+            self._inv_scale_t = torch.ones(1, dtype=torch.float32, device=device)
 
         # nccl_ub: place the flat buckets in an ncclMemAlloc-backed pool so
         # RCCL registers them as user buffers (zero-copy collectives over
@@ -468,6 +485,10 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
         self.grad_sync()
 
+        if self.capturable and self.device.type == "cuda":
+            self._step_capturable(grad_scaler)
+            return loss
+
         if grad_scaler is not None:
             inv_scale = float(grad_scaler._get_scale_async().double().reciprocal())
             found = torch.zeros(1, dtype=torch.float32, device=self.device)
@@ -495,6 +516,11 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
         self._step += 1
 
+        # pipelined per-bucket update: bucket i's shard-copy + all-gather is
+        # issued on the comm stream right after ITS Adam launch, so the
+        # gather of bucket i overlaps the Adam math of bucket i+1 (the
+        # reference's pipeline_size=2 bucket pipeline,
+        # distributed_fused_adam.py:2505-2586)
         for b in self.buckets:
             group = b.group
             beta1, beta2 = group["betas"]
@@ -514,43 +540,92 @@ class DistributedFusedAdam(torch.optim.Optimizer):
                                exp_avg, exp_avg_sq)
             self._set_moments(b, exp_avg, exp_avg_sq)
             self._set_master(b, master)
+            self._issue_bucket_param_sync(b)
 
-        # param sync: shard -> param bucket, then all_gather. On GPU the
-        # gathers run async on the comm stream with a stream-order dependency
-        # back to the main stream — the next forward's kernels queue behind
-        # them while the host races ahead (overlap_param_sync semantics
-        # without model hooks; per-bucket lazy waits land in a later round).
-        use_comm_stream = self._use_stream and self.world_size > 1
-        if use_comm_stream:
-            self._comm_stream.wait_stream(torch.cuda.current_stream())
-        for b in self.buckets:
-            if b.master_shard is None:
-                continue  # remainder mode: _set_master already wrote the shard
-            lo = self.rank * b.shard_size
-            shard = b.param_data[lo:lo + b.shard_size]
-            shard.copy_(b.master_shard.to(b.param_data.dtype))
-        if self.world_size > 1:
-            ctx = torch.cuda.stream(self._comm_stream) if use_comm_stream else _null()
-            with ctx:
-                for b in self.buckets:
-                    lo = self.rank * b.shard_size
-                    shard = b.param_data[lo:lo + b.shard_size]
-                    if _backend_supports_rs(self.process_group):
-                        b.param_sync_work = dist.all_gather_into_tensor(
-                            b.param_data, shard, group=self.process_group, async_op=True)
-                    else:
-                        chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
-                        b.param_sync_work = dist.all_gather(
-                            chunks, shard.contiguous(), group=self.process_group,
-                            async_op=True)
-                        b._param_chunks = chunks
-            if not self.overlap_param_sync:
-                for b in self.buckets:
-                    self._finish_param_sync_bucket(b)
-                if use_comm_stream:
-                    torch.cuda.current_stream().wait_stream(self._comm_stream)
+        if self.world_size > 1 and not self.overlap_param_sync:
+            for b in self.buckets:
+                self._finish_param_sync_bucket(b)
+            if self._use_stream:
+                torch.cuda.current_stream().wait_stream(self._comm_stream)
         self._reset_buckets_after_step()
         return loss
+
+    def _issue_bucket_param_sync(self, b):
+        """Copy this bucket's updated master into its param shard and launch
+        the all-gather on the comm stream (ordered after the main-stream work
+        issued so far, i.e. this bucket's Adam kernel)."""
+        lo = self.rank * b.shard_size
+        shard = b.param_data[lo:lo + b.shard_size]
+        if b.master_shard is not None:
+            shard.copy_(b.master_shard.to(b.param_data.dtype))
+        # else: remainder mode — _set_master already wrote the shard bits
+        if self.world_size == 1:
+            return
+        use_comm_stream = self._use_stream
+        if use_comm_stream:
+            self._comm_stream.wait_stream(torch.cuda.current_stream())
+        ctx = torch.cuda.stream(self._comm_stream) if use_comm_stream else _null()
+        with ctx:
+            if _backend_supports_rs(self.process_group):
+                b.param_sync_work = dist.all_gather_into_tensor(
+                    b.param_data, shard, group=self.process_group, async_op=True)
+            else:
+                chunks = [torch.empty_like(shard) for _ in range(self.world_size)]
+                b.param_sync_work = dist.all_gather(
+                    chunks, shard.contiguous(), group=self.process_group,
+                    async_op=True)
+                b._param_chunks = chunks
+
+    def _step_capturable(self, grad_scaler):
+        """hipGraph-capturable step: no host reads, every decision is a
+        device op. The found-inf gate is the kernel-side noop early-exit
+        (AdamCapturableFunctor, csrc/multi_tensor_adam.hip:81-122); lr/step/
+        inv_scale are read from device pointers at kernel time."""
+        amp_C = get_ext("amp_C")
+        if grad_scaler is not None:
+            scale = grad_scaler._get_scale_async()
+            self._inv_scale_t.copy_(scale.double().reciprocal().float())
+            found = torch.zeros(1, dtype=torch.float32, device=self.device)
+            for b in self.buckets:
+                found += (~torch.isfinite(b.grad_shard)).sum()
+            if self.world_size > 1:
+                dist.all_reduce(found, group=self.process_group)
+            if self.redundant_process_group is not None:
+                dist.all_reduce(found, group=self.redundant_process_group)
+            found = (found > 0).to(torch.float32)
+            self._noop.copy_(found.to(torch.int32))
+            try:
+                state = grad_scaler._per_optimizer_states[id(self)]
+                state["found_inf_per_device"] = {self.device: found}
+            except (AttributeError, KeyError):
+                pass
+        else:
+            self._noop.zero_()
+            self._inv_scale_t.fill_(1.0)
+        # step advances only on non-overflow iterations (device-side)
+        self._step_t.add_(1 - self._noop)
+
+        for b in self.buckets:
+            group = b.group
+            beta1, beta2 = group["betas"]
+            bias_correction = 1 if group["bias_correction"] else 0
+            multi_tensor_applier(
+                amp_C.multi_tensor_adam_capturable, self._noop,
+                [[b.grad_shard], [b.master_shard], [b.exp_avg], [b.exp_avg_sq]],
+                group["lr_t"], beta1, beta2, group["eps"], self._step_t,
+                self.adam_w_mode, bias_correction, group["weight_decay"],
+                self._inv_scale_t,
+            )
+            # on overflow the kernel early-exited, so this copy/gather
+            # redistributes unchanged values — safe, and keeps the captured
+            # op sequence identical every iteration
+            self._issue_bucket_param_sync(b)
+        if self.world_size > 1 and not self.overlap_param_sync:
+            for b in self.buckets:
+                self._finish_param_sync_bucket(b)
+            if self._use_stream:
+                torch.cuda.current_stream().wait_stream(self._comm_stream)
+        self._reset_buckets_after_step()
 
     def _reset_buckets_after_step(self):
         for b in self.buckets:
@@ -583,8 +658,13 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         ``load_state_dict`` can reshard onto any world size with the same
         param list and bucket_cap."""
         groups_sd = [
-            {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
+            {k: v for k, v in g.items() if k not in ("params", "lr_t")}
+            for g in self.param_groups
         ]
+        if self.capturable:
+            # the live counter is the device tensor (host _step is not
+            # advanced by graph replays)
+            self._step = int(self._step_t.item())
         if gather_on_root:
             buckets_sd = []
             for b in self.buckets:
@@ -619,8 +699,12 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         if "buckets" not in sd:
             raise ValueError("expected a DistributedFusedAdam state dict")
         self._step = sd["step"]
+        if self.capturable:
+            self._step_t.fill_(int(sd["step"]))
         for g, gsd in zip(self.param_groups, sd["param_groups"]):
             g.update(gsd)
+            if self.capturable:
+                g["lr_t"].fill_(float(g["lr"]))
         if sd.get("gathered"):
             # reshard a gathered (full) checkpoint onto this world size
             for b, bsd in zip(self.buckets, sd["buckets"]):
