@@ -66,12 +66,17 @@ class FlashAttentionFunction(torch.autograd.Function):
     the round-2 follow-up.
     """
 
-    CHUNK = 256  # q rows recomputed per tile
+    CHUNK = 256  # q rows recomputed per tile (CPU fallback)
 
-    # fused MFMA backward validated on hardware round 2
-    # (profiles/validate_fmha_r2.log); the chunked-recompute composition
-    # below remains the CPU / debug fallback
-    use_fused_backward = True
+    # Backward route. The hand-written MFMA backward (csrc/fmha.hip
+    # fmha_bwd_*) is numerics-validated on hardware but its column-strided
+    # B-fragment loads cap it at ~190 TF — measured SLOWER than recomputing
+    # through hipBLASLt batched GEMMs (~1 PF): probe_fmha_perf.log r2 shows
+    # raw fused bwd 2.48 ms vs composed bwd ~1.3 ms at the BERT shape. So
+    # the default GPU backward is the batched-GEMM recompute below;
+    # use_fused_backward=True re-enables the MFMA kernels (for future
+    # LDS-staged revisions).
+    use_fused_backward = False
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
@@ -90,6 +95,25 @@ class FlashAttentionFunction(torch.autograd.Function):
             ext = get_ext("mfma")
             dq, dk, dv = ext.fmha_bwd(dout, q, k, v, out, lse, causal, float(scale))
             return dq, dk, dv, None, None
+        if q.is_cuda:
+            # batched-GEMM recompute: 5 hipBLASLt GEMMs + 2 elementwise
+            # passes; P recomputed from (q, k, lse) — standard flash
+            # identities, D_i = rowsum(dO * O)
+            S = q.shape[-2]
+            delta = (dout.float() * out.float()).sum(-1, keepdim=True)
+            s = torch.matmul(q, k.transpose(-1, -2))
+            p32 = torch.exp(s.float() * scale - lse.unsqueeze(-1))
+            if causal:
+                mask = torch.triu(
+                    torch.ones(S, S, dtype=torch.bool, device=q.device), 1)
+                p32 = p32.masked_fill(mask, 0.0)
+            p = p32.to(q.dtype)
+            dv = torch.matmul(p.transpose(-1, -2), dout)
+            dp = torch.matmul(dout, v.transpose(-1, -2))
+            ds = (p32 * (dp.float() - delta)).to(q.dtype)
+            dq = torch.matmul(ds, k) * scale
+            dk = torch.matmul(ds.transpose(-1, -2), q) * scale
+            return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
         S = q.shape[-2]
         qf, kf, vf = q.float(), k.float(), v.float()
         dof = dout.float()

@@ -54,3 +54,33 @@ def test_fmha_fused_backward_matches_eager(causal):
     torch.testing.assert_close(dq.float(), qr.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dk.float(), kr.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(dv.float(), vr.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_flash_autograd_gemm_recompute_backward(causal):
+    """Default GPU backward (hipBLASLt batched-GEMM recompute) vs fp32
+    autograd reference."""
+    from apex_amd.transformer import flash_attention
+
+    B, H, S, D = 2, 4, 256, 64
+    torch.manual_seed(3)
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    dout = torch.randn_like(q)
+    out = flash_attention(q, k, v, causal=causal)
+    out.backward(dout)
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    s = torch.matmul(qr, kr.transpose(-1, -2)) / (D ** 0.5)
+    if causal:
+        s = s.masked_fill(torch.triu(torch.ones(S, S, dtype=torch.bool,
+                                                device="cuda"), 1), float("-inf"))
+    ref = torch.matmul(torch.softmax(s, -1), vr)
+    ref.backward(dout.float())
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(q.grad.float(), qr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), kr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), vr.grad, rtol=5e-2, atol=5e-2)
