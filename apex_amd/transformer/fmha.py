@@ -96,24 +96,20 @@ class FlashAttentionFunction(torch.autograd.Function):
             dq, dk, dv = ext.fmha_bwd(dout, q, k, v, out, lse, causal, float(scale))
             return dq, dk, dv, None, None
         if q.is_cuda:
-            # batched-GEMM recompute: 5 hipBLASLt GEMMs + 2 elementwise
-            # passes; P recomputed from (q, k, lse) — standard flash
-            # identities, D_i = rowsum(dO * O)
-            S = q.shape[-2]
-            delta = (dout.float() * out.float()).sum(-1, keepdim=True)
-            s = torch.matmul(q, k.transpose(-1, -2))
-            p32 = torch.exp(s.float() * scale - lse.unsqueeze(-1))
-            if causal:
-                mask = torch.triu(
-                    torch.ones(S, S, dtype=torch.bool, device=q.device), 1)
-                p32 = p32.masked_fill(mask, 0.0)
-            p = p32.to(q.dtype)
+            # batched-GEMM recompute: 5 hipBLASLt GEMMs + 3 single-pass
+            # fused kernels (delta / p / ds from csrc/fmha.hip) — P is
+            # recomputed from (q, k, lse); the scale for dq/dk is folded
+            # into the ds kernel
+            ext = get_ext("mfma")
+            delta = ext.fmha_delta(dout, out)                 # [B,H,S] fp32
+            s = torch.matmul(q, k.transpose(-1, -2))          # bf16 GEMM
+            p = ext.fmha_p(s, lse, float(scale), causal)      # bf16, 1 pass
             dv = torch.matmul(p.transpose(-1, -2), dout)
             dp = torch.matmul(dout, v.transpose(-1, -2))
-            ds = (p32 * (dp.float() - delta)).to(q.dtype)
-            dq = torch.matmul(ds, k) * scale
-            dk = torch.matmul(ds.transpose(-1, -2), q) * scale
-            return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+            ds = ext.fmha_ds(p, dp, delta, float(scale))      # bf16, 1 pass
+            dq = torch.matmul(ds, k)
+            dk = torch.matmul(ds.transpose(-1, -2), q)
+            return dq, dk, dv, None, None
         S = q.shape[-2]
         qf, kf, vf = q.float(), k.float(), v.float()
         dof = dout.float()

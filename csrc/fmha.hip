@@ -379,7 +379,146 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   }
 }
 
+// ---------------- GEMM-recompute backward helpers ----------------
+// The production backward is 5 hipBLASLt batched GEMMs (torch.matmul) plus
+// these three single-pass kernels — one HBM pass each instead of the eager
+// fp32 mul/exp/mask/cast chains that cost ~6 extra S x S passes.
+
+// delta[row] = sum_d dout[row,d] * out[row,d]; one wave per row.
+template <int D>
+__global__ void __launch_bounds__(256) fmha_delta_kernel(
+    const short* __restrict__ dO, const short* __restrict__ O, float* __restrict__ delta,
+    long rows) {
+  const int lane = threadIdx.x & 63;
+  const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  if (row >= rows) return;
+  constexpr int PER = D / 64;  // elements per lane (1 for D=64, 2 for D=128)
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < PER; ++j) {
+    const long i = row * D + j * 64 + lane;
+    const __hip_bfloat16 a = *reinterpret_cast<const __hip_bfloat16*>(dO + i);
+    const __hip_bfloat16 b = *reinterpret_cast<const __hip_bfloat16*>(O + i);
+    acc = fmaf(__bfloat162float(a), __bfloat162float(b), acc);
+  }
+#pragma unroll
+  for (int m = 1; m < 64; m <<= 1) acc += __shfl_xor(acc, m);
+  if (lane == 0) delta[row] = acc;
+}
+
+// p[row, c] = exp(s[row, c] * scale - lse[row]) (0 where causal-masked);
+// rows are (bh, q) pairs, c the kv column. 8-wide bf16 vectors (S % 32 == 0).
+template <bool CAUSAL>
+__global__ void __launch_bounds__(256) fmha_p_kernel(
+    const short* __restrict__ Sm, const float* __restrict__ lse, short* __restrict__ P,
+    long rows, long S, float scale) {
+  const long total_vec = rows * (S / 8);
+  for (long vi = (long)blockIdx.x * blockDim.x + threadIdx.x; vi < total_vec;
+       vi += (long)gridDim.x * blockDim.x) {
+    const long row = vi / (S / 8);
+    const long c0 = (vi % (S / 8)) * 8;
+    const float l = lse[row];
+    const long qrow = row % S;  // q index within the head (rows = BH*S)
+    bf16x8 sv = *reinterpret_cast<const bf16x8*>(Sm + row * S + c0);
+    bf16x8 pv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const __hip_bfloat16 sb = *reinterpret_cast<const __hip_bfloat16*>(&sv[j]);
+      float p = __expf(__bfloat162float(sb) * scale - l);
+      if (CAUSAL && (c0 + j) > qrow) p = 0.f;
+      const __hip_bfloat16 pb = __float2bfloat16(p);
+      pv[j] = *reinterpret_cast<const short*>(&pb);
+    }
+    *reinterpret_cast<bf16x8*>(P + row * S + c0) = pv;
+  }
+}
+
+// ds[row, c] = p[row, c] * (dp[row, c] - delta[row]) * scale
+__global__ void __launch_bounds__(256) fmha_ds_kernel(
+    const short* __restrict__ P, const short* __restrict__ dP, const float* __restrict__ delta,
+    short* __restrict__ dS, long rows, long S, float scale) {
+  const long total_vec = rows * (S / 8);
+  for (long vi = (long)blockIdx.x * blockDim.x + threadIdx.x; vi < total_vec;
+       vi += (long)gridDim.x * blockDim.x) {
+    const long row = vi / (S / 8);
+    const long c0 = (vi % (S / 8)) * 8;
+    const float d = delta[row];
+    bf16x8 pv = *reinterpret_cast<const bf16x8*>(P + row * S + c0);
+    bf16x8 dpv = *reinterpret_cast<const bf16x8*>(dP + row * S + c0);
+    bf16x8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const __hip_bfloat16 pb = *reinterpret_cast<const __hip_bfloat16*>(&pv[j]);
+      const __hip_bfloat16 db = *reinterpret_cast<const __hip_bfloat16*>(&dpv[j]);
+      const float ds = __bfloat162float(pb) * (__bfloat162float(db) - d) * scale;
+      const __hip_bfloat16 ob = __float2bfloat16(ds);
+      ov[j] = *reinterpret_cast<const short*>(&ob);
+    }
+    *reinterpret_cast<bf16x8*>(dS + row * S + c0) = ov;
+  }
+}
+
 }  // namespace
+
+// delta = rowsum(dout * out) in fp32, shape [B, H, S]
+at::Tensor fmha_delta(at::Tensor dout, at::Tensor out) {
+  auto doc = dout.contiguous(), oc = out.contiguous();
+  const long D = doc.size(-1);
+  const long rows = doc.numel() / D;
+  TORCH_CHECK(D == 64 || D == 128, "fmha_delta: head_dim 64/128");
+  auto delta = at::empty(doc.sizes().slice(0, doc.dim() - 1),
+                         doc.options().dtype(at::kFloat));
+  auto stream = current_stream();
+  const long grid = std::min<long>((rows * 64 + 255) / 256, 65535);
+  if (D == 64)
+    hipLaunchKernelGGL((fmha_delta_kernel<64>), dim3((uint32_t)grid), dim3(256), 0, stream,
+                       (const short*)doc.data_ptr(), (const short*)oc.data_ptr(),
+                       delta.data_ptr<float>(), rows);
+  else
+    hipLaunchKernelGGL((fmha_delta_kernel<128>), dim3((uint32_t)grid), dim3(256), 0, stream,
+                       (const short*)doc.data_ptr(), (const short*)oc.data_ptr(),
+                       delta.data_ptr<float>(), rows);
+  HIP_CHECK(hipGetLastError());
+  return delta;
+}
+
+// p = exp(s * scale - lse[row]) (causal-masked); s [B,H,S,S] bf16, lse [B,H,S]
+at::Tensor fmha_p(at::Tensor s, at::Tensor lse, double scale, bool causal) {
+  auto sc = s.contiguous();
+  auto lc = lse.contiguous();
+  TORCH_CHECK(sc.scalar_type() == at::ScalarType::BFloat16, "fmha_p: bf16 s");
+  const long S = sc.size(-1);
+  TORCH_CHECK(S % 8 == 0, "fmha_p: S % 8");
+  const long rows = sc.numel() / S;
+  auto p = at::empty_like(sc);
+  auto stream = current_stream();
+  const long grid = std::min<long>((rows * (S / 8) + 255) / 256, 65535);
+  if (causal)
+    hipLaunchKernelGGL((fmha_p_kernel<true>), dim3((uint32_t)grid), dim3(256), 0, stream,
+                       (const short*)sc.data_ptr(), lc.data_ptr<float>(),
+                       (short*)p.data_ptr(), rows, S, (float)scale);
+  else
+    hipLaunchKernelGGL((fmha_p_kernel<false>), dim3((uint32_t)grid), dim3(256), 0, stream,
+                       (const short*)sc.data_ptr(), lc.data_ptr<float>(),
+                       (short*)p.data_ptr(), rows, S, (float)scale);
+  HIP_CHECK(hipGetLastError());
+  return p;
+}
+
+// ds = p * (dp - delta[row]) * scale
+at::Tensor fmha_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale) {
+  auto pc = p.contiguous(), dpc = dp.contiguous(), dc = delta.contiguous();
+  const long S = pc.size(-1);
+  const long rows = pc.numel() / S;
+  auto ds = at::empty_like(pc);
+  auto stream = current_stream();
+  const long grid = std::min<long>((rows * (S / 8) + 255) / 256, 65535);
+  hipLaunchKernelGGL(fmha_ds_kernel, dim3((uint32_t)grid), dim3(256), 0, stream,
+                     (const short*)pc.data_ptr(), (const short*)dpc.data_ptr(),
+                     dc.data_ptr<float>(), (short*)ds.data_ptr(), rows, S, (float)scale);
+  HIP_CHECK(hipGetLastError());
+  return ds;
+}
 
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
                                  double scale) {

@@ -15,6 +15,7 @@
 // Backward mirrors it; dgamma/dbeta use the NHWC column-reduction pattern.
 #include "common.h"
 
+#include <unordered_map>
 #include <vector>
 
 namespace {
@@ -222,12 +223,76 @@ int gn_splits(long work) {
   return (int)std::max<long>(1, std::min<long>(32, (work + 16383) / 16384));
 }
 
+// ---- one-pass persistent forward (reference analogue:
+// group_norm_nhwc_fwd_one_pass_kernel.cuh) ----
+// One workgroup per (n, g): the slab (HW x cpg) is streamed from HBM ONCE
+// into dynamic LDS (fp32 accumulation on the fly), stats reduced in-block,
+// then the normalize+affine+SiLU pass re-reads the slab from LDS — HBM
+// traffic drops from 3 slab passes (two-pass: stats read + apply read +
+// write) to 2 (read + write). Eligible when the slab fits the LDS budget;
+// the generic two-pass remains the fallback for everything else.
+constexpr long GN_ONEPASS_LDS_BYTES = 128 * 1024;  // of 160 KB per CU
+
+template <typename T, bool AFFINE, bool SILU>
+__global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
+    const T* __restrict__ x, T* __restrict__ y, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, const float* __restrict__ w, const float* __restrict__ b,
+    long HW, long C, long G, float eps) {
+  extern __shared__ short lds_raw[];  // slab as bf16/fp16 bit patterns (or float when T=float)
+  T* slab = reinterpret_cast<T*>(lds_raw);
+  const long ng = blockIdx.x;
+  const long n = ng / G, g = ng % G;
+  const long cpg = C / G;
+  const long c0 = g * cpg;
+  const long count = HW * cpg;
+  const T* xp = x + n * HW * C + c0;
+
+  // load + accumulate. Consecutive threads take consecutive channels within
+  // a row (cpg-wide contiguous segments in NHWC).
+  float sum = 0.f, sq = 0.f;
+  for (long i = threadIdx.x; i < count; i += blockDim.x) {
+    const long r = i / cpg;
+    const long c = i % cpg;
+    const T raw = xp[r * C + c];
+    slab[i] = raw;
+    const float v = to_float(raw);
+    sum += v;
+    sq = fmaf(v, v, sq);
+  }
+  __shared__ float smem[GN_BLOCK / WAVE_SIZE];
+  sum = block_reduce_sum(sum, smem);
+  sq = block_reduce_sum(sq, smem);
+  __shared__ float s_mu, s_rs;
+  if (threadIdx.x == 0) {
+    const float mu = sum / count;
+    const float var = fmaxf(sq / count - mu * mu, 0.f);
+    s_mu = mu;
+    s_rs = rsqrtf(var + eps);
+    mean_out[ng] = mu;
+    rstd_out[ng] = s_rs;
+  }
+  __syncthreads();
+  const float mu = s_mu, rs = s_rs;
+
+  T* yp = y + n * HW * C + c0;
+  for (long i = threadIdx.x; i < count; i += blockDim.x) {
+    const long r = i / cpg;
+    const long c = i % cpg;
+    float v = (to_float(slab[i]) - mu) * rs;
+    if (AFFINE) v = fmaf(v, w[c0 + c], b[c0 + c]);
+    if (SILU) v = v / (1.f + __expf(-v));
+    yp[r * C + c] = from_float<T>(v);
+  }
+}
+
 }  // namespace
 
 // x: [N, H, W, C] NHWC. Returns (y, mean[N*G], rstd[N*G]).
+// passes: 0 = auto heuristic, 1 = force one-pass (if eligible), 2 = force
+// two-pass — mirrors the reference's `passes` knob (group_norm.py:213-232).
 std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tensor> weight,
                                             c10::optional<at::Tensor> bias, long G, double eps,
-                                            bool silu) {
+                                            bool silu, long passes) {
   auto xc = x.contiguous();
   const long N = xc.size(0), C = xc.size(-1);
   const long HW = xc.numel() / (N * C);
@@ -244,7 +309,40 @@ std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tens
   auto b32 = affine ? bias->to(at::kFloat).contiguous() : at::Tensor();
   auto stream = current_stream();
 
+  // one-pass when the (HW x cpg) slab fits the LDS budget and there are
+  // enough (n, g) workgroups to occupy the 256 CUs
+  const long slab_bytes = HW * cpg * (long)xc.element_size();
+  const bool eligible = slab_bytes <= GN_ONEPASS_LDS_BYTES;
+  const bool one_pass = eligible && (passes == 1 || (passes == 0 && N * G >= 128));
+
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "group_norm_nhwc_fwd", ([&] {
+    if (one_pass) {
+      auto launch1 = [&](auto aff, auto sl) {
+        auto kfn = gn_fwd_onepass_kernel<scalar_t, decltype(aff)::value, decltype(sl)::value>;
+        if (slab_bytes > 64 * 1024) {
+          static std::unordered_map<const void*, bool> raised;
+          if (!raised[(const void*)kfn]) {
+            HIP_CHECK(hipFuncSetAttribute((const void*)kfn,
+                                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                                          (int)GN_ONEPASS_LDS_BYTES));
+            raised[(const void*)kfn] = true;
+          }
+        }
+        hipLaunchKernelGGL(kfn, dim3((uint32_t)(N * G)), dim3(GN_BLOCK), (size_t)slab_bytes,
+                           stream, (const scalar_t*)xc.data_ptr(), (scalar_t*)y.data_ptr(),
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           affine ? w32.data_ptr<float>() : nullptr,
+                           affine ? b32.data_ptr<float>() : nullptr, HW, C, G, (float)eps);
+      };
+      using Tt = std::true_type;
+      using Ff = std::false_type;
+      if (affine && silu) launch1(Tt{}, Tt{});
+      else if (affine) launch1(Tt{}, Ff{});
+      else if (silu) launch1(Ff{}, Tt{});
+      else launch1(Ff{}, Ff{});
+      HIP_CHECK(hipGetLastError());
+      return;
+    }
     hipLaunchKernelGGL((gn_fwd_stats_kernel<scalar_t>), dim3((uint32_t)(N * G), S),
                        dim3(GN_BLOCK), 0, stream, (const scalar_t*)xc.data_ptr(),
                        part.data_ptr<float>(), N, HW, C, G, S);
@@ -346,6 +444,8 @@ std::vector<at::Tensor> group_norm_nhwc_bwd(at::Tensor dy, at::Tensor x, at::Ten
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("fwd", &group_norm_nhwc_fwd, "NHWC GroupNorm forward (+SiLU) -> (y, mean, rstd)");
+  m.def("fwd", &group_norm_nhwc_fwd, "NHWC GroupNorm forward (+SiLU) -> (y, mean, rstd)",
+        py::arg("x"), py::arg("weight"), py::arg("bias"), py::arg("G"), py::arg("eps"),
+        py::arg("silu"), py::arg("passes") = 0);
   m.def("bwd", &group_norm_nhwc_bwd, "NHWC GroupNorm backward -> (dx, dgamma, dbeta)");
 }

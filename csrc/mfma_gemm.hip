@@ -325,12 +325,18 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
                                  double scale);
 std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor out, at::Tensor lse, bool causal, double scale);
+at::Tensor fmha_delta(at::Tensor dout, at::Tensor out);
+at::Tensor fmha_p(at::Tensor s, at::Tensor lse, double scale, bool causal);
+at::Tensor fmha_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fmha_fwd", &fmha_fwd,
-        "flash-attention fwd (bf16, D=64/128) -> (out, lse) [EXPERIMENTAL]");
+        "flash-attention fwd (bf16, D=64/128) -> (out, lse)");
   m.def("fmha_bwd", &fmha_bwd,
-        "flash-attention bwd -> (dq, dk, dv) [EXPERIMENTAL]");
+        "flash-attention MFMA bwd -> (dq, dk, dv)");
+  m.def("fmha_delta", &fmha_delta, "rowsum(dout*out) fp32 (one pass)");
+  m.def("fmha_p", &fmha_p, "p = exp(s*scale - lse[row]) (+causal mask), bf16 one pass");
+  m.def("fmha_ds", &fmha_ds, "ds = p*(dp - delta[row])*scale, bf16 one pass");
   m.def("gemm_bias_gelu", &gemm_bias_gelu_mfma, "bf16 MFMA GEMM + bias + tanh-GELU (fused)");
   m.def("gemm_bias", &gemm_bias_mfma, "bf16 MFMA GEMM + bias");
   m.def("mfma_tile_probe", &mfma_tile_probe, "single-tile fragment-layout verification");

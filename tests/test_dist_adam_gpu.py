@@ -136,11 +136,13 @@ def test_dist_adam_capturable_graph_replay():
     ref(xs[0]).pow(2).mean().backward()
     ropt.step()
 
-    # capture one step with grads staged in the buckets
+    # capture one step with grads staged in the buckets (capture RECORDS
+    # without executing — replay immediately to apply this iteration)
     backward_into_buckets(xs[1])
     g = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g):
         opt.step()
+    g.replay()
     ropt.zero_grad()
     ref(xs[1]).pow(2).mean().backward()
     ropt.step()
