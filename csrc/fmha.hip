@@ -423,7 +423,8 @@ __global__ void __launch_bounds__(256) fmha_p_kernel(
     bf16x8 pv;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const __hip_bfloat16 sb = *reinterpret_cast<const __hip_bfloat16*>(&sv[j]);
+      const short sraw = sv[j];
+      const __hip_bfloat16 sb = *reinterpret_cast<const __hip_bfloat16*>(&sraw);
       float p = __expf(__bfloat162float(sb) * scale - l);
       if (CAUSAL && (c0 + j) > qrow) p = 0.f;
       const __hip_bfloat16 pb = __float2bfloat16(p);
@@ -448,8 +449,10 @@ __global__ void __launch_bounds__(256) fmha_ds_kernel(
     bf16x8 ov;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const __hip_bfloat16 pb = *reinterpret_cast<const __hip_bfloat16*>(&pv[j]);
-      const __hip_bfloat16 db = *reinterpret_cast<const __hip_bfloat16*>(&dpv[j]);
+      const short praw = pv[j];
+      const short draw = dpv[j];
+      const __hip_bfloat16 pb = *reinterpret_cast<const __hip_bfloat16*>(&praw);
+      const __hip_bfloat16 db = *reinterpret_cast<const __hip_bfloat16*>(&draw);
       const float ds = __bfloat162float(pb) * (__bfloat162float(db) - d) * scale;
       const __hip_bfloat16 ob = __float2bfloat16(ds);
       ov[j] = *reinterpret_cast<const short*>(&ob);
