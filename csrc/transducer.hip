@@ -17,11 +17,54 @@ namespace {
 
 constexpr int TJ_BLOCK = 256;
 
-template <typename T, bool RELU>
+// ---- Philox4x32-10 counter-based RNG for the fused joint dropout
+// (reference analogue: apex/contrib/csrc/transducer/philox.cuh). The
+// counter is the flat OUTPUT index, so forward and both backward kernels
+// regenerate the identical mask from (seed, index) — mask-free backward.
+__device__ __forceinline__ void philox_round(uint32_t& c0, uint32_t& c1, uint32_t& c2,
+                                             uint32_t& c3, uint32_t k0, uint32_t k1) {
+  const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  const uint32_t hi0 = __umulhi(M0, c0), lo0 = M0 * c0;
+  const uint32_t hi1 = __umulhi(M1, c2), lo1 = M1 * c2;
+  c0 = hi1 ^ c1 ^ k0;
+  c1 = lo1;
+  c2 = hi0 ^ c3 ^ k1;
+  c3 = lo0;
+}
+
+__device__ __forceinline__ float philox_uniform(unsigned long long seed, long idx) {
+  uint32_t c0 = (uint32_t)((unsigned long)idx >> 2);
+  uint32_t c1 = (uint32_t)((unsigned long)idx >> 34);
+  uint32_t c2 = 0u, c3 = 0u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  uint32_t res;
+  switch (idx & 3) {
+    case 0: res = c0; break;
+    case 1: res = c1; break;
+    case 2: res = c2; break;
+    default: res = c3; break;
+  }
+  // 24-bit mantissa -> [0, 1)
+  return (res >> 8) * (1.0f / 16777216.0f);
+}
+
+// d *= mask(idx) / (1-p)  (0 when dropped)
+__device__ __forceinline__ float apply_dropout(float v, unsigned long long seed, long idx,
+                                               float p, float rinv) {
+  return (philox_uniform(seed, idx) >= p) ? v * rinv : 0.f;
+}
+
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_kernel(
     const T* __restrict__ f, const T* __restrict__ g, T* __restrict__ out,
     const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
-    long H) {
+    long H, float p, float rinv, unsigned long long seed) {
   const long total = B * Tm * U * H;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
@@ -33,17 +76,18 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_kernel(
     if (t < f_len[b] && u < g_len[b]) {
       v = to_float(f[(b * Tm + t) * H + h]) + to_float(g[(b * U + u) * H + h]);
       if (RELU) v = fmaxf(v, 0.f);
+      if (DROPOUT) v = apply_dropout(v, seed, i, p, rinv);
     }
     out[i] = from_float<T>(v);
   }
 }
 
 // grad_f[b,t,h] = sum_u dout[b,t,u,h] (masked); one block per (b,t) row.
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_f_kernel(
     const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ df,
     const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
-    long H) {
+    long H, float p, float rinv, unsigned long long seed) {
   const long bt = blockIdx.x;
   const long b = bt / Tm, t = bt % Tm;
   const int ulen = (t < f_len[b]) ? g_len[b] : 0;
@@ -52,6 +96,7 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_f_kernel(
     for (long u = 0; u < ulen; ++u) {
       const long idx = ((b * Tm + t) * U + u) * H + h;
       float d = to_float(dout[idx]);
+      if (DROPOUT) d = apply_dropout(d, seed, idx, p, rinv);
       if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
       acc += d;
     }
@@ -59,11 +104,11 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_f_kernel(
   }
 }
 
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_g_kernel(
     const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ dg,
     const int* __restrict__ f_len, const int* __restrict__ g_len, long B, long Tm, long U,
-    long H) {
+    long H, float p, float rinv, unsigned long long seed) {
   const long bu = blockIdx.x;
   const long b = bu / U, u = bu % U;
   const int tlen = (u < g_len[b]) ? f_len[b] : 0;
@@ -72,6 +117,7 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_g_kernel(
     for (long t = 0; t < tlen; ++t) {
       const long idx = ((b * Tm + t) * U + u) * H + h;
       float d = to_float(dout[idx]);
+      if (DROPOUT) d = apply_dropout(d, seed, idx, p, rinv);
       if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
       acc += d;
     }
@@ -81,12 +127,12 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_g_kernel(
 
 // ---- packed joint: out rows = sum_b f_len[b]*g_len[b] (batch_offset is
 // the EXCLUSIVE prefix of f_len*g_len) ----
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_packed_kernel(
     const T* __restrict__ f, const T* __restrict__ g, T* __restrict__ out,
     const int* __restrict__ f_len, const int* __restrict__ g_len,
     const long* __restrict__ off /* [B+1] exclusive */, long B, long Tm, long U, long H,
-    long total_rows) {
+    long total_rows, float p, float rinv, unsigned long long seed) {
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total_rows * H;
        i += (long)gridDim.x * blockDim.x) {
     const long row = i / H;
@@ -104,15 +150,17 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_fwd_packed_kernel(
     const long u = local % gl;
     float v = to_float(f[(b * Tm + t) * H + h]) + to_float(g[(b * U + u) * H + h]);
     if (RELU) v = fmaxf(v, 0.f);
+    if (DROPOUT) v = apply_dropout(v, seed, i, p, rinv);
     out[i] = from_float<T>(v);
   }
 }
 
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_f_kernel(
     const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ df,
     const int* __restrict__ f_len, const int* __restrict__ g_len,
-    const long* __restrict__ off, long B, long Tm, long U, long H) {
+    const long* __restrict__ off, long B, long Tm, long U, long H, float p, float rinv,
+    unsigned long long seed) {
   const long bt = blockIdx.x;
   const long b = bt / Tm, t = bt % Tm;
   const int ulen = (t < f_len[b]) ? g_len[b] : 0;
@@ -122,6 +170,7 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_f_kernel(
     for (long u = 0; u < ulen; ++u) {
       const long idx = (base + u) * H + h;
       float d = to_float(dout[idx]);
+      if (DROPOUT) d = apply_dropout(d, seed, idx, p, rinv);
       if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
       acc += d;
     }
@@ -129,11 +178,12 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_f_kernel(
   }
 }
 
-template <typename T, bool RELU>
+template <typename T, bool RELU, bool DROPOUT>
 __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_g_kernel(
     const T* __restrict__ dout, const T* __restrict__ out, T* __restrict__ dg,
     const int* __restrict__ f_len, const int* __restrict__ g_len,
-    const long* __restrict__ off, long B, long Tm, long U, long H) {
+    const long* __restrict__ off, long B, long Tm, long U, long H, float p, float rinv,
+    unsigned long long seed) {
   const long bu = blockIdx.x;
   const long b = bu / U, u = bu % U;
   const int tlen = (u < g_len[b]) ? f_len[b] : 0;
@@ -142,6 +192,7 @@ __global__ void __launch_bounds__(TJ_BLOCK) joint_bwd_packed_g_kernel(
     for (long t = 0; t < tlen; ++t) {
       const long idx = (off[b] + t * (long)g_len[b] + u) * H + h;
       float d = to_float(dout[idx]);
+      if (DROPOUT) d = apply_dropout(d, seed, idx, p, rinv);
       if (RELU && to_float(out[idx]) <= 0.f) d = 0.f;
       acc += d;
     }
@@ -284,7 +335,8 @@ __global__ void __launch_bounds__(TJ_BLOCK) rnnt_grad_kernel(
 }  // namespace
 
 std::vector<at::Tensor> transducer_joint_forward(at::Tensor f, at::Tensor g, at::Tensor f_len,
-                                                 at::Tensor g_len, bool relu) {
+                                                 at::Tensor g_len, bool relu,
+                                                 double dropout_prob, long seed) {
   auto fc = f.contiguous();
   auto gc = g.contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
@@ -293,18 +345,22 @@ std::vector<at::Tensor> transducer_joint_forward(at::Tensor f, at::Tensor g, at:
   auto out = at::empty({B, Tm, U, H}, fc.options());
   const long total = out.numel();
   const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
+  const float p = (float)dropout_prob;
+  const float rinv = p > 0.f ? 1.f / (1.f - p) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
   APEX_DISPATCH_FLOAT_HALF_BF(fc.scalar_type(), "transducer_joint_forward", ([&] {
-    if (relu) {
-      hipLaunchKernelGGL((joint_fwd_kernel<scalar_t, true>), dim3(grid), dim3(TJ_BLOCK), 0,
-                         current_stream(), (const scalar_t*)fc.data_ptr(),
-                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-    } else {
-      hipLaunchKernelGGL((joint_fwd_kernel<scalar_t, false>), dim3(grid), dim3(TJ_BLOCK), 0,
-                         current_stream(), (const scalar_t*)fc.data_ptr(),
-                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-    }
+    auto launch = [&](auto rl, auto dp) {
+      hipLaunchKernelGGL((joint_fwd_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+                         dim3(grid), dim3(TJ_BLOCK), 0, current_stream(),
+                         (const scalar_t*)fc.data_ptr(), (const scalar_t*)gc.data_ptr(),
+                         (scalar_t*)out.data_ptr(), fl.data_ptr<int>(), gl.data_ptr<int>(), B,
+                         Tm, U, H, p, rinv, sd);
+    };
+    using Tt = std::true_type; using Ff = std::false_type;
+    if (relu && p > 0.f) launch(Tt{}, Tt{});
+    else if (relu) launch(Tt{}, Ff{});
+    else if (p > 0.f) launch(Ff{}, Tt{});
+    else launch(Ff{}, Ff{});
   }()));
   HIP_CHECK(hipGetLastError());
   return {out};
@@ -312,32 +368,34 @@ std::vector<at::Tensor> transducer_joint_forward(at::Tensor f, at::Tensor g, at:
 
 std::vector<at::Tensor> transducer_joint_backward(at::Tensor grad_out, at::Tensor out,
                                                   at::Tensor f_len, at::Tensor g_len, long B,
-                                                  long Tm, long U, long H, bool relu) {
+                                                  long Tm, long U, long H, bool relu,
+                                                  double dropout_prob, long seed) {
   auto dout = grad_out.contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
   auto gl = g_len.to(at::kInt).contiguous();
   auto df = at::empty({B, Tm, H}, dout.options());
   auto dg = at::empty({B, U, H}, dout.options());
+  const float p = (float)dropout_prob;
+  const float rinv = p > 0.f ? 1.f / (1.f - p) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
   APEX_DISPATCH_FLOAT_HALF_BF(dout.scalar_type(), "transducer_joint_backward", ([&] {
-    if (relu) {
-      hipLaunchKernelGGL((joint_bwd_f_kernel<scalar_t, true>), dim3((uint32_t)(B * Tm)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-      hipLaunchKernelGGL((joint_bwd_g_kernel<scalar_t, true>), dim3((uint32_t)(B * U)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-    } else {
-      hipLaunchKernelGGL((joint_bwd_f_kernel<scalar_t, false>), dim3((uint32_t)(B * Tm)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-      hipLaunchKernelGGL((joint_bwd_g_kernel<scalar_t, false>), dim3((uint32_t)(B * U)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
-                         fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm, U, H);
-    }
+    auto launch = [&](auto rl, auto dp) {
+      hipLaunchKernelGGL((joint_bwd_f_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+                         dim3((uint32_t)(B * Tm)), dim3(TJ_BLOCK), 0, current_stream(),
+                         (const scalar_t*)dout.data_ptr(), (const scalar_t*)out.data_ptr(),
+                         (scalar_t*)df.data_ptr(), fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm,
+                         U, H, p, rinv, sd);
+      hipLaunchKernelGGL((joint_bwd_g_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+                         dim3((uint32_t)(B * U)), dim3(TJ_BLOCK), 0, current_stream(),
+                         (const scalar_t*)dout.data_ptr(), (const scalar_t*)out.data_ptr(),
+                         (scalar_t*)dg.data_ptr(), fl.data_ptr<int>(), gl.data_ptr<int>(), B, Tm,
+                         U, H, p, rinv, sd);
+    };
+    using Tt = std::true_type; using Ff = std::false_type;
+    if (relu && p > 0.f) launch(Tt{}, Tt{});
+    else if (relu) launch(Tt{}, Ff{});
+    else if (p > 0.f) launch(Ff{}, Tt{});
+    else launch(Ff{}, Ff{});
   }()));
   HIP_CHECK(hipGetLastError());
   return {df, dg};
@@ -434,7 +492,8 @@ at::Tensor transducer_loss_backward(at::Tensor x, at::Tensor label, at::Tensor a
 std::vector<at::Tensor> transducer_joint_forward_packed(at::Tensor f, at::Tensor g,
                                                         at::Tensor f_len, at::Tensor g_len,
                                                         at::Tensor batch_offset,
-                                                        long packed_batch, bool relu) {
+                                                        long packed_batch, bool relu,
+                                                        double dropout_prob, long seed) {
   auto fc = f.contiguous();
   auto gc = g.contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
@@ -444,20 +503,22 @@ std::vector<at::Tensor> transducer_joint_forward_packed(at::Tensor f, at::Tensor
   auto out = at::empty({packed_batch, H}, fc.options());
   const long total = packed_batch * H;
   const int grid = (int)std::min<long>((total + TJ_BLOCK - 1) / TJ_BLOCK, 8192);
+  const float p = (float)dropout_prob;
+  const float rinv = p > 0.f ? 1.f / (1.f - p) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
   APEX_DISPATCH_FLOAT_HALF_BF(fc.scalar_type(), "transducer_joint_forward_packed", ([&] {
-    if (relu) {
-      hipLaunchKernelGGL((joint_fwd_packed_kernel<scalar_t, true>), dim3(grid), dim3(TJ_BLOCK),
-                         0, current_stream(), (const scalar_t*)fc.data_ptr(),
-                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H, packed_batch);
-    } else {
-      hipLaunchKernelGGL((joint_fwd_packed_kernel<scalar_t, false>), dim3(grid), dim3(TJ_BLOCK),
-                         0, current_stream(), (const scalar_t*)fc.data_ptr(),
-                         (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H, packed_batch);
-    }
+    auto launch = [&](auto rl, auto dp) {
+      hipLaunchKernelGGL(
+          (joint_fwd_packed_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+          dim3(grid), dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)fc.data_ptr(),
+          (const scalar_t*)gc.data_ptr(), (scalar_t*)out.data_ptr(), fl.data_ptr<int>(),
+          glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm, U, H, packed_batch, p, rinv, sd);
+    };
+    using Tt = std::true_type; using Ff = std::false_type;
+    if (relu && p > 0.f) launch(Tt{}, Tt{});
+    else if (relu) launch(Tt{}, Ff{});
+    else if (p > 0.f) launch(Ff{}, Tt{});
+    else launch(Ff{}, Ff{});
   }()));
   HIP_CHECK(hipGetLastError());
   return {out};
@@ -466,47 +527,56 @@ std::vector<at::Tensor> transducer_joint_forward_packed(at::Tensor f, at::Tensor
 std::vector<at::Tensor> transducer_joint_backward_packed(at::Tensor grad_out, at::Tensor out,
                                                          at::Tensor f_len, at::Tensor g_len,
                                                          at::Tensor batch_offset, long B, long Tm,
-                                                         long U, long H, bool relu) {
+                                                         long U, long H, bool relu,
+                                                         double dropout_prob, long seed) {
   auto dout = grad_out.contiguous();
   auto fl = f_len.to(at::kInt).contiguous();
   auto glen = g_len.to(at::kInt).contiguous();
   auto off = batch_offset.to(at::kLong).contiguous();
   auto df = at::empty({B, Tm, H}, dout.options());
   auto dg = at::empty({B, U, H}, dout.options());
+  const float p = (float)dropout_prob;
+  const float rinv = p > 0.f ? 1.f / (1.f - p) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
   APEX_DISPATCH_FLOAT_HALF_BF(dout.scalar_type(), "transducer_joint_backward_packed", ([&] {
-    if (relu) {
-      hipLaunchKernelGGL((joint_bwd_packed_f_kernel<scalar_t, true>), dim3((uint32_t)(B * Tm)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H);
-      hipLaunchKernelGGL((joint_bwd_packed_g_kernel<scalar_t, true>), dim3((uint32_t)(B * U)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H);
-    } else {
-      hipLaunchKernelGGL((joint_bwd_packed_f_kernel<scalar_t, false>), dim3((uint32_t)(B * Tm)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)df.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H);
-      hipLaunchKernelGGL((joint_bwd_packed_g_kernel<scalar_t, false>), dim3((uint32_t)(B * U)),
-                         dim3(TJ_BLOCK), 0, current_stream(), (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)out.data_ptr(), (scalar_t*)dg.data_ptr(),
-                         fl.data_ptr<int>(), glen.data_ptr<int>(), off.data_ptr<long>(), B, Tm,
-                         U, H);
-    }
+    auto launch = [&](auto rl, auto dp) {
+      hipLaunchKernelGGL(
+          (joint_bwd_packed_f_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+          dim3((uint32_t)(B * Tm)), dim3(TJ_BLOCK), 0, current_stream(),
+          (const scalar_t*)dout.data_ptr(), (const scalar_t*)out.data_ptr(),
+          (scalar_t*)df.data_ptr(), fl.data_ptr<int>(), glen.data_ptr<int>(),
+          off.data_ptr<long>(), B, Tm, U, H, p, rinv, sd);
+      hipLaunchKernelGGL(
+          (joint_bwd_packed_g_kernel<scalar_t, decltype(rl)::value, decltype(dp)::value>),
+          dim3((uint32_t)(B * U)), dim3(TJ_BLOCK), 0, current_stream(),
+          (const scalar_t*)dout.data_ptr(), (const scalar_t*)out.data_ptr(),
+          (scalar_t*)dg.data_ptr(), fl.data_ptr<int>(), glen.data_ptr<int>(),
+          off.data_ptr<long>(), B, Tm, U, H, p, rinv, sd);
+    };
+    using Tt = std::true_type; using Ff = std::false_type;
+    if (relu && p > 0.f) launch(Tt{}, Tt{});
+    else if (relu) launch(Tt{}, Ff{});
+    else if (p > 0.f) launch(Ff{}, Tt{});
+    else launch(Ff{}, Ff{});
   }()));
   HIP_CHECK(hipGetLastError());
   return {df, dg};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("joint_forward", &transducer_joint_forward);
-  m.def("joint_backward", &transducer_joint_backward);
+  m.def("joint_forward", &transducer_joint_forward, py::arg("f"), py::arg("g"),
+        py::arg("f_len"), py::arg("g_len"), py::arg("relu"), py::arg("dropout_prob") = 0.0,
+        py::arg("seed") = 0);
+  m.def("joint_backward", &transducer_joint_backward, py::arg("grad_out"), py::arg("out"),
+        py::arg("f_len"), py::arg("g_len"), py::arg("B"), py::arg("Tm"), py::arg("U"),
+        py::arg("H"), py::arg("relu"), py::arg("dropout_prob") = 0.0, py::arg("seed") = 0);
   m.def("loss_forward", &transducer_loss_forward);
   m.def("loss_backward", &transducer_loss_backward);
-  m.def("joint_forward_packed", &transducer_joint_forward_packed);
-  m.def("joint_backward_packed", &transducer_joint_backward_packed);
+  m.def("joint_forward_packed", &transducer_joint_forward_packed, py::arg("f"), py::arg("g"),
+        py::arg("f_len"), py::arg("g_len"), py::arg("batch_offset"), py::arg("packed_batch"),
+        py::arg("relu"), py::arg("dropout_prob") = 0.0, py::arg("seed") = 0);
+  m.def("joint_backward_packed", &transducer_joint_backward_packed, py::arg("grad_out"),
+        py::arg("out"), py::arg("f_len"), py::arg("g_len"), py::arg("batch_offset"),
+        py::arg("B"), py::arg("Tm"), py::arg("U"), py::arg("H"), py::arg("relu"),
+        py::arg("dropout_prob") = 0.0, py::arg("seed") = 0);
 }
