@@ -2,10 +2,13 @@
 
 API parity with the reference ``apex.contrib.transducer``
 (apex/contrib/transducer/transducer.py: TransducerJoint:6, TransducerLoss:88).
-Round-1 scope notes (documented gaps, not silent fallbacks):
+Notes:
 - packed layouts are supported (``pack_output`` with ``batch_offset`` =
   inclusive cumsum of f_len*g_len, and ``packed_input`` for the loss);
-- fused dropout inside the joint runs as a torch dropout on the joint output;
+- dropout is FUSED into the joint kernels (round 2): a Philox counter keyed
+  on the flat output index generates the mask inside forward AND both
+  backward reductions — mask-free backward, exactly the reference's
+  philox.cuh design (apex/contrib/csrc/transducer/philox.cuh);
 - ``fuse_softmax_backward`` is accepted; the loss consumes log-probs and
   returns grads w.r.t. them (the log_softmax backward is chained by autograd
   rather than fused into the loss kernel).
@@ -18,19 +21,25 @@ from ..._ext import get_ext
 
 class TransducerJointFunc(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, f, g, f_len, g_len, relu, batch_offset=None, packed_batch=0):
+    def forward(ctx, f, g, f_len, g_len, relu, batch_offset=None, packed_batch=0,
+                dropout_prob=0.0):
         ext = get_ext("transducer")
+        # philox seed drawn from torch's CPU RNG: deterministic under
+        # torch.manual_seed, shared by fwd and the mask-free backward
+        seed = int(torch.randint(0, 2 ** 62, (1,)).item()) if dropout_prob > 0 else 0
         if batch_offset is not None:
             # exclusive offsets from the reference's inclusive cumsum
             off = torch.cat([batch_offset.new_zeros(1), batch_offset[:-1]])
-            (out,) = ext.joint_forward_packed(f, g, f_len, g_len, off, packed_batch, relu)
+            (out,) = ext.joint_forward_packed(f, g, f_len, g_len, off, packed_batch, relu,
+                                              dropout_prob, seed)
             ctx.off = off
         else:
-            (out,) = ext.joint_forward(f, g, f_len, g_len, relu)
+            (out,) = ext.joint_forward(f, g, f_len, g_len, relu, dropout_prob, seed)
             ctx.off = None
         ctx.save_for_backward(out, f_len, g_len)
         ctx.dims = (f.size(0), f.size(1), g.size(1), f.size(2))
         ctx.relu = relu
+        ctx.dropout = (dropout_prob, seed)
         return out
 
     @staticmethod
@@ -38,12 +47,14 @@ class TransducerJointFunc(torch.autograd.Function):
         ext = get_ext("transducer")
         out, f_len, g_len = ctx.saved_tensors
         B, T, U, H = ctx.dims
+        p, seed = ctx.dropout
         if ctx.off is not None:
             df, dg = ext.joint_backward_packed(grad_out, out, f_len, g_len, ctx.off, B, T, U, H,
-                                               ctx.relu)
+                                               ctx.relu, p, seed)
         else:
-            df, dg = ext.joint_backward(grad_out, out, f_len, g_len, B, T, U, H, ctx.relu)
-        return df, dg, None, None, None, None, None
+            df, dg = ext.joint_backward(grad_out, out, f_len, g_len, B, T, U, H, ctx.relu,
+                                        p, seed)
+        return df, dg, None, None, None, None, None, None
 
 
 class TransducerJoint(torch.nn.Module):
@@ -57,14 +68,12 @@ class TransducerJoint(torch.nn.Module):
         self.mask_probe = [] if (relu or dropout) and probe_mask else None
 
     def forward(self, f, g, f_len, g_len, batch_offset=None, packed_batch=0):
+        p = self.dropout_prob if (self.dropout and self.training) else 0.0
         if self.pack_output:
             assert batch_offset is not None and packed_batch > 0, \
                 "pack_output needs batch_offset (cumsum of f_len*g_len) and packed_batch"
-            out = TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, batch_offset,
-                                            packed_batch)
-            if self.dropout and self.training:
-                out = torch.nn.functional.dropout(out, p=self.dropout_prob)
-            return out
+            return TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, batch_offset,
+                                             packed_batch, p)
         if not f.is_cuda:
             # reference math on CPU
             out = f.unsqueeze(2) + g.unsqueeze(1)
@@ -73,11 +82,10 @@ class TransducerJoint(torch.nn.Module):
             out = out * (mask_t & mask_u)
             if self.relu:
                 out = torch.relu(out)
-        else:
-            out = TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, None, 0)
-        if self.dropout and self.training:
-            out = torch.nn.functional.dropout(out, p=self.dropout_prob)
-        return out
+            if p > 0:
+                out = torch.nn.functional.dropout(out, p=p)
+            return out
+        return TransducerJointFunc.apply(f, g, f_len, g_len, self.relu, None, 0, p)
 
 
 class TransducerLossFunc(torch.autograd.Function):

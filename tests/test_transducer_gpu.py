@@ -165,3 +165,66 @@ def test_transducer_loss_packed_matches_dense():
         ref = x_dense.grad[b, :int(f_len[b]), :int(y_len[b]) + 1, :].reshape(-1, V)
         torch.testing.assert_close(x_packed.grad[off:off + n], ref, rtol=1e-4, atol=1e-5)
         off += n
+
+
+def test_joint_fused_dropout():
+    """Fused philox dropout: scaled kept values, ~p drop rate, deterministic
+    under torch.manual_seed, and the backward mask matches the forward mask
+    (grads flow only through kept elements)."""
+    from apex_amd.contrib.transducer import TransducerJoint
+
+    torch.manual_seed(42)
+    B, T, U, H = 4, 24, 12, 64
+    f = torch.randn(B, T, H, device="cuda", requires_grad=True)
+    g = torch.randn(B, U, H, device="cuda", requires_grad=True)
+    f_len = torch.full((B,), T, dtype=torch.int32, device="cuda")
+    g_len = torch.full((B,), U, dtype=torch.int32, device="cuda")
+    p = 0.3
+
+    joint = TransducerJoint(dropout=True, dropout_prob=p).train()
+    torch.manual_seed(7)
+    out = joint(f, g, f_len, g_len)
+
+    base = f.detach().unsqueeze(2) + g.detach().unsqueeze(1)
+    kept = out != 0
+    # kept values are base / (1-p)
+    torch.testing.assert_close(out[kept], (base / (1 - p))[kept], rtol=1e-5, atol=1e-5)
+    rate = 1.0 - kept.float().mean().item()
+    assert abs(rate - p) < 0.02, f"drop rate {rate} vs p {p}"
+
+    # determinism: same torch seed -> same mask
+    torch.manual_seed(7)
+    out2 = joint(f, g, f_len, g_len)
+    assert torch.equal(out.detach(), out2.detach())
+
+    # backward: grads flow exactly through the kept elements with 1/(1-p)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    fr = f.detach().clone().requires_grad_(True)
+    gr = g.detach().clone().requires_grad_(True)
+    ref = (fr.unsqueeze(2) + gr.unsqueeze(1)) * kept.float() / (1 - p)
+    ref.backward(dout)
+    torch.testing.assert_close(f.grad, fr.grad, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(g.grad, gr.grad, rtol=1e-4, atol=1e-4)
+
+
+def test_joint_fused_dropout_packed():
+    from apex_amd.contrib.transducer import TransducerJoint
+
+    torch.manual_seed(1)
+    B, T, U, H = 3, 10, 6, 32
+    f = torch.randn(B, T, H, device="cuda", requires_grad=True)
+    g = torch.randn(B, U, H, device="cuda", requires_grad=True)
+    f_len = torch.tensor([10, 7, 5], dtype=torch.int32, device="cuda")
+    g_len = torch.tensor([6, 4, 3], dtype=torch.int32, device="cuda")
+    batch_offset = torch.cumsum(f_len.long() * g_len.long(), 0)
+    packed = int(batch_offset[-1])
+    p = 0.25
+    joint = TransducerJoint(pack_output=True, dropout=True, dropout_prob=p).train()
+    out = joint(f, g, f_len, g_len, batch_offset=batch_offset, packed_batch=packed)
+    assert out.shape == (packed, H)
+    kept = out != 0
+    rate = 1.0 - kept.float().mean().item()
+    assert abs(rate - p) < 0.05
+    out.sum().backward()
+    assert torch.isfinite(f.grad).all() and torch.isfinite(g.grad).all()
