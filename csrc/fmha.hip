@@ -568,8 +568,8 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
   auto dq = at::empty_like(qc);
   auto dk = at::empty_like(kc);
   auto dv = at::empty_like(vc);
-  // delta = rowsum(dO * O) in fp32 — a cheap eager reduction
-  auto delta = (doc.to(at::kFloat) * oc.to(at::kFloat)).sum(-1).contiguous();
+  // delta = rowsum(dO * O) in fp32 — one fused pass (fmha_delta kernel)
+  auto delta = fmha_delta(doc, oc);
   auto stream = current_stream();
   dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
   dim3 block(FM_WAVES * 64);

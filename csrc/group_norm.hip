@@ -223,23 +223,24 @@ int gn_splits(long work) {
   return (int)std::max<long>(1, std::min<long>(32, (work + 16383) / 16384));
 }
 
-// ---- one-pass persistent forward (reference analogue:
+// ---- one-pass forward (reference analogue:
 // group_norm_nhwc_fwd_one_pass_kernel.cuh) ----
-// One workgroup per (n, g): the slab (HW x cpg) is streamed from HBM ONCE
-// into dynamic LDS (fp32 accumulation on the fly), stats reduced in-block,
-// then the normalize+affine+SiLU pass re-reads the slab from LDS — HBM
-// traffic drops from 3 slab passes (two-pass: stats read + apply read +
-// write) to 2 (read + write). Eligible when the slab fits the LDS budget;
-// the generic two-pass remains the fallback for everything else.
-constexpr long GN_ONEPASS_LDS_BYTES = 128 * 1024;  // of 160 KB per CU
+// One workgroup per (n, g): the slab (HW x cpg) is read for the stats, then
+// re-read for the normalize pass. Unlike the two-pass form — where ALL
+// slabs are swept between the two reads so the apply kernel always misses
+// L2 — the re-read here happens while the block's own slab (<= ~1 MB) is
+// still resident in its XCD's 4 MB L2, so HBM sees the data once. An LDS
+// slab was tried first and was SLOWER (profiles/probe_group_norm.log r2
+// call 5): 80-128 KB of LDS per workgroup caps the CU at 1 block / 4 waves,
+// too little latency hiding for a streaming kernel. No LDS, full occupancy,
+// L2 provides the persistence.
+constexpr long GN_ONEPASS_SLAB_BYTES = 1 * 1024 * 1024;  // per-block L2 budget
 
 template <typename T, bool AFFINE, bool SILU>
 __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
     const T* __restrict__ x, T* __restrict__ y, float* __restrict__ mean_out,
     float* __restrict__ rstd_out, const float* __restrict__ w, const float* __restrict__ b,
     long HW, long C, long G, float eps) {
-  extern __shared__ short lds_raw[];  // slab as bf16/fp16 bit patterns (or float when T=float)
-  T* slab = reinterpret_cast<T*>(lds_raw);
   const long ng = blockIdx.x;
   const long n = ng / G, g = ng % G;
   const long cpg = C / G;
@@ -247,15 +248,13 @@ __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
   const long count = HW * cpg;
   const T* xp = x + n * HW * C + c0;
 
-  // load + accumulate. Consecutive threads take consecutive channels within
-  // a row (cpg-wide contiguous segments in NHWC).
+  // pass A: stats. Consecutive threads take consecutive channels within a
+  // row (cpg-wide contiguous segments in NHWC).
   float sum = 0.f, sq = 0.f;
   for (long i = threadIdx.x; i < count; i += blockDim.x) {
     const long r = i / cpg;
     const long c = i % cpg;
-    const T raw = xp[r * C + c];
-    slab[i] = raw;
-    const float v = to_float(raw);
+    const float v = to_float(xp[r * C + c]);
     sum += v;
     sq = fmaf(v, v, sq);
   }
@@ -274,11 +273,12 @@ __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
   __syncthreads();
   const float mu = s_mu, rs = s_rs;
 
+  // pass B: normalize from the (L2-resident) slab
   T* yp = y + n * HW * C + c0;
   for (long i = threadIdx.x; i < count; i += blockDim.x) {
     const long r = i / cpg;
     const long c = i % cpg;
-    float v = (to_float(slab[i]) - mu) * rs;
+    float v = (to_float(xp[r * C + c]) - mu) * rs;
     if (AFFINE) v = fmaf(v, w[c0 + c], b[c0 + c]);
     if (SILU) v = v / (1.f + __expf(-v));
     yp[r * C + c] = from_float<T>(v);
@@ -309,27 +309,18 @@ std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tens
   auto b32 = affine ? bias->to(at::kFloat).contiguous() : at::Tensor();
   auto stream = current_stream();
 
-  // one-pass when the (HW x cpg) slab fits the LDS budget and there are
-  // enough (n, g) workgroups to occupy the 256 CUs
+  // one-pass when the (HW x cpg) slab stays L2-resident per block and there
+  // are enough (n, g) workgroups to occupy the 256 CUs
   const long slab_bytes = HW * cpg * (long)xc.element_size();
-  const bool eligible = slab_bytes <= GN_ONEPASS_LDS_BYTES;
-  const bool one_pass = eligible && (passes == 1 || (passes == 0 && N * G >= 128));
+  const bool eligible = slab_bytes <= GN_ONEPASS_SLAB_BYTES;
+  const bool one_pass = eligible && (passes == 1 || (passes == 0 && N * G >= 256));
 
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "group_norm_nhwc_fwd", ([&] {
     if (one_pass) {
       auto launch1 = [&](auto aff, auto sl) {
         auto kfn = gn_fwd_onepass_kernel<scalar_t, decltype(aff)::value, decltype(sl)::value>;
-        if (slab_bytes > 64 * 1024) {
-          static std::unordered_map<const void*, bool> raised;
-          if (!raised[(const void*)kfn]) {
-            HIP_CHECK(hipFuncSetAttribute((const void*)kfn,
-                                          hipFuncAttributeMaxDynamicSharedMemorySize,
-                                          (int)GN_ONEPASS_LDS_BYTES));
-            raised[(const void*)kfn] = true;
-          }
-        }
-        hipLaunchKernelGGL(kfn, dim3((uint32_t)(N * G)), dim3(GN_BLOCK), (size_t)slab_bytes,
-                           stream, (const scalar_t*)xc.data_ptr(), (scalar_t*)y.data_ptr(),
+        hipLaunchKernelGGL(kfn, dim3((uint32_t)(N * G)), dim3(GN_BLOCK), 0, stream,
+                           (const scalar_t*)xc.data_ptr(), (scalar_t*)y.data_ptr(),
                            mean.data_ptr<float>(), rstd.data_ptr<float>(),
                            affine ? w32.data_ptr<float>() : nullptr,
                            affine ? b32.data_ptr<float>() : nullptr, HW, C, G, (float)eps);

@@ -36,9 +36,9 @@ def main():
         w = torch.randn(C, device="cuda", dtype=torch.float32)
         b = torch.randn(C, device="cuda", dtype=torch.float32)
         slab_kb = H * W * (C // G) * 2 / 1024
-        one_pass_active = slab_kb <= 128 and N * G >= 128
+        one_pass_active = slab_kb <= 1024 and N * G >= 256
 
-        t1 = timeit(lambda: gn.fwd(x, w, b, G, 1e-5, True, 1)) if one_pass_active or slab_kb <= 128 else float("nan")
+        t1 = timeit(lambda: gn.fwd(x, w, b, G, 1e-5, True, 1)) if slab_kb <= 1024 else float("nan")
         t2 = timeit(lambda: gn.fwd(x, w, b, G, 1e-5, True, 2))
         t = t1 if one_pass_active else t2
         # reference numerics (fp32 eager)
