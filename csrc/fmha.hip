@@ -30,6 +30,26 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 constexpr int FM_ROWS = 16;   // q rows per wave
 constexpr int FM_WAVES = 4;   // waves per workgroup
 constexpr int FM_BN = 32;     // kv tile (one MFMA K step for P@V)
+constexpr int FM_PAD = 8;     // LDS tile row padding (shorts) -> 16B-aligned rows
+
+// Stage one 32 x D bf16 tile from HBM into LDS (row-major, padded rows),
+// wave-cooperative and fully vectorized (bf16x8 both sides). Replaces the
+// per-fragment scalar column-strided HBM gathers that capped the round-1
+// kernels at ~190 TF — B-fragments are then read from LDS.
+template <int D>
+__device__ __forceinline__ void fm_stage_tile(short* __restrict__ dst,
+                                              const short* __restrict__ src) {
+  constexpr int VPR = D / 8;    // vectors per row
+  constexpr int NV = FM_BN * VPR;
+  const int lane = threadIdx.x & 63;
+#pragma unroll
+  for (int v = lane; v < NV; v += 64) {
+    const int r = v / VPR;
+    const int c0 = (v % VPR) * 8;
+    *reinterpret_cast<bf16x8*>(dst + r * (D + FM_PAD) + c0) =
+        *reinterpret_cast<const bf16x8*>(src + (long)r * D + c0);
+  }
+}
 
 __device__ __forceinline__ float row_reduce_max16(float v) {
 #pragma unroll
@@ -60,7 +80,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const short* v_ptr = V + (long)bh * S * D;
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
+  __shared__ short lds_v[FM_WAVES][FM_BN * (D + FM_PAD)];
   short* pbuf = lds_p[wave];
+  short* vbuf = lds_v[wave];
 
   // Q rows for this wave, as A-fragments, resident for the whole pass
   bf16x8 aq[NK];
@@ -84,6 +106,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
+    fm_stage_tile<D>(vbuf, v_ptr + (long)kv0 * D);
     // ---- S = scale * Q K^T for this 16 x 32 tile (two 16x16 halves) ----
     f32x4 s_half[2];
 #pragma unroll
@@ -152,12 +175,12 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
     // ---- acc += P V (one MFMA per 16 head-dim columns) ----
 #pragma unroll
     for (int d = 0; d < ND; ++d) {
-      // B[k=kv][c=dim]: lane -> c = lane%16, k = (lane/16)*8 + j
+      // B[k=kv][c=dim] from the staged LDS tile: lane -> c = lane%16,
+      // k = (lane/16)*8 + j (8 short reads, conflict-free rows)
       bf16x8 bv;
 #pragma unroll
       for (int jj = 0; jj < 8; ++jj) {
-        const int kvr = kv0 + (lane >> 4) * 8 + jj;
-        bv[jj] = v_ptr[(long)kvr * D + d * 16 + (lane & 15)];
+        bv[jj] = vbuf[((lane >> 4) * 8 + jj) * (D + FM_PAD) + d * 16 + (lane & 15)];
       }
       acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc[d], 0, 0, 0);
     }
@@ -210,7 +233,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const float* delta = DELTA + (long)bh * S;
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
+  __shared__ short lds_k[FM_WAVES][FM_BN * (D + FM_PAD)];
   short* pbuf = lds_p[wave];
+  short* kbuf = lds_k[wave];
 
   bf16x8 aq[NK], ado[NK];
   const int a_row = q0 + (lane & 15);
@@ -227,15 +252,18 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
+    fm_stage_tile<D>(kbuf, k_ptr + (long)kv0 * D);
+    __builtin_amdgcn_s_waitcnt(0);  // staged K visible to this wave
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       f32x4 s = f32x4{0.f, 0.f, 0.f, 0.f};
       f32x4 dp = f32x4{0.f, 0.f, 0.f, 0.f};
       const int k_row = kv0 + j * 16 + (lane & 15);
+      const int k_lrow = j * 16 + (lane & 15);
 #pragma unroll
       for (int c = 0; c < NK; ++c) {
         const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
-            k_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+            kbuf + k_lrow * (D + FM_PAD) + c * 32 + (lane >> 4) * 8);
         s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[c], bk, s, 0, 0, 0);
         const bf16x8 bvt = *reinterpret_cast<const bf16x8*>(
             v_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
@@ -259,7 +287,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
       bf16x8 bK;
 #pragma unroll
       for (int jj = 0; jj < 8; ++jj)
-        bK[jj] = k_ptr[(long)(kv0 + (lane >> 4) * 8 + jj) * D + d * 16 + (lane & 15)];
+        bK[jj] = kbuf[((lane >> 4) * 8 + jj) * (D + FM_PAD) + d * 16 + (lane & 15)];
       dq_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, bK, dq_acc[d], 0, 0, 0);
     }
     __builtin_amdgcn_s_waitcnt(0);
@@ -298,8 +326,12 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
   __shared__ short lds_ds[FM_WAVES][FM_ROWS * FM_BN];
+  __shared__ short lds_q[FM_WAVES][FM_BN * (D + FM_PAD)];
+  __shared__ short lds_do[FM_WAVES][FM_BN * (D + FM_PAD)];
   short* pbuf = lds_p[wave];
   short* dsbuf = lds_ds[wave];
+  short* qbuf = lds_q[wave];
+  short* dobuf = lds_do[wave];
 
   bf16x8 ak[NK], av[NK];
   const int a_row = kv0 + (lane & 15);
@@ -318,18 +350,21 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int q_start = CAUSAL ? (kv0 / FM_BN) * FM_BN : 0;
 
   for (int q0 = q_start; q0 < S; q0 += FM_BN) {
+    fm_stage_tile<D>(qbuf, q_ptr + (long)q0 * D);
+    fm_stage_tile<D>(dobuf, do_ptr + (long)q0 * D);
+    __builtin_amdgcn_s_waitcnt(0);  // staged tiles visible to this wave
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       f32x4 sT = f32x4{0.f, 0.f, 0.f, 0.f};
       f32x4 dpT = f32x4{0.f, 0.f, 0.f, 0.f};
-      const int q_row = q0 + j * 16 + (lane & 15);
+      const int q_lrow = j * 16 + (lane & 15);
 #pragma unroll
       for (int c = 0; c < NK; ++c) {
         const bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-            q_ptr + (long)q_row * D + c * 32 + (lane >> 4) * 8);
+            qbuf + q_lrow * (D + FM_PAD) + c * 32 + (lane >> 4) * 8);
         sT = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[c], bq, sT, 0, 0, 0);
         const bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
-            do_ptr + (long)q_row * D + c * 32 + (lane >> 4) * 8);
+            dobuf + q_lrow * (D + FM_PAD) + c * 32 + (lane >> 4) * 8);
         dpT = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av[c], bdo, dpT, 0, 0, 0);
       }
 #pragma unroll
@@ -355,9 +390,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
       bf16x8 b_do, b_q;
 #pragma unroll
       for (int jj = 0; jj < 8; ++jj) {
-        const long qr = q0 + (lane >> 4) * 8 + jj;
-        b_do[jj] = do_ptr[qr * D + d * 16 + (lane & 15)];
-        b_q[jj] = q_ptr[qr * D + d * 16 + (lane & 15)];
+        const int qlr = (lane >> 4) * 8 + jj;
+        b_do[jj] = dobuf[qlr * (D + FM_PAD) + d * 16 + (lane & 15)];
+        b_q[jj] = qbuf[qlr * (D + FM_PAD) + d * 16 + (lane & 15)];
       }
       dv_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_do, dv_acc[d], 0, 0, 0);
       dk_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_q, dk_acc[d], 0, 0, 0);
