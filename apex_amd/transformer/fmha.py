@@ -68,15 +68,14 @@ class FlashAttentionFunction(torch.autograd.Function):
 
     CHUNK = 256  # q rows recomputed per tile (CPU fallback)
 
-    # Backward route. The hand-written MFMA backward (csrc/fmha.hip
-    # fmha_bwd_*) is numerics-validated on hardware but its column-strided
-    # B-fragment loads cap it at ~190 TF — measured SLOWER than recomputing
-    # through hipBLASLt batched GEMMs (~1 PF): probe_fmha_perf.log r2 shows
-    # raw fused bwd 2.48 ms vs composed bwd ~1.3 ms at the BERT shape. So
-    # the default GPU backward is the batched-GEMM recompute below;
-    # use_fused_backward=True re-enables the MFMA kernels (for future
-    # LDS-staged revisions).
-    use_fused_backward = False
+    # Backward route. After the round-2 LDS-staging rework (wave-cooperative
+    # tile staging, vectorized HBM + conflict-free LDS B-fragments) the MFMA
+    # backward beats both alternatives at every probed shape
+    # (profiles/probe_fmha_perf4.log: 1.21 ms vs composed ~1.30 ms and
+    # GEMM-recompute ~1.90 ms at B128 H12 S512 D64). False selects the
+    # hipBLASLt batched-GEMM recompute (5 GEMMs + 3 fused single-pass
+    # kernels) — the debug/fallback route.
+    use_fused_backward = True
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):

@@ -109,11 +109,12 @@ def main():
     use_syncbn = False
     if args.model == "resnet50":
         model = resnet50(num_classes=1000)
-        use_syncbn = has_ext("syncbn")
+        use_syncbn = has_ext("syncbn") and not os.environ.get("APEX_BENCH_NO_SYNCBN")
         if use_syncbn:
             model = convert_syncbn_model(model)
         model = model.to(device)
-        if not use_cpu:
+        nhwc = not use_cpu and not os.environ.get("APEX_BENCH_NCHW")
+        if nhwc:
             # NHWC end to end: MIOpen bf16 convs and the welford_*_c_last
             # SyncBN kernels both prefer channels_last
             model = model.to(memory_format=torch.channels_last)
@@ -121,7 +122,7 @@ def main():
         model, opt = amp.initialize(model, opt, opt_level="O1",
                                     cast_model_type=torch.bfloat16, loss_scale=1.0, verbosity=0)
         x = torch.randn(args.batch, 3, args.image_size, args.image_size, device=device)
-        if not use_cpu:
+        if nhwc:
             x = x.contiguous(memory_format=torch.channels_last)
         y = torch.randint(0, 1000, (args.batch,), device=device)
         criterion = torch.nn.CrossEntropyLoss()

@@ -309,11 +309,14 @@ std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tens
   auto b32 = affine ? bias->to(at::kFloat).contiguous() : at::Tensor();
   auto stream = current_stream();
 
-  // one-pass when the (HW x cpg) slab stays L2-resident per block and there
-  // are enough (n, g) workgroups to occupy the 256 CUs
+  // MEASURED (profiles/probe_group_norm2.log): the one-pass form is slower
+  // than two-pass at nearly every diffusion shape on this pool (0.4-1.1x) —
+  // per-element div/mod addressing and 2-byte loads dominate before the
+  // saved HBM pass can pay. Auto therefore keeps two-pass; passes=1 opts in
+  // explicitly (the reference's knob surface is preserved).
   const long slab_bytes = HW * cpg * (long)xc.element_size();
   const bool eligible = slab_bytes <= GN_ONEPASS_SLAB_BYTES;
-  const bool one_pass = eligible && (passes == 1 || (passes == 0 && N * G >= 256));
+  const bool one_pass = eligible && passes == 1;
 
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "group_norm_nhwc_fwd", ([&] {
     if (one_pass) {

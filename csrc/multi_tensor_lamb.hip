@@ -170,3 +170,62 @@ void multi_tensor_lamb_mp_cuda(long chunk_size, at::Tensor noop_flag, TensorList
     }()));
   }()));
 }
+
+// ---- sharded (ZeRO) entry points for DistributedFusedLAMB ----
+// Reference analogue: the legacy two-phase csrc/multi_tensor_lamb_stage_1.cu
+// / stage_2.cu bindings the MLPerf-BERT DistributedFusedLAMB pipeline uses
+// (distributed_fused_lamb.py:1105-1133) — here the phases are split so the
+// caller can all-reduce the per-SEGMENT param/update norms across ranks
+// between them (each rank holds only a shard of every tensor).
+
+// stage 1 only: update written into g. global_grad_norm is a DEVICE scalar
+// (no host sync on the norm path).
+void multi_tensor_lamb_stage1_cuda(long chunk_size, at::Tensor noop_flag,
+                                   TensorLists tensor_lists, double beta1, double beta2,
+                                   double eps, long step, long bias_correction,
+                                   double weight_decay, long grad_averaging, long mode,
+                                   at::Tensor global_grad_norm, double max_grad_norm) {
+  float bc1_recip = 1.f, bc2_recip = 1.f;
+  if (bias_correction == 1) {
+    bc1_recip = (float)(1.0 / (1.0 - std::pow(beta1, (double)step)));
+    bc2_recip = (float)(1.0 / (1.0 - std::pow(beta2, (double)step)));
+  }
+  const float beta3 = grad_averaging ? (float)(1.0 - beta1) : 1.0f;
+  auto gnorm = global_grad_norm.to(at::kFloat);
+  const auto g_t = tensor_lists[0][0].scalar_type();
+  const auto p_t = tensor_lists[1][0].scalar_type();
+  APEX_DISPATCH_FLOAT_HALF_BF(p_t, "multi_tensor_lamb_stage1", ([&] {
+    using param_scalar = scalar_t;
+    APEX_DISPATCH_FLOAT_HALF_BF(g_t, "multi_tensor_lamb_stage1", ([&] {
+      using grad_scalar = scalar_t;
+      multi_tensor_apply<4>(chunk_size, noop_flag, tensor_lists,
+                            LambStage1<param_scalar, grad_scalar, false>(), (float)beta1,
+                            (float)beta2, beta3, bc1_recip, bc2_recip, (float)eps, (int)mode,
+                            (float)weight_decay, gnorm.data_ptr<float>(), (float)max_grad_norm,
+                            (const float*)nullptr, (const float*)nullptr);
+    }()));
+  }()));
+}
+
+// stage 2 only: p -= trust_ratio * update with caller-supplied per-listed-
+// tensor norms (DEVICE fp32 [n_tensors], already reduced across ranks).
+void multi_tensor_lamb_stage2_cuda(long chunk_size, at::Tensor noop_flag,
+                                   TensorLists tensor_lists, at::Tensor param_norms,
+                                   at::Tensor update_norms, double lr, double weight_decay,
+                                   bool use_nvlamb) {
+  auto pn = param_norms.to(at::kFloat).contiguous();
+  auto un = update_norms.to(at::kFloat).contiguous();
+  const auto p_t = tensor_lists[0][0].scalar_type();
+  const auto g_t = tensor_lists[1][0].scalar_type();
+  APEX_DISPATCH_FLOAT_HALF_BF(p_t, "multi_tensor_lamb_stage2", ([&] {
+    using param_scalar = scalar_t;
+    APEX_DISPATCH_FLOAT_HALF_BF(g_t, "multi_tensor_lamb_stage2", ([&] {
+      using grad_scalar = scalar_t;
+      multi_tensor_apply<2>(chunk_size, noop_flag, tensor_lists,
+                            LambStage2<param_scalar, grad_scalar, false>(),
+                            pn.data_ptr<float>(), un.data_ptr<float>(), (const float*)nullptr,
+                            (float)lr, (float)weight_decay, (int)use_nvlamb,
+                            (const float*)nullptr);
+    }()));
+  }()));
+}
