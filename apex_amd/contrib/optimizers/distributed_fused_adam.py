@@ -348,9 +348,11 @@ class DistributedFusedAdam(torch.optim.Optimizer):
 
     # ---------- grad sync ----------
     def _start_bucket_grad_sync(self, b):
+        self._on_bucket_pre_reduce(b)
         if self.world_size == 1 and self.redundant_process_group is None:
             b.grad_shard.copy_(b.grad_data.float())
             b.synced = True
+            self._on_bucket_grad_synced(b)
             return
         if self.world_size == 1:  # pure replication: reduce across replicas only
             if self.average_grad_sync:
@@ -358,6 +360,7 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             dist.all_reduce(b.grad_data, group=self.redundant_process_group)
             b.grad_shard.copy_(b.grad_data.float())
             b.synced = True
+            self._on_bucket_grad_synced(b)
             return
         if self._use_stream:
             self._comm_stream.wait_stream(torch.cuda.current_stream())
@@ -394,6 +397,17 @@ class DistributedFusedAdam(torch.optim.Optimizer):
             b.grad_shard.copy_(b._sync_shard.float())
             b.sync_work = None
         b.synced = True
+        self._on_bucket_grad_synced(b)
+
+    def _on_bucket_grad_synced(self, b):
+        """Subclass hook: runs right after a bucket's reduced grads land in
+        ``grad_shard`` (DistributedFusedLAMB accumulates its grad-norm
+        partials here, overlapped with the remaining backward)."""
+
+    def _on_bucket_pre_reduce(self, b):
+        """Subclass hook: runs just before a bucket's grad collective is
+        issued, with the LOCAL accumulated grads still intact in grad_data
+        (DistributedFusedLAMB's clip_after_ar=False norm taps here)."""
 
     def grad_sync(self):
         """Finish all outstanding gradient reductions."""
