@@ -100,9 +100,15 @@ def exhaustive_search(weight2d, m=4, n=2, max_sweeps=8):
     """Stripe-pair exhaustive permutation search.
 
     For every pair of m-wide stripes, evaluates ALL unique repartitions of
-    their 2m columns (batched topk over the 35 candidates at m=4) and keeps
-    the best; sweeps until a full pass makes no improvement. Deterministic,
-    no randomness; strictly monotone in kept magnitude.
+    their 2m columns and keeps the best; sweeps until a full pass makes no
+    improvement. Deterministic, no randomness; strictly monotone in kept
+    magnitude.
+
+    On GPU (m=4, n=2) each sweep scores EVERY stripe pair x every
+    repartition in ONE kernel launch (apex_amd._permutation_search — the
+    MI355X analogue of the reference's permutation_search_kernels.cu) and
+    greedily applies the best non-overlapping improvements; one
+    device-to-host copy per sweep instead of one sync per pair.
 
     Returns the permutation of the input channels (dim 1).
     """
@@ -112,6 +118,9 @@ def exhaustive_search(weight2d, m=4, n=2, max_sweeps=8):
     nstripes = cols // m
     parts = _stripe_pair_partitions(m).to(w.device)  # [P, 2m]
     perm = torch.arange(cols, device=w.device)
+
+    if w.is_cuda and m == 4 and n == 2 and nstripes >= 2:
+        return _exhaustive_search_gpu(w, perm, parts, m, nstripes, max_sweeps)
 
     def pair_kept(cols2m):
         # cols2m: [rows, 2m] -> kept magnitude per candidate partition [P]
@@ -131,6 +140,44 @@ def exhaustive_search(weight2d, m=4, n=2, max_sweeps=8):
                     perm[i * m:(i + 1) * m] = new[:m]
                     perm[j * m:(j + 1) * m] = new[m:]
                     improved = True
+        if not improved:
+            break
+    return perm.cpu()
+
+
+def _exhaustive_search_gpu(w, perm, parts, m, nstripes, max_sweeps):
+    from ..._ext import get_ext
+
+    ps = get_ext("permutation_search")
+    pairs = [(i, j) for i in range(nstripes - 1) for j in range(i + 1, nstripes)]
+    pair_t = torch.tensor(pairs, dtype=torch.long, device=w.device)  # [np, 2]
+    for _ in range(max_sweeps):
+        # gather the 8 permuted column ids of every pair: [np, 8]
+        stripes = perm.view(nstripes, m)
+        cols8 = torch.cat([stripes[pair_t[:, 0]], stripes[pair_t[:, 1]]], dim=1)
+        scores = ps.stripe_pair_scores(w, cols8, parts)        # [np, 35]
+        best_val, best_p = scores.max(dim=1)
+        gain = best_val - scores[:, 0]
+        order = torch.argsort(gain, descending=True)
+        # one host copy per sweep
+        order_l = order.cpu().tolist()
+        gain_l = gain.cpu().tolist()
+        bp_l = best_p.cpu().tolist()
+        used = set()
+        improved = False
+        for k in order_l:
+            if gain_l[k] <= 1e-6:
+                break
+            i, j = pairs[k]
+            if i in used or j in used:
+                continue  # stripes already rewritten this sweep
+            used.add(i)
+            used.add(j)
+            idx = cols8[k]
+            new = idx[parts[bp_l[k]]]
+            perm[i * m:(i + 1) * m] = new[:m]
+            perm[j * m:(j + 1) * m] = new[m:]
+            improved = True
         if not improved:
             break
     return perm.cpu()

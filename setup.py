@@ -58,6 +58,7 @@ ext_modules = [
     hip_ext("apex_amd._fused_dense", ["csrc/fused_dense.hip"], libraries=["hipblaslt"]),
     hip_ext("apex_amd._mlp", ["csrc/mlp.hip"], libraries=["hipblaslt"]),
     hip_ext("apex_amd._xentropy", ["csrc/xentropy.hip"]),
+    hip_ext("apex_amd._permutation_search", ["csrc/permutation_search.hip"]),
     hip_ext("apex_amd._focal_loss", ["csrc/focal_loss.hip"]),
     hip_ext("apex_amd._index_mul_2d", ["csrc/index_mul_2d.hip"]),
     hip_ext("apex_amd._group_norm", ["csrc/group_norm.hip"]),
