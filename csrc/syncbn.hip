@@ -79,12 +79,35 @@ __global__ void __launch_bounds__(BN_BLOCK) welford_nchw_kernel(
 template <typename T>
 __global__ void __launch_bounds__(BN_BLOCK) welford_nhwc_kernel(
     const T* __restrict__ x, WelfordData* __restrict__ part, long rows, long C, int S) {
-  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const int s = blockIdx.y;
-  if (c >= C) return;
   const long per = (rows + S - 1) / S;
   const long lo = (long)s * per;
   const long hi = min(lo + per, rows);
+  if (C < BN_BLOCK && (BN_BLOCK % C) == 0 && gridDim.x == 1) {
+    // small-C cooperative form (round-2 fix: one-thread-per-channel left
+    // 97% of the machine idle at ResNet's C=64 stages): R = BN_BLOCK/C row
+    // lanes share each channel — full coalescing (consecutive threads =
+    // consecutive channels within a row) and BN_BLOCK active lanes.
+    const int R = BN_BLOCK / (int)C;
+    const int c = threadIdx.x % (int)C;
+    const int k = threadIdx.x / (int)C;
+    float mean = 0.f, m2 = 0.f, count = 0.f;
+    for (long r = lo + k; r < hi; r += R)
+      welford_add(to_float(x[r * C + c]), mean, m2, count);
+    __shared__ WelfordData lanes[BN_BLOCK];
+    lanes[threadIdx.x] = {mean, m2, count};
+    __syncthreads();
+    if (k == 0) {
+      for (int kk = 1; kk < R; ++kk) {
+        const WelfordData w = lanes[kk * (int)C + c];
+        welford_combine(mean, m2, count, w.mean, w.m2, w.count);
+      }
+      part[(long)c * S + s] = {mean, m2, count};
+    }
+    return;
+  }
+  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   float mean = 0.f, m2 = 0.f, count = 0.f;
   for (long r = lo; r < hi; ++r) welford_add(to_float(x[r * C + c]), mean, m2, count);
   part[c * S + s] = {mean, m2, count};
@@ -165,12 +188,38 @@ __global__ void __launch_bounds__(BN_BLOCK) reduce_bn_partials_kernel(
       part[(c * S + s) * 2 + 1] = s2;
     }
   } else {
-    const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
     const int s = blockIdx.y;
-    if (c >= C) return;
     const long rows = N * HW;
     const long per = (rows + S - 1) / S;
     const long lo = (long)s * per, hi = min(lo + per, rows);
+    if (C < BN_BLOCK && (BN_BLOCK % C) == 0 && gridDim.x == 1) {
+      // small-C cooperative form (see welford_nhwc_kernel)
+      const int R = BN_BLOCK / (int)C;
+      const int c = threadIdx.x % (int)C;
+      const int k = threadIdx.x / (int)C;
+      const float mu = mean[c];
+      float s1 = 0.f, s2 = 0.f;
+      for (long r = lo + k; r < hi; r += R) {
+        float d = to_float(dy[r * C + c]);
+        s1 += d;
+        s2 = fmaf(d, to_float(x[r * C + c]) - mu, s2);
+      }
+      __shared__ float lanes[BN_BLOCK][2];
+      lanes[threadIdx.x][0] = s1;
+      lanes[threadIdx.x][1] = s2;
+      __syncthreads();
+      if (k == 0) {
+        for (int kk = 1; kk < R; ++kk) {
+          s1 += lanes[kk * (int)C + c][0];
+          s2 += lanes[kk * (int)C + c][1];
+        }
+        part[((long)c * S + s) * 2] = s1;
+        part[((long)c * S + s) * 2 + 1] = s2;
+      }
+      return;
+    }
+    const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
     float s1 = 0.f, s2 = 0.f;
     const float mu = mean[c];
     for (long r = lo; r < hi; ++r) {
@@ -241,6 +290,16 @@ int pick_splits(long work_per_channel, long C) {
   return (int)std::max<long>(1, std::min<long>(s, 64));
 }
 
+// NHWC stats grids: gridDim.x = ceil(C/BN_BLOCK) blocks of channels, S row
+// splits. The small-C cooperative kernels keep all BN_BLOCK lanes busy, so
+// split rows until ~2048 workgroups (8 XCDs x 256 CUs need >> 256).
+int pick_splits_nhwc(long rows, long C) {
+  const long bx = std::max<long>((C + BN_BLOCK - 1) / BN_BLOCK, 1);
+  long s = 2048 / bx;
+  s = std::min<long>(s, std::max<long>(rows / 256, 1));
+  return (int)std::max<long>(1, std::min<long>(s, 2048));
+}
+
 std::vector<at::Tensor> welford_impl(const at::Tensor& input, bool nhwc) {
   auto x = input.contiguous();
   long N, C, HW;
@@ -265,7 +324,7 @@ std::vector<at::Tensor> welford_impl(const at::Tensor& input, bool nhwc) {
     HIP_CHECK(hipGetLastError());
   } else {
     const long rows = N * HW;
-    const int S = pick_splits(rows, C);
+    const int S = pick_splits_nhwc(rows, C);
     auto part = at::empty({C, S, 3}, fopts);
     APEX_DISPATCH_FLOAT_HALF_BF(x.scalar_type(), "welford_mean_var_c_last", ([&] {
       hipLaunchKernelGGL((welford_nhwc_kernel<scalar_t>),
@@ -335,7 +394,7 @@ std::vector<at::Tensor> reduce_bn_impl(const at::Tensor& grad_out, const at::Ten
   auto gb = affine ? at::empty({C}, fopts) : at::Tensor();
   auto stream = current_stream();
 
-  const int S = pick_splits(N * HW, C);
+  const int S = nhwc ? pick_splits_nhwc(N * HW, C) : pick_splits(N * HW, C);
   auto part = at::empty({C, S, 2}, fopts);
 
   APEX_DISPATCH_FLOAT_HALF_BF(x.scalar_type(), "reduce_bn", ([&] {
