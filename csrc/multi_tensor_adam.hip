@@ -5,9 +5,13 @@
 // non-capturable path, in-kernel for capturable).
 //
 // MI355X roofline note: fp32 Adam moves 28 B/element (r g,p,m,v; w p,m,v) →
-// a 350M-param step is ~9.8 GB ≈ 1.6 ms at 6.3 TB/s achievable HBM BW. The
-// kernel is pure streaming: 8/16-byte vector accesses, fp32 math in
-// registers, one workgroup per 64K chunk.
+// a 350M-param step is ~9.8 GB. The r2 streaming sweep
+// (profiles/probe_adam_tune2.log) measured 5.49 TB/s as the best ANY
+// block/ILP config reaches for this 4-stream read + 3-stream write mix
+// (1024/4 won; grid caps hurt), so ~1.8 ms is the practical floor — the
+// measured 1.85-2.0 ms step is at ~95% of that. The kernel is pure
+// streaming: 8/16-byte vector accesses, fp32 math in registers, one
+// workgroup per 64K chunk.
 #include "amp_C.h"
 #include "multi_tensor_apply.h"
 
