@@ -368,3 +368,43 @@ def test_self_mha_flash_route_matches_composed_gpu():
         tr.flash_attention_supported = orig
     torch.testing.assert_close(y1.float(), y2.float(), rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(gx1.float(), x.grad.float(), rtol=5e-2, atol=5e-2)
+
+
+def test_flash_route_actually_engages():
+    """Counter probe: the bundled models' default GPU path must call
+    flash_attention (guards against silently falling back to composed)."""
+    import apex_amd.transformer as tr
+    from apex_amd.models.transformer import (BertModel, GPTModel,
+                                             TransformerLMConfig)
+    from apex_amd.contrib.fast_multihead_attn import SelfMultiheadAttn
+
+    calls = {"n": 0}
+    orig = tr.flash_attention
+
+    def counted(*a, **k):
+        calls["n"] += 1
+        return orig(*a, **k)
+
+    tr.flash_attention = counted
+    try:
+        cfg = TransformerLMConfig(vocab_size=128, hidden=128, layers=2, heads=2,
+                                  seq_len=64, causal=False)
+        m = BertModel(cfg).cuda().bfloat16()
+        tokens = torch.randint(0, 128, (2, 64), device="cuda")
+        m(tokens)
+        assert calls["n"] == 2, f"BERT flash route not engaged ({calls['n']})"
+
+        calls["n"] = 0
+        cfg2 = TransformerLMConfig(vocab_size=128, hidden=128, layers=2, heads=2,
+                                   seq_len=64, causal=True)
+        g = GPTModel(cfg2).cuda().bfloat16()
+        g(tokens)
+        assert calls["n"] == 2, f"GPT flash route not engaged ({calls['n']})"
+
+        calls["n"] = 0
+        mha = SelfMultiheadAttn(128, 2, dropout=0.0).cuda().bfloat16()
+        x = torch.randn(64, 2, 128, device="cuda", dtype=torch.bfloat16)
+        mha(x, attn_mask="causal")
+        assert calls["n"] == 1, f"SelfMultiheadAttn flash route not engaged ({calls['n']})"
+    finally:
+        tr.flash_attention = orig
