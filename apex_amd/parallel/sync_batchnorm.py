@@ -111,46 +111,63 @@ class SyncBatchnormFunction(torch.autograd.Function):
         use_kernels = input.is_cuda
         c = input.shape[-1] if channel_last else input.shape[1]
         count = input.numel() // c
+        track = track_running_stats and running_mean is not None
 
-        if use_kernels:
+        if use_kernels and world_size == 1:
+            # single-process fast path: welford + invstd + running-EMA in one
+            # fused call (the eager composition was ~10 launches per layer —
+            # host-launch-bound at ResNet's 53 BN layers)
             syncbn = get_ext("syncbn")
-            if channel_last:
-                mean, var_biased = syncbn.welford_mean_var_c_last(input)
-            else:
-                mean, var_biased = syncbn.welford_mean_var(input)
+            rm = running_mean if track else None
+            rv = running_var if track else None
+            mean, var_biased, inv_std = syncbn.bn_stats(
+                input, channel_last, eps, rm, rv, momentum, count)
+            total_count = count
         else:
-            x2d = _to_channels_second(input.float(), channel_last)
-            mean = x2d.mean(dim=1)
-            var_biased = x2d.var(dim=1, unbiased=False)
-
-        if world_size > 1:
-            counts = torch.full((1,), count, dtype=mean.dtype, device=mean.device)
-            combined = torch.cat([mean, var_biased, counts])
-            combined_list = [torch.empty_like(combined) for _ in range(world_size)]
-            dist.all_gather(combined_list, combined, group=process_group)
-            mean_all = torch.stack([c_[:c] for c_ in combined_list])
-            var_all = torch.stack([c_[c:2 * c] for c_ in combined_list])
-            count_all = torch.stack([c_[2 * c:] for c_ in combined_list]).view(-1)
             if use_kernels:
                 syncbn = get_ext("syncbn")
-                mean, var_biased = syncbn.welford_parallel(mean_all, var_all, count_all.to(torch.int32))
+                if channel_last:
+                    mean, var_biased = syncbn.welford_mean_var_c_last(input)
+                else:
+                    mean, var_biased = syncbn.welford_mean_var(input)
             else:
-                tot = count_all.sum()
-                w = count_all / tot
-                mean_g = (mean_all * w.unsqueeze(1)).sum(0)
-                var_g = (var_all * w.unsqueeze(1)).sum(0) + (((mean_all - mean_g) ** 2) * w.unsqueeze(1)).sum(0)
-                mean, var_biased = mean_g, var_g
-            total_count = int(count * world_size)
-        else:
-            total_count = count
+                x2d = _to_channels_second(input.float(), channel_last)
+                mean = x2d.mean(dim=1)
+                var_biased = x2d.var(dim=1, unbiased=False)
 
-        inv_std = 1.0 / torch.sqrt(var_biased + eps)
+            if world_size > 1:
+                counts = torch.full((1,), count, dtype=mean.dtype, device=mean.device)
+                combined = torch.cat([mean, var_biased, counts])
+                combined_list = [torch.empty_like(combined) for _ in range(world_size)]
+                dist.all_gather(combined_list, combined, group=process_group)
+                mean_all = torch.stack([c_[:c] for c_ in combined_list])
+                var_all = torch.stack([c_[c:2 * c] for c_ in combined_list])
+                count_all = torch.stack([c_[2 * c:] for c_ in combined_list]).view(-1)
+                total_count = int(count * world_size)
+                if use_kernels:
+                    rm = running_mean if track else None
+                    rv = running_var if track else None
+                    mean, var_biased, inv_std = syncbn.bn_stats_parallel(
+                        mean_all, var_all, count_all.to(torch.int32), eps, rm, rv,
+                        momentum, total_count)
+                else:
+                    tot = count_all.sum()
+                    w = count_all / tot
+                    mean_g = (mean_all * w.unsqueeze(1)).sum(0)
+                    var_g = (var_all * w.unsqueeze(1)).sum(0) + (((mean_all - mean_g) ** 2) * w.unsqueeze(1)).sum(0)
+                    mean, var_biased = mean_g, var_g
+            else:
+                total_count = count
 
-        if track_running_stats and running_mean is not None:
-            with torch.no_grad():
-                unbiased = var_biased * (total_count / max(total_count - 1, 1))
-                running_mean.mul_(1 - momentum).add_(mean.to(running_mean.dtype), alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased.to(running_var.dtype), alpha=momentum)
+            if use_kernels and world_size > 1:
+                pass  # fused: invstd + running stats already done in-kernel
+            else:
+                inv_std = 1.0 / torch.sqrt(var_biased + eps)
+                if track:
+                    with torch.no_grad():
+                        unbiased = var_biased * (total_count / max(total_count - 1, 1))
+                        running_mean.mul_(1 - momentum).add_(mean.to(running_mean.dtype), alpha=momentum)
+                        running_var.mul_(1 - momentum).add_(unbiased.to(running_var.dtype), alpha=momentum)
 
         ctx.process_group = process_group
         ctx.channel_last = channel_last

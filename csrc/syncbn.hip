@@ -127,6 +127,27 @@ __global__ void welford_merge_kernel(const WelfordData* __restrict__ part, float
   var_out[c] = count > 0.f ? m2 / count : 0.f;
 }
 
+// stats epilogue: invstd = rsqrt(var + eps) and (optionally) the running
+// mean/var EMA update — fused so the Python layer launches ONE kernel
+// instead of the ~6 eager ops torch would emit per BatchNorm layer (round-2
+// fix: the NHWC ResNet step was host-launch-bound, ~16 eager launches x 53
+// BN layers).
+__global__ void bn_stats_epilogue_kernel(const float* __restrict__ mean,
+                                         const float* __restrict__ var,
+                                         float* __restrict__ invstd, float eps, long C,
+                                         float* __restrict__ running_mean,
+                                         float* __restrict__ running_var, float momentum,
+                                         float unbiased_factor) {
+  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float v = var[c];
+  invstd[c] = rsqrtf(v + eps);
+  if (running_mean) {
+    running_mean[c] = running_mean[c] * (1.f - momentum) + mean[c] * momentum;
+    running_var[c] = running_var[c] * (1.f - momentum) + v * unbiased_factor * momentum;
+  }
+}
+
 // merge per-process (mean, var_biased, count) rows (the cross-GPU stat merge)
 __global__ void welford_parallel_kernel(const float* __restrict__ mean_all,
                                         const float* __restrict__ var_all,
@@ -464,6 +485,59 @@ at::Tensor bn_bwd_impl(const at::Tensor& grad_out, const at::Tensor& input,
 std::vector<at::Tensor> welford_mean_var(at::Tensor input) { return welford_impl(input, false); }
 std::vector<at::Tensor> welford_mean_var_c_last(at::Tensor input) { return welford_impl(input, true); }
 
+// fused stats: welford + invstd + running-stat EMA in-kernel.
+// Returns (mean, var_biased, invstd); running stats updated in place when
+// given (count = elements per channel for the unbiased correction).
+std::vector<at::Tensor> bn_stats(at::Tensor input, bool nhwc, double eps,
+                                 c10::optional<at::Tensor> running_mean,
+                                 c10::optional<at::Tensor> running_var, double momentum,
+                                 long total_count) {
+  auto mv = welford_impl(input, nhwc);
+  auto mean = mv[0];
+  auto var = mv[1];
+  const long C = mean.numel();
+  auto invstd = at::empty_like(mean);
+  const bool track = running_mean.has_value() && running_mean->defined();
+  TORCH_CHECK(!track || (running_mean->scalar_type() == at::ScalarType::Float &&
+                         running_var->scalar_type() == at::ScalarType::Float),
+              "bn_stats: running stats must be fp32");
+  const float unbiased = total_count > 1 ? (float)total_count / (float)(total_count - 1) : 1.f;
+  hipLaunchKernelGGL(bn_stats_epilogue_kernel, dim3((uint32_t)((C + BN_BLOCK - 1) / BN_BLOCK)),
+                     dim3(BN_BLOCK), 0, current_stream(), mean.data_ptr<float>(),
+                     var.data_ptr<float>(), invstd.data_ptr<float>(), (float)eps, C,
+                     track ? running_mean->data_ptr<float>() : nullptr,
+                     track ? running_var->data_ptr<float>() : nullptr, (float)momentum,
+                     unbiased);
+  HIP_CHECK(hipGetLastError());
+  return {mean, var, invstd};
+}
+
+std::vector<at::Tensor> welford_parallel(at::Tensor mean_all, at::Tensor var_all,
+                                         at::Tensor counts);
+
+// fused cross-process variant: welford_parallel merge + invstd + running EMA
+std::vector<at::Tensor> bn_stats_parallel(at::Tensor mean_all, at::Tensor var_all,
+                                          at::Tensor counts, double eps,
+                                          c10::optional<at::Tensor> running_mean,
+                                          c10::optional<at::Tensor> running_var,
+                                          double momentum, long total_count) {
+  auto mv = welford_parallel(mean_all, var_all, counts);
+  auto mean = mv[0];
+  auto var = mv[1];
+  const long C = mean.numel();
+  auto invstd = at::empty_like(mean);
+  const bool track = running_mean.has_value() && running_mean->defined();
+  const float unbiased = total_count > 1 ? (float)total_count / (float)(total_count - 1) : 1.f;
+  hipLaunchKernelGGL(bn_stats_epilogue_kernel, dim3((uint32_t)((C + BN_BLOCK - 1) / BN_BLOCK)),
+                     dim3(BN_BLOCK), 0, current_stream(), mean.data_ptr<float>(),
+                     var.data_ptr<float>(), invstd.data_ptr<float>(), (float)eps, C,
+                     track ? running_mean->data_ptr<float>() : nullptr,
+                     track ? running_var->data_ptr<float>() : nullptr, (float)momentum,
+                     unbiased);
+  HIP_CHECK(hipGetLastError());
+  return {mean, var, invstd};
+}
+
 std::vector<at::Tensor> welford_parallel(at::Tensor mean_all, at::Tensor var_all,
                                          at::Tensor counts) {
   TORCH_CHECK(mean_all.dim() == 2, "welford_parallel expects [world, C]");
@@ -518,6 +592,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("welford_mean_var", &welford_mean_var, "per-channel Welford mean/biased-var (NCHW)");
   m.def("welford_mean_var_c_last", &welford_mean_var_c_last, "NHWC variant");
   m.def("welford_parallel", &welford_parallel, "merge per-process (mean,var,count) rows");
+  m.def("bn_stats", &bn_stats,
+        "fused welford + invstd + running-EMA -> (mean, var_biased, invstd)");
+  m.def("bn_stats_parallel", &bn_stats_parallel,
+        "fused cross-process merge + invstd + running-EMA");
   m.def("batchnorm_forward", &batchnorm_forward, "BN apply (NCHW)");
   m.def("batchnorm_forward_c_last", &batchnorm_forward_c_last, "BN apply (NHWC, optional ReLU)");
   m.def("reduce_bn", &reduce_bn, "per-channel sum_dy/sum_dy_xmu + weight/bias grads (NCHW)");
