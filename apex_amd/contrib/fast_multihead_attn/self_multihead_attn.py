@@ -63,7 +63,7 @@ class SelfMultiheadAttn(torch.nn.Module):
                 and (attn_mask is None or attn_mask == "causal")
                 and flash_attention_supported(q, dropout=dropout_active)):
             # MFMA flash path: hardware-validated round 2, no S x S matrix
-            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+            ctx = flash_attention(q, k, v,
                                   causal=(attn_mask == "causal"), scale=self.scaling)
             ctx = ctx.permute(2, 0, 1, 3).reshape(s * b, h)
             out = fused_dense_function(ctx, self.out_proj_weight, self.out_proj_bias)

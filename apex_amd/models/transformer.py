@@ -76,8 +76,9 @@ class FusedSelfAttention(nn.Module):
         scale = 1.0 / math.sqrt(self.hd)
         if mask is None and flash_attention_supported(q):
             # MFMA flash kernel: no S x S matrix, no separate softmax pass
-            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
-                                  causal=self.causal, scale=scale)
+            # strided BSHD views pass straight through (the kernels read
+            # through (b,h,s) strides — no .contiguous() copies)
+            ctx = flash_attention(q, k, v, causal=self.causal, scale=scale)
         else:
             scores = torch.matmul(q, k.transpose(-2, -1))  # [b, nh, s, s]
             if self.causal:
@@ -254,8 +255,7 @@ class LlamaAttention(nn.Module):
         from ..transformer import flash_attention, flash_attention_supported
 
         if LlamaAttention.use_flash and flash_attention_supported(q):
-            ctx = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
-                                  causal=True, scale=1.0 / math.sqrt(self.hd))
+            ctx = flash_attention(q, k, v, causal=True, scale=1.0 / math.sqrt(self.hd))
         else:
             scores = torch.matmul(q, k.transpose(-2, -1))
             probs = scaled_upper_triang_masked_softmax(

@@ -38,7 +38,7 @@ constexpr int FM_PAD = 8;     // LDS tile row padding (shorts) -> 16B-aligned ro
 // kernels at ~190 TF — B-fragments are then read from LDS.
 template <int D>
 __device__ __forceinline__ void fm_stage_tile(short* __restrict__ dst,
-                                              const short* __restrict__ src) {
+                                              const short* __restrict__ src, long row_stride) {
   constexpr int VPR = D / 8;    // vectors per row
   constexpr int NV = FM_BN * VPR;
   const int lane = threadIdx.x & 63;
@@ -47,7 +47,7 @@ __device__ __forceinline__ void fm_stage_tile(short* __restrict__ dst,
     const int r = v / VPR;
     const int c0 = (v % VPR) * 8;
     *reinterpret_cast<bf16x8*>(dst + r * (D + FM_PAD) + c0) =
-        *reinterpret_cast<const bf16x8*>(src + (long)r * D + c0);
+        *reinterpret_cast<const bf16x8*>(src + (long)r * row_stride + c0);
   }
 }
 
@@ -66,7 +66,8 @@ __device__ __forceinline__ float row_reduce_sum16(float v) {
 template <bool CAUSAL, int D>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
-    short* __restrict__ O, float* __restrict__ LSE, int S, float scale) {
+    short* __restrict__ O, float* __restrict__ LSE, int S, float scale, int H,
+    long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs) {
   constexpr int NK = D / 32;  // K chunks for Q@K^T
   constexpr int ND = D / 16;  // 16-col output groups for P@V
   const int lane = threadIdx.x & 63;
@@ -75,9 +76,10 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
   if (q0 >= S) return;
 
-  const short* q_ptr = Q + (long)bh * S * D;
-  const short* k_ptr = K + (long)bh * S * D;
-  const short* v_ptr = V + (long)bh * S * D;
+  const long bb = bh / H, hh = bh % H;
+  const short* q_ptr = Q + bb * qb + hh * qh;
+  const short* k_ptr = K + bb * kb + hh * kh;
+  const short* v_ptr = V + bb * vb + hh * vh;
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
   __shared__ short lds_v[FM_WAVES][FM_BN * (D + FM_PAD)];
@@ -89,7 +91,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int a_row = q0 + (lane & 15);
 #pragma unroll
   for (int c = 0; c < NK; ++c)
-    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
 
   f32x4 acc[ND];
 #pragma unroll
@@ -106,7 +108,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile<D>(vbuf, v_ptr + (long)kv0 * D);
+    fm_stage_tile<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
     // ---- S = scale * Q K^T for this 16 x 32 tile (two 16x16 halves) ----
     f32x4 s_half[2];
 #pragma unroll
@@ -116,7 +118,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
 #pragma unroll
       for (int c = 0; c < NK; ++c) {
         const bf16x8 bk = *reinterpret_cast<const bf16x8*>(
-            k_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+            k_ptr + (long)k_row * ks + c * 32 + (lane >> 4) * 8);
         s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[c], bk, s, 0, 0, 0);
       }
       s_half[j] = s;
@@ -217,7 +219,9 @@ template <bool CAUSAL, int D>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
-    const float* __restrict__ DELTA, short* __restrict__ dQ, int S, float scale) {
+    const float* __restrict__ DELTA, short* __restrict__ dQ, int S, float scale, int H,
+    long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
+    long ob, long oh, long os) {
   constexpr int NK = D / 32;
   constexpr int ND = D / 16;
   const int lane = threadIdx.x & 63;
@@ -225,10 +229,11 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int bh = blockIdx.y;
   const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
   if (q0 >= S) return;
-  const short* q_ptr = Q + (long)bh * S * D;
-  const short* k_ptr = K + (long)bh * S * D;
-  const short* v_ptr = V + (long)bh * S * D;
-  const short* do_ptr = dO + (long)bh * S * D;
+  const long bb = bh / H, hh = bh % H;
+  const short* q_ptr = Q + bb * qb + hh * qh;
+  const short* k_ptr = K + bb * kb + hh * kh;
+  const short* v_ptr = V + bb * vb + hh * vh;
+  const short* do_ptr = dO + bb * ob + hh * oh;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -241,8 +246,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int a_row = q0 + (lane & 15);
 #pragma unroll
   for (int c = 0; c < NK; ++c) {
-    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
-    ado[c] = *reinterpret_cast<const bf16x8*>(do_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
+    ado[c] = *reinterpret_cast<const bf16x8*>(do_ptr + (long)a_row * os + c * 32 + (lane >> 4) * 8);
   }
   f32x4 dq_acc[ND];
 #pragma unroll
@@ -252,7 +257,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile<D>(kbuf, k_ptr + (long)kv0 * D);
+    fm_stage_tile<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
     __builtin_amdgcn_s_waitcnt(0);  // staged K visible to this wave
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
@@ -266,7 +271,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
             kbuf + k_lrow * (D + FM_PAD) + c * 32 + (lane >> 4) * 8);
         s = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[c], bk, s, 0, 0, 0);
         const bf16x8 bvt = *reinterpret_cast<const bf16x8*>(
-            v_ptr + (long)k_row * D + c * 32 + (lane >> 4) * 8);
+            v_ptr + (long)k_row * vs + c * 32 + (lane >> 4) * 8);
         dp = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ado[c], bvt, dp, 0, 0, 0);
       }
 #pragma unroll
@@ -309,7 +314,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
     const float* __restrict__ DELTA, short* __restrict__ dK, short* __restrict__ dV,
-    int S, float scale) {
+    int S, float scale, int H,
+    long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
+    long ob, long oh, long os) {
   constexpr int NK = D / 32;
   constexpr int ND = D / 16;
   const int lane = threadIdx.x & 63;
@@ -317,10 +324,11 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int bh = blockIdx.y;
   const int kv0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
   if (kv0 >= S) return;
-  const short* q_ptr = Q + (long)bh * S * D;
-  const short* k_ptr = K + (long)bh * S * D;
-  const short* v_ptr = V + (long)bh * S * D;
-  const short* do_ptr = dO + (long)bh * S * D;
+  const long bb = bh / H, hh = bh % H;
+  const short* q_ptr = Q + bb * qb + hh * qh;
+  const short* k_ptr = K + bb * kb + hh * kh;
+  const short* v_ptr = V + bb * vb + hh * vh;
+  const short* do_ptr = dO + bb * ob + hh * oh;
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
@@ -337,8 +345,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int a_row = kv0 + (lane & 15);
 #pragma unroll
   for (int c = 0; c < NK; ++c) {
-    ak[c] = *reinterpret_cast<const bf16x8*>(k_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
-    av[c] = *reinterpret_cast<const bf16x8*>(v_ptr + (long)a_row * D + c * 32 + (lane >> 4) * 8);
+    ak[c] = *reinterpret_cast<const bf16x8*>(k_ptr + (long)a_row * ks + c * 32 + (lane >> 4) * 8);
+    av[c] = *reinterpret_cast<const bf16x8*>(v_ptr + (long)a_row * vs + c * 32 + (lane >> 4) * 8);
   }
   f32x4 dv_acc[ND], dk_acc[ND];
 #pragma unroll
@@ -350,8 +358,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int q_start = CAUSAL ? (kv0 / FM_BN) * FM_BN : 0;
 
   for (int q0 = q_start; q0 < S; q0 += FM_BN) {
-    fm_stage_tile<D>(qbuf, q_ptr + (long)q0 * D);
-    fm_stage_tile<D>(dobuf, do_ptr + (long)q0 * D);
+    fm_stage_tile<D>(qbuf, q_ptr + (long)q0 * qs, qs);
+    fm_stage_tile<D>(dobuf, do_ptr + (long)q0 * os, os);
     __builtin_amdgcn_s_waitcnt(0);  // staged tiles visible to this wave
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
@@ -419,21 +427,25 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
 // these three single-pass kernels — one HBM pass each instead of the eager
 // fp32 mul/exp/mask/cast chains that cost ~6 extra S x S passes.
 
-// delta[row] = sum_d dout[row,d] * out[row,d]; one wave per row.
+// delta[row] = sum_d dout[row,d] * out[row,d]; one wave per row. dO may be
+// a strided BSHD view (ob/oh/os strides); O is packed.
 template <int D>
 __global__ void __launch_bounds__(256) fmha_delta_kernel(
     const short* __restrict__ dO, const short* __restrict__ O, float* __restrict__ delta,
-    long rows) {
+    long rows, int H, int S, long ob, long oh, long os) {
   const int lane = threadIdx.x & 63;
   const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   if (row >= rows) return;
   constexpr int PER = D / 64;  // elements per lane (1 for D=64, 2 for D=128)
+  const long bb = row / ((long)H * S);
+  const long rem = row % ((long)H * S);
+  const long dob = bb * ob + (rem / S) * oh + (rem % S) * os;
   float acc = 0.f;
 #pragma unroll
   for (int j = 0; j < PER; ++j) {
-    const long i = row * D + j * 64 + lane;
-    const __hip_bfloat16 a = *reinterpret_cast<const __hip_bfloat16*>(dO + i);
-    const __hip_bfloat16 b = *reinterpret_cast<const __hip_bfloat16*>(O + i);
+    const __hip_bfloat16 a = *reinterpret_cast<const __hip_bfloat16*>(dO + dob + j * 64 + lane);
+    const __hip_bfloat16 b =
+        *reinterpret_cast<const __hip_bfloat16*>(O + row * D + j * 64 + lane);
     acc = fmaf(__bfloat162float(a), __bfloat162float(b), acc);
   }
 #pragma unroll
@@ -498,11 +510,18 @@ __global__ void __launch_bounds__(256) fmha_ds_kernel(
 
 }  // namespace
 
-// delta = rowsum(dout * out) in fp32, shape [B, H, S]
+// delta = rowsum(dout * out) in fp32, shape [B, H, S]; dout may be a
+// strided [B,H,S,D] view (D contiguous)
 at::Tensor fmha_delta(at::Tensor dout, at::Tensor out) {
-  auto doc = dout.contiguous(), oc = out.contiguous();
+  auto doc = (dout.dim() == 4 && dout.stride(3) == 1) ? dout : dout.contiguous();
+  auto oc = out.contiguous();
   const long D = doc.size(-1);
   const long rows = doc.numel() / D;
+  const int H = doc.dim() == 4 ? (int)doc.size(1) : 1;
+  const int S = doc.dim() == 4 ? (int)doc.size(2) : (int)doc.size(-2);
+  const long ob = doc.dim() == 4 ? doc.stride(0) : rows ? doc.numel() : 0;
+  const long oh = doc.dim() == 4 ? doc.stride(1) : 0;
+  const long os = doc.dim() == 4 ? doc.stride(2) : D;
   TORCH_CHECK(D == 64 || D == 128, "fmha_delta: head_dim 64/128");
   auto delta = at::empty(doc.sizes().slice(0, doc.dim() - 1),
                          doc.options().dtype(at::kFloat));
@@ -511,11 +530,11 @@ at::Tensor fmha_delta(at::Tensor dout, at::Tensor out) {
   if (D == 64)
     hipLaunchKernelGGL((fmha_delta_kernel<64>), dim3((uint32_t)grid), dim3(256), 0, stream,
                        (const short*)doc.data_ptr(), (const short*)oc.data_ptr(),
-                       delta.data_ptr<float>(), rows);
+                       delta.data_ptr<float>(), rows, H, S, ob, oh, os);
   else
     hipLaunchKernelGGL((fmha_delta_kernel<128>), dim3((uint32_t)grid), dim3(256), 0, stream,
                        (const short*)doc.data_ptr(), (const short*)oc.data_ptr(),
-                       delta.data_ptr<float>(), rows);
+                       delta.data_ptr<float>(), rows, H, S, ob, oh, os);
   HIP_CHECK(hipGetLastError());
   return delta;
 }
@@ -558,17 +577,31 @@ at::Tensor fmha_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale) 
   return ds;
 }
 
+namespace {
+// strided [B,H,S,D] access: last dim must be contiguous and 16B-alignable;
+// the kernels read rows through (b_stride, h_stride, s_stride), so BSHD
+// views (e.g. q/k/v slices of a packed QKV projection) pass WITHOUT the
+// .contiguous() copies that cost the round-1 BERT step ~5% (432
+// direct_copy calls per trace).
+inline bool fm_strided_ok(const at::Tensor& t) {
+  return t.stride(3) == 1 && (t.stride(2) % 8) == 0 && (t.stride(1) % 8) == 0 &&
+         (t.stride(0) % 8) == 0;
+}
+}  // namespace
+
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
                                  double scale) {
   TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_fwd: bf16 only");
   TORCH_CHECK(q.dim() == 4, "fmha_fwd: [B, H, S, D]");
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  auto qc = fm_strided_ok(q) ? q : q.contiguous();
+  auto kc = fm_strided_ok(k) ? k : k.contiguous();
+  auto vc = fm_strided_ok(v) ? v : v.contiguous();
   const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
   TORCH_CHECK(kc.sizes() == qc.sizes() && vc.sizes() == qc.sizes(),
               "fmha_fwd: q/k/v shapes must match (no MQA yet)");
   TORCH_CHECK(D == 64 || D == 128, "fmha_fwd: head_dim must be 64 or 128");
   TORCH_CHECK(S % 32 == 0, "fmha_fwd: seq_len must be a multiple of 32");
-  auto out = at::empty_like(qc);
+  auto out = at::empty({B, H, S, D}, qc.options());
   auto lse = at::empty({B, H, S}, qc.options().dtype(at::kFloat));
   auto stream = current_stream();
   dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
@@ -579,7 +612,10 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
   hipLaunchKernelGGL((fmha_fwd_kernel<CAUSAL, DD>), grid, block, 0, stream,                \
                      (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),             \
                      (const short*)vc.data_ptr(), (short*)out.data_ptr(),                  \
-                     lse.data_ptr<float>(), S, sc)
+                     lse.data_ptr<float>(), S, sc, H,                                      \
+                     qc.stride(0), qc.stride(1), qc.stride(2),                             \
+                     kc.stride(0), kc.stride(1), kc.stride(2),                             \
+                     vc.stride(0), vc.stride(1), vc.stride(2))
   if (causal) {
     if (D == 64) FMHA_LAUNCH(true, 64);
     else FMHA_LAUNCH(true, 128);
@@ -595,15 +631,19 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
 std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor out, at::Tensor lse, bool causal, double scale) {
   TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_bwd: bf16 only");
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
-  auto doc = dout.contiguous(), oc = out.contiguous(), lsec = lse.contiguous();
+  auto qc = fm_strided_ok(q) ? q : q.contiguous();
+  auto kc = fm_strided_ok(k) ? k : k.contiguous();
+  auto vc = fm_strided_ok(v) ? v : v.contiguous();
+  auto doc = fm_strided_ok(dout) ? dout : dout.contiguous();
+  auto oc = out.contiguous(), lsec = lse.contiguous();
   const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
   TORCH_CHECK(D == 64 || D == 128, "fmha_bwd: head_dim must be 64 or 128");
   TORCH_CHECK(S % 32 == 0, "fmha_bwd: seq_len must be a multiple of 32");
-  auto dq = at::empty_like(qc);
-  auto dk = at::empty_like(kc);
-  auto dv = at::empty_like(vc);
-  // delta = rowsum(dO * O) in fp32 — one fused pass (fmha_delta kernel)
+  // grad outputs are packed (autograd restrides them back through the view)
+  auto dq = at::empty({B, H, S, D}, qc.options());
+  auto dk = at::empty({B, H, S, D}, kc.options());
+  auto dv = at::empty({B, H, S, D}, vc.options());
+  // delta = rowsum(dO * O) in fp32 — one fused pass (stride-aware)
   auto delta = fmha_delta(doc, oc);
   auto stream = current_stream();
   dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
@@ -616,12 +656,20 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
-                       (short*)dq.data_ptr(), S, sc);                                      \
+                       (short*)dq.data_ptr(), S, sc, H,                                    \
+                       qc.stride(0), qc.stride(1), qc.stride(2),                           \
+                       kc.stride(0), kc.stride(1), kc.stride(2),                           \
+                       vc.stride(0), vc.stride(1), vc.stride(2),                           \
+                       doc.stride(0), doc.stride(1), doc.stride(2));                       \
     hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD>), grid, block, 0, stream,          \
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
-                       (short*)dk.data_ptr(), (short*)dv.data_ptr(), S, sc);               \
+                       (short*)dk.data_ptr(), (short*)dv.data_ptr(), S, sc, H,             \
+                       qc.stride(0), qc.stride(1), qc.stride(2),                           \
+                       kc.stride(0), kc.stride(1), kc.stride(2),                           \
+                       vc.stride(0), vc.stride(1), vc.stride(2),                           \
+                       doc.stride(0), doc.stride(1), doc.stride(2));                       \
   } while (0)
   if (causal) {
     if (D == 64) FMHA_BWD_LAUNCH(true, 64);
