@@ -113,16 +113,25 @@ class SyncBatchnormFunction(torch.autograd.Function):
         count = input.numel() // c
         track = track_running_stats and running_mean is not None
 
+        # the in-kernel EMA writes fp32 running stats; fall back to the eager
+        # update for exotic dtypes (e.g. a model cast wholesale to bf16)
+        fused_track = track and running_mean.dtype == torch.float32
+
         if use_kernels and world_size == 1:
             # single-process fast path: welford + invstd + running-EMA in one
             # fused call (the eager composition was ~10 launches per layer —
             # host-launch-bound at ResNet's 53 BN layers)
             syncbn = get_ext("syncbn")
-            rm = running_mean if track else None
-            rv = running_var if track else None
+            rm = running_mean if fused_track else None
+            rv = running_var if fused_track else None
             mean, var_biased, inv_std = syncbn.bn_stats(
                 input, channel_last, eps, rm, rv, momentum, count)
             total_count = count
+            if track and not fused_track:
+                with torch.no_grad():
+                    unbiased = var_biased * (total_count / max(total_count - 1, 1))
+                    running_mean.mul_(1 - momentum).add_(mean.to(running_mean.dtype), alpha=momentum)
+                    running_var.mul_(1 - momentum).add_(unbiased.to(running_var.dtype), alpha=momentum)
         else:
             if use_kernels:
                 syncbn = get_ext("syncbn")
@@ -145,11 +154,18 @@ class SyncBatchnormFunction(torch.autograd.Function):
                 count_all = torch.stack([c_[2 * c:] for c_ in combined_list]).view(-1)
                 total_count = int(count * world_size)
                 if use_kernels:
-                    rm = running_mean if track else None
-                    rv = running_var if track else None
+                    rm = running_mean if fused_track else None
+                    rv = running_var if fused_track else None
                     mean, var_biased, inv_std = syncbn.bn_stats_parallel(
                         mean_all, var_all, count_all.to(torch.int32), eps, rm, rv,
                         momentum, total_count)
+                    if track and not fused_track:
+                        with torch.no_grad():
+                            unbiased = var_biased * (total_count / max(total_count - 1, 1))
+                            running_mean.mul_(1 - momentum).add_(
+                                mean.to(running_mean.dtype), alpha=momentum)
+                            running_var.mul_(1 - momentum).add_(
+                                unbiased.to(running_var.dtype), alpha=momentum)
                 else:
                     tot = count_all.sum()
                     w = count_all / tot
