@@ -6,10 +6,10 @@ import sys
 from collections import defaultdict
 
 
-def main(d, prefix):
+def main(d, prefix, marker="AdamFunctor"):
     kt = list(csv.DictReader(open(f"{d}/{prefix}_kernel_trace.csv")))
     evs = sorted((int(r["Start_Timestamp"]), int(r["End_Timestamp"]), r["Kernel_Name"]) for r in kt)
-    adam = [e for s, e, n in evs if "AdamFunctor" in n]
+    adam = [e for s, e, n in evs if marker in n]
     # optimizer launches cluster at step ends; keep the last of each cluster
     step_ends = []
     for t in adam:
@@ -71,6 +71,28 @@ def main(d, prefix):
     for fn, t in sorted(tot.items(), key=lambda kv: -kv[1])[:12]:
         print(f"  {t:8.2f} ms x{cc[fn]:6} {fn}")
 
+    # per-kernel-family GPU time inside the steady-state window
+    fam = defaultdict(float)
+    fc = defaultdict(int)
+    for s, e, n in window:
+        key = ("naive_conv" if "naive_conv" in n else
+               "ck_conv" if "ck::" in n else
+               "Tensile" if "Cijk" in n else
+               "miopen" if ("miopen" in n.lower() or "Winograd" in n or "Conv" in n) else
+               "welford/bn" if ("welford" in n or "bn_" in n or "reduce_bn" in n
+                                or "batchnorm" in n.lower()) else
+               "sgd/adam/mta" if ("multi_tensor" in n or "Functor" in n) else
+               "pool" if "pool" in n.lower() else
+               "elementwise/copy" if ("elementwise" in n or "copy" in n or "Fill" in n) else
+               "fmha" if "fmha" in n else
+               "softmax/ln" if ("softmax" in n or "ln_" in n or "xentropy" in n) else
+               "other")
+        fam[key] += (e - s) / 1e6
+        fc[key] += 1
+    print("\nkernel families in window (GPU ms):")
+    for k, t in sorted(fam.items(), key=lambda kv: -kv[1]):
+        print(f"  {t:8.2f} ms x{fc[k]:6} {k}")
+
 
 if __name__ == "__main__":
-    main(sys.argv[1], sys.argv[2])
+    main(sys.argv[1], sys.argv[2], sys.argv[3] if len(sys.argv) > 3 else "AdamFunctor")
