@@ -203,6 +203,28 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph whole-step capture (resnet/O1-static path): the traced step is
+    # launch-bound — ~1050 dispatches, 57% GPU-busy (gpurun_out/
+    # trace_rnncw.txt) — and with static loss scale there is NO host sync in
+    # the step (no overflow .item()), SyncBN stats are one fused device
+    # launch, and inputs are static buffers. Replay removes the host launch
+    # overhead entirely. Fallback to eager on any capture failure. The O2
+    # transformer paths keep eager (their unscale reads the overflow flag).
+    graphed = False
+    if (args.model == "resnet50" and not use_cpu and not distributed
+            and os.environ.get("APEX_BENCH_GRAPH", "1") != "0"):
+        try:
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                step()
+            g.replay()  # capture records without executing; apply one step
+            torch.cuda.synchronize()
+            step = g.replay  # noqa: F811
+            graphed = True
+        except Exception as e:
+            print(f"# hipGraph capture unavailable, eager path: {e}", flush=True)
+
     if distributed:
         dist.barrier()
     if not use_cpu:
@@ -250,6 +272,7 @@ def main():
                               "gpt2": "FusedLAMB", "llama": "FusedAdam",
                               "transformer_lg": "FusedAdam"}[args.model],
                 "syncbn": use_syncbn,
+                "hipgraph_step": graphed if args.model == "resnet50" else None,
                 "parallelism": f"dp{world}",
             },
         }
