@@ -46,13 +46,15 @@ __all__ = ["initialize", "scale_loss", "master_params", "state_dict", "load_stat
 
 class OptProperties:
     def __init__(self, opt_level, cast_model_type, patch_torch_functions,
-                 keep_batchnorm_fp32, master_weights, loss_scale):
+                 keep_batchnorm_fp32, master_weights, loss_scale,
+                 overflow_check=True):
         self.opt_level = opt_level
         self.cast_model_type = cast_model_type
         self.patch_torch_functions = patch_torch_functions
         self.keep_batchnorm_fp32 = keep_batchnorm_fp32
         self.master_weights = master_weights
         self.loss_scale = loss_scale
+        self.overflow_check = overflow_check
 
 
 _OPT_LEVELS = {
@@ -270,9 +272,17 @@ def initialize(
     verbosity=1,
     min_loss_scale=None,
     max_loss_scale=2.0 ** 24,
+    overflow_check=True,
 ):
     """Initialize amp. Returns (models, optimizers) with the same
-    list-or-single structure the caller passed (reference behavior)."""
+    list-or-single structure the caller passed (reference behavior).
+
+    ``overflow_check=False`` (static loss scale only, apex_amd extension for
+    bf16 + hipGraph capture): skip the per-iteration device-to-host overflow
+    read — scale_loss then performs NO host sync, so a whole training step
+    (fwd + bwd + FusedAdam(capturable=True).step) records into a hipGraph.
+    bf16 with scale 1.0 has no practical overflow path; steps are never
+    skipped in this mode."""
     _amp_state.verbosity = verbosity
 
     models_was_list = isinstance(models, list)
@@ -309,8 +319,11 @@ def initialize(
     if loss_scale is not None:
         props["loss_scale"] = loss_scale if loss_scale == "dynamic" else float(loss_scale)
 
+    if not overflow_check and props["loss_scale"] == "dynamic":
+        raise ValueError("overflow_check=False requires a static loss_scale")
     opt_properties = OptProperties(opt_level, props["cast_model_type"], props["patch_torch_functions"],
-                                   props["keep_batchnorm_fp32"], props["master_weights"], props["loss_scale"])
+                                   props["keep_batchnorm_fp32"], props["master_weights"],
+                                   props["loss_scale"], overflow_check)
 
     maybe_print(f"apex_amd.amp: opt_level={opt_level}, cast_model_type={props['cast_model_type']}, "
                 f"keep_batchnorm_fp32={props['keep_batchnorm_fp32']}, master_weights={props['master_weights']}, "
@@ -458,14 +471,15 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
         yield loss
         if delay_unscale:
             return
+        check = _amp_state.opt_properties.overflow_check
         # still materialize masters for O2 with static scale 1.0
         if _amp_state.opt_properties.master_weights:
             for opt in opt_list:
                 model_grads, master_grads = _materialize_master_grads(opt, 1.0)
-                ov = scaler.unscale_grads(model_grads, master_grads)
+                ov = scaler.unscale_grads(model_grads, master_grads, check=check)
                 for p in opt._amp_stash.all_fp16_params:
                     p.grad = None  # consumed into masters (see dynamic path)
-                if ov:
+                if check and ov:
                     opt._amp_skip_next_step = True
         return
 

@@ -109,13 +109,15 @@ class LossScaler:
                 self.update_scale()
         return self._has_overflow
 
-    def unscale_grads(self, grads_in, grads_out, scale_override=None):
+    def unscale_grads(self, grads_in, grads_out, scale_override=None, check=True):
         """out = in * (1/scale), with isfinite check setting the overflow flag.
 
         Standalone call: returns True if an overflow was detected
         (host-synchronizing on GPU) and ticks the dynamic scale iff
         ``scale_override`` is None. Inside a begin_unscale()/finish_unscale()
         bracket: only accumulates the overflow flag; no tick, no host sync.
+        ``check=False`` (static scale): skip the device-to-host overflow read
+        entirely — the launch stays async/hipGraph-capturable.
         """
         bracketed = self._in_iteration
         update_state = (scale_override is None) and not bracketed
@@ -138,6 +140,9 @@ class LossScaler:
             )
             if bracketed:
                 return False  # decision deferred to finish_unscale
+            if not check:
+                self._has_overflow = False
+                return False
             if self.dynamic and update_state:
                 # on-device scale update; host reads only the skip decision
                 amp_C.update_scale_hysteresis(

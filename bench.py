@@ -149,29 +149,41 @@ def main():
         from apex_amd.optimizers import FusedAdam, FusedLAMB
         from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
 
+        # FusedAdam models run hipGraph-capturable (device lr/step: bias
+        # corrections stay correct across graph replays); FusedLAMB (gpt2)
+        # keeps the eager step (its host-side step count would freeze)
+        capturable = (args.model in ("bert", "llama", "transformer_lg")
+                      and not use_cpu and not distributed
+                      and os.environ.get("APEX_BENCH_GRAPH", "1") != "0")
         if args.model == "bert":
             cfg = bert_base_config(seq_len=args.seq_len)
             model = BertModel(cfg).to(device)
-            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01,
+                            capturable=capturable)
             config_model = "bert-base"
         elif args.model == "llama":
             cfg = llama_small_config(seq_len=args.seq_len)
             model = LlamaModel(cfg).to(device)
-            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01,
+                            capturable=capturable)
             config_model = "llama-small(rope+swiglu+rmsnorm)"
         elif args.model == "transformer_lg":
             cfg = transformer_large_config(seq_len=args.seq_len)
             model = TransformerLargeModel(cfg).to(device)
-            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01)
+            opt = FusedAdam(model.parameters(), lr=1e-4, weight_decay=0.01,
+                            capturable=capturable)
             config_model = "transformer-large(fast_multihead_attn)"
         else:
             cfg = gpt2_345m_config(seq_len=min(args.seq_len, 1024))
             model = GPTModel(cfg).to(device)
             opt = FusedLAMB(model.parameters(), lr=1e-4, weight_decay=0.01)
             config_model = "gpt2-345m"
+        # overflow_check=False (static bf16 scale): scale_loss performs no
+        # host sync, so the whole O2 step is hipGraph-capturable
         model, opt = amp.initialize(model, opt, opt_level="O2",
                                     cast_model_type=torch.bfloat16, loss_scale=1.0,
-                                    keep_batchnorm_fp32=False, verbosity=0)
+                                    keep_batchnorm_fp32=False, verbosity=0,
+                                    overflow_check=not capturable)
         tokens = torch.randint(0, cfg.vocab_size, (args.batch, cfg.seq_len), device=device)
         # BERT trains unpadded/full attention here (synthetic fixed-length
         # batches): mask=None routes through the MFMA flash kernel; padded
@@ -211,7 +223,8 @@ def main():
     # overhead entirely. Fallback to eager on any capture failure. The O2
     # transformer paths keep eager (their unscale reads the overflow flag).
     graphed = False
-    if (args.model == "resnet50" and not use_cpu and not distributed
+    if (args.model in ("resnet50", "bert", "llama", "transformer_lg")
+            and not use_cpu and not distributed
             and os.environ.get("APEX_BENCH_GRAPH", "1") != "0"):
         try:
             torch.cuda.synchronize()
@@ -272,7 +285,7 @@ def main():
                               "gpt2": "FusedLAMB", "llama": "FusedAdam",
                               "transformer_lg": "FusedAdam"}[args.model],
                 "syncbn": use_syncbn,
-                "hipgraph_step": graphed if args.model == "resnet50" else None,
+                "hipgraph_step": graphed,
                 "parallelism": f"dp{world}",
             },
         }
