@@ -242,3 +242,37 @@ def test_o2_static_scale1_clears_model_grads_and_skips_on_overflow():
     opt.step()
     for b, a in zip(before, amp.master_params(opt)):
         torch.testing.assert_close(b, a)
+
+
+def test_overflow_check_off_static_scale():
+    """overflow_check=False (static scale): no step skipping, masters still
+    maintained; dynamic scale + overflow_check=False is rejected."""
+    import pytest as _pytest
+    import torch
+    from apex_amd import amp
+    from apex_amd.amp._amp_state import _amp_state
+
+    _amp_state.reset()
+    model = torch.nn.Linear(8, 8)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    with _pytest.raises(ValueError):
+        amp.initialize(model, opt, opt_level="O2", loss_scale="dynamic",
+                       overflow_check=False, verbosity=0)
+
+    _amp_state.reset()
+    model = torch.nn.Linear(8, 8)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    model, opt = amp.initialize(model, opt, opt_level="O2", loss_scale=1.0,
+                                cast_model_type=torch.bfloat16,
+                                keep_batchnorm_fp32=False, verbosity=0,
+                                overflow_check=False)
+    x = torch.randn(4, 8)
+    loss = model(x).pow(2).mean()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    # grads landed in the fp32 masters, no skip flag set
+    stash = opt._amp_stash
+    assert any(p.grad is not None and p.grad.abs().sum() > 0
+               for p in stash.all_fp32_from_fp16_params)
+    assert not getattr(opt, "_amp_skip_next_step", False)
+    opt.step()
