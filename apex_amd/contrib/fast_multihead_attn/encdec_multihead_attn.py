@@ -55,6 +55,21 @@ class EncdecMultiheadAttn(torch.nn.Module):
         k = kv[:, :, 0].permute(1, 2, 0, 3)
         v = kv[:, :, 1].permute(1, 2, 0, 3)
 
+        from ...transformer import flash_attention, flash_attention_supported
+
+        dropout_active = self.dropout if (is_training and self.training) else 0.0
+        if (not need_weights and key_padding_mask is None and attn_mask is None
+                and flash_attention_supported(q, dropout=dropout_active, k=k)):
+            # MFMA flash cross-attention (Sq != Skv supported): no sq x sk
+            # matrix, strided BSHD views pass copy-free
+            ctx = flash_attention(q, k, v, causal=False, scale=self.scaling)
+            ctx = ctx.permute(2, 0, 1, 3).reshape(sq * b, h)
+            out = fused_dense_function(ctx, self.out_proj_weight, self.out_proj_bias)
+            out = out.reshape(sq, b, h)
+            if self.include_norm_add:
+                out = out + residual
+            return out, None
+
         scores = torch.matmul(q, k.transpose(-2, -1))  # [b, nh, sq, sk]
         if attn_mask is not None:
             probs = scaled_masked_softmax(scores.contiguous(), attn_mask, self.scaling)

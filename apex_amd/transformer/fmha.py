@@ -19,14 +19,16 @@ import torch
 from .._ext import get_ext
 
 
-def flash_attention_supported(q, dropout=0.0):
-    """True when the MFMA flash kernel can take this tensor: CUDA bf16,
-    head_dim 64/128, seq_len % 32 == 0, no attention dropout."""
+def flash_attention_supported(q, dropout=0.0, k=None):
+    """True when the MFMA flash kernel can take these tensors: CUDA bf16,
+    head_dim 64/128, seq lens % 32 == 0, no attention dropout. Pass ``k``
+    for cross-attention (Skv may differ from Sq)."""
     return (
         q.is_cuda
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
         and q.shape[-2] % 32 == 0
+        and (k is None or k.shape[-2] % 32 == 0)
         and dropout == 0.0
     )
 

@@ -66,7 +66,7 @@ __device__ __forceinline__ float row_reduce_sum16(float v) {
 template <bool CAUSAL, int D>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
-    short* __restrict__ O, float* __restrict__ LSE, int S, float scale, int H,
+    short* __restrict__ O, float* __restrict__ LSE, int S, int SKV, float scale, int H,
     long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs) {
   constexpr int NK = D / 32;  // K chunks for Q@K^T
   constexpr int ND = D / 16;  // 16-col output groups for P@V
@@ -105,7 +105,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
 
   // rows this lane's D-fragments correspond to (replicated over col lanes)
   const int my_r0 = (lane >> 4) * 4;  // + q
-  const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
+  const int kv_end = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
     fm_stage_tile<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
@@ -219,7 +219,7 @@ template <bool CAUSAL, int D>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
-    const float* __restrict__ DELTA, short* __restrict__ dQ, int S, float scale, int H,
+    const float* __restrict__ DELTA, short* __restrict__ dQ, int S, int SKV, float scale, int H,
     long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
     long ob, long oh, long os) {
   constexpr int NK = D / 32;
@@ -254,7 +254,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   for (int d = 0; d < ND; ++d) dq_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int my_r0 = (lane >> 4) * 4;
-  const int kv_end = CAUSAL ? min(S, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : S;
+  const int kv_end = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
     fm_stage_tile<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
@@ -314,7 +314,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
     const float* __restrict__ DELTA, short* __restrict__ dK, short* __restrict__ dV,
-    int S, float scale, int H,
+    int S, int SKV, float scale, int H,
     long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
     long ob, long oh, long os) {
   constexpr int NK = D / 32;
@@ -323,7 +323,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int wave = threadIdx.x >> 6;
   const int bh = blockIdx.y;
   const int kv0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
-  if (kv0 >= S) return;
+  if (kv0 >= SKV) return;
   const long bb = bh / H, hh = bh % H;
   const short* q_ptr = Q + bb * qb + hh * qh;
   const short* k_ptr = K + bb * kb + hh * kh;
@@ -597,10 +597,13 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
   auto kc = fm_strided_ok(k) ? k : k.contiguous();
   auto vc = fm_strided_ok(v) ? v : v.contiguous();
   const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
-  TORCH_CHECK(kc.sizes() == qc.sizes() && vc.sizes() == qc.sizes(),
-              "fmha_fwd: q/k/v shapes must match (no MQA yet)");
+  const int SKV = kc.size(2);
+  TORCH_CHECK(kc.size(0) == B && kc.size(1) == H && kc.size(3) == D &&
+                  vc.sizes() == kc.sizes(),
+              "fmha_fwd: k/v must be [B,H,Skv,D] matching q's B/H/D (no MQA yet)");
   TORCH_CHECK(D == 64 || D == 128, "fmha_fwd: head_dim must be 64 or 128");
-  TORCH_CHECK(S % 32 == 0, "fmha_fwd: seq_len must be a multiple of 32");
+  TORCH_CHECK(S % 32 == 0 && SKV % 32 == 0, "fmha_fwd: seq lens must be multiples of 32");
+  TORCH_CHECK(!causal || S == SKV, "fmha_fwd: causal needs Sq == Skv");
   auto out = at::empty({B, H, S, D}, qc.options());
   auto lse = at::empty({B, H, S}, qc.options().dtype(at::kFloat));
   auto stream = current_stream();
@@ -612,7 +615,7 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
   hipLaunchKernelGGL((fmha_fwd_kernel<CAUSAL, DD>), grid, block, 0, stream,                \
                      (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),             \
                      (const short*)vc.data_ptr(), (short*)out.data_ptr(),                  \
-                     lse.data_ptr<float>(), S, sc, H,                                      \
+                     lse.data_ptr<float>(), S, SKV, sc, H,                                 \
                      qc.stride(0), qc.stride(1), qc.stride(2),                             \
                      kc.stride(0), kc.stride(1), kc.stride(2),                             \
                      vc.stride(0), vc.stride(1), vc.stride(2))
@@ -637,16 +640,19 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
   auto doc = fm_strided_ok(dout) ? dout : dout.contiguous();
   auto oc = out.contiguous(), lsec = lse.contiguous();
   const int B = qc.size(0), H = qc.size(1), S = qc.size(2), D = qc.size(3);
+  const int SKV = kc.size(2);
   TORCH_CHECK(D == 64 || D == 128, "fmha_bwd: head_dim must be 64 or 128");
-  TORCH_CHECK(S % 32 == 0, "fmha_bwd: seq_len must be a multiple of 32");
+  TORCH_CHECK(S % 32 == 0 && SKV % 32 == 0, "fmha_bwd: seq lens must be multiples of 32");
+  TORCH_CHECK(!causal || S == SKV, "fmha_bwd: causal needs Sq == Skv");
   // grad outputs are packed (autograd restrides them back through the view)
   auto dq = at::empty({B, H, S, D}, qc.options());
-  auto dk = at::empty({B, H, S, D}, kc.options());
-  auto dv = at::empty({B, H, S, D}, vc.options());
+  auto dk = at::empty({B, H, SKV, D}, kc.options());
+  auto dv = at::empty({B, H, SKV, D}, vc.options());
   // delta = rowsum(dO * O) in fp32 — one fused pass (stride-aware)
   auto delta = fmha_delta(doc, oc);
   auto stream = current_stream();
   dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
+  dim3 grid_kv((SKV + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
   dim3 block(FM_WAVES * 64);
   const float sc = (float)scale;
 
@@ -656,16 +662,16 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
-                       (short*)dq.data_ptr(), S, sc, H,                                    \
+                       (short*)dq.data_ptr(), S, SKV, sc, H,                               \
                        qc.stride(0), qc.stride(1), qc.stride(2),                           \
                        kc.stride(0), kc.stride(1), kc.stride(2),                           \
                        vc.stride(0), vc.stride(1), vc.stride(2),                           \
                        doc.stride(0), doc.stride(1), doc.stride(2));                       \
-    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD>), grid, block, 0, stream,          \
+    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD>), grid_kv, block, 0, stream,       \
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
-                       (short*)dk.data_ptr(), (short*)dv.data_ptr(), S, sc, H,             \
+                       (short*)dk.data_ptr(), (short*)dv.data_ptr(), S, SKV, sc, H,        \
                        qc.stride(0), qc.stride(1), qc.stride(2),                           \
                        kc.stride(0), kc.stride(1), kc.stride(2),                           \
                        vc.stride(0), vc.stride(1), vc.stride(2),                           \
