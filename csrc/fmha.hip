@@ -412,11 +412,12 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     const long kv_g = kv0 + my_r0 + qi;
 #pragma unroll
     for (int d = 0; d < ND; ++d) {
+      // dK/dV are packed [B, H, SKV, D] — the head stride is SKV, not Sq
       const __hip_bfloat16 ov = __float2bfloat16(dv_acc[d][qi]);
-      dV[((long)bh * S + kv_g) * D + d * 16 + (lane & 15)] =
+      dV[((long)bh * SKV + kv_g) * D + d * 16 + (lane & 15)] =
           *reinterpret_cast<const short*>(&ov);
       const __hip_bfloat16 ok = __float2bfloat16(dk_acc[d][qi] * scale);
-      dK[((long)bh * S + kv_g) * D + d * 16 + (lane & 15)] =
+      dK[((long)bh * SKV + kv_g) * D + d * 16 + (lane & 15)] =
           *reinterpret_cast<const short*>(&ok);
     }
   }
