@@ -51,6 +51,24 @@ __device__ __forceinline__ void fm_stage_tile(short* __restrict__ dst,
   }
 }
 
+// block-cooperative variant: ONE shared tile per workgroup (all 256 threads
+// stage; callers barrier). Quarters the staging traffic and LDS footprint of
+// the per-wave form — the four waves of a workgroup consume the same tiles.
+template <int D>
+__device__ __forceinline__ void fm_stage_tile_block(short* __restrict__ dst,
+                                                    const short* __restrict__ src,
+                                                    long row_stride) {
+  constexpr int VPR = D / 8;
+  constexpr int NV = FM_BN * VPR;
+#pragma unroll
+  for (int v = threadIdx.x; v < NV; v += FM_WAVES * 64) {
+    const int r = v / VPR;
+    const int c0 = (v % VPR) * 8;
+    *reinterpret_cast<bf16x8*>(dst + r * (D + FM_PAD) + c0) =
+        *reinterpret_cast<const bf16x8*>(src + (long)r * row_stride + c0);
+  }
+}
+
 __device__ __forceinline__ float row_reduce_max16(float v) {
 #pragma unroll
   for (int m = 1; m < 16; m <<= 1) v = fmaxf(v, __shfl_xor(v, m));
@@ -74,7 +92,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int wave = threadIdx.x >> 6;
   const int bh = blockIdx.y;
   const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
-  if (q0 >= S) return;
+  // waves past Sq stay RESIDENT (they co-stage the shared V tile and hit the
+  // block barriers) but skip all math and stores
+  const bool active = q0 < S;
 
   const long bb = bh / H, hh = bh % H;
   const short* q_ptr = Q + bb * qb + hh * qh;
@@ -82,16 +102,18 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const short* v_ptr = V + bb * vb + hh * vh;
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_v[FM_WAVES][FM_BN * (D + FM_PAD)];
+  __shared__ short lds_v[FM_BN * (D + FM_PAD)];  // ONE tile per workgroup
   short* pbuf = lds_p[wave];
-  short* vbuf = lds_v[wave];
+  short* vbuf = lds_v;
 
   // Q rows for this wave, as A-fragments, resident for the whole pass
   bf16x8 aq[NK];
-  const int a_row = q0 + (lane & 15);
+  const int a_row = min(q0 + (lane & 15), S - 1);
+  if (active) {
 #pragma unroll
-  for (int c = 0; c < NK; ++c)
-    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
+    for (int c = 0; c < NK; ++c)
+      aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
+  }
 
   f32x4 acc[ND];
 #pragma unroll
@@ -105,10 +127,19 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
 
   // rows this lane's D-fragments correspond to (replicated over col lanes)
   const int my_r0 = (lane >> 4) * 4;  // + q
-  const int kv_end = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
+  // block-uniform bound (last wave's rows); a wave's extra diagonal tiles
+  // are fully masked, which leaves its online-softmax state unchanged
+  const int q0_wg = blockIdx.x * (FM_WAVES * FM_ROWS);
+  const int kv_end = CAUSAL
+      ? min(SKV, ((q0_wg + FM_WAVES * FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
+    fm_stage_tile_block<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
+    __syncthreads();  // staged tile visible block-wide
+    if (!active) {
+      __syncthreads();
+      continue;
+    }
     // ---- S = scale * Q K^T for this 16 x 32 tile (two 16x16 halves) ----
     f32x4 s_half[2];
 #pragma unroll
@@ -186,9 +217,10 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
       }
       acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc[d], 0, 0, 0);
     }
-    __builtin_amdgcn_s_waitcnt(0);  // LDS reads done before next tile's stores
+    __syncthreads();  // all waves done reading before the next tile's stores
   }
 
+  if (!active) return;
   // ---- epilogue: O = acc / l, LSE = m + log(l) ----
 #pragma unroll
   for (int q = 0; q < 4; ++q) {
@@ -228,7 +260,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int wave = threadIdx.x >> 6;
   const int bh = blockIdx.y;
   const int q0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
-  if (q0 >= S) return;
+  const bool active = q0 < S;  // inactive waves co-stage + barrier only
   const long bb = bh / H, hh = bh % H;
   const short* q_ptr = Q + bb * qb + hh * qh;
   const short* k_ptr = K + bb * kb + hh * kh;
@@ -238,27 +270,36 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const float* delta = DELTA + (long)bh * S;
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_k[FM_WAVES][FM_BN * (D + FM_PAD)];
+  __shared__ short lds_k[FM_BN * (D + FM_PAD)];  // ONE shared K tile per WG
   short* pbuf = lds_p[wave];
-  short* kbuf = lds_k[wave];
+  short* kbuf = lds_k;
 
   bf16x8 aq[NK], ado[NK];
-  const int a_row = q0 + (lane & 15);
+  const int a_row = min(q0 + (lane & 15), S - 1);
+  if (active) {
 #pragma unroll
-  for (int c = 0; c < NK; ++c) {
-    aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
-    ado[c] = *reinterpret_cast<const bf16x8*>(do_ptr + (long)a_row * os + c * 32 + (lane >> 4) * 8);
+    for (int c = 0; c < NK; ++c) {
+      aq[c] = *reinterpret_cast<const bf16x8*>(q_ptr + (long)a_row * qs + c * 32 + (lane >> 4) * 8);
+      ado[c] = *reinterpret_cast<const bf16x8*>(do_ptr + (long)a_row * os + c * 32 + (lane >> 4) * 8);
+    }
   }
   f32x4 dq_acc[ND];
 #pragma unroll
   for (int d = 0; d < ND; ++d) dq_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int my_r0 = (lane >> 4) * 4;
-  const int kv_end = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
+  // block-uniform bound: extra diagonal tiles give p = 0 -> ds = 0
+  const int q0_wg = blockIdx.x * (FM_WAVES * FM_ROWS);
+  const int kv_end = CAUSAL
+      ? min(SKV, ((q0_wg + FM_WAVES * FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
-    __builtin_amdgcn_s_waitcnt(0);  // staged K visible to this wave
+    fm_stage_tile_block<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
+    __syncthreads();  // staged K visible block-wide
+    if (!active) {
+      __syncthreads();
+      continue;
+    }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       f32x4 s = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -295,8 +336,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
         bK[jj] = kbuf[((lane >> 4) * 8 + jj) * (D + FM_PAD) + d * 16 + (lane & 15)];
       dq_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, bK, dq_acc[d], 0, 0, 0);
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();  // all waves done with kbuf before the next stage
   }
+  if (!active) return;
 #pragma unroll
   for (int qi = 0; qi < 4; ++qi) {
     const long row_g = q0 + my_r0 + qi;
@@ -323,7 +365,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int wave = threadIdx.x >> 6;
   const int bh = blockIdx.y;
   const int kv0 = blockIdx.x * (FM_WAVES * FM_ROWS) + wave * FM_ROWS;
-  if (kv0 >= SKV) return;
+  const bool active = kv0 < SKV;  // inactive waves co-stage + barrier only
   const long bb = bh / H, hh = bh % H;
   const short* q_ptr = Q + bb * qb + hh * qh;
   const short* k_ptr = K + bb * kb + hh * kh;
@@ -334,19 +376,21 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
 
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
   __shared__ short lds_ds[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_q[FM_WAVES][FM_BN * (D + FM_PAD)];
-  __shared__ short lds_do[FM_WAVES][FM_BN * (D + FM_PAD)];
+  __shared__ short lds_q[FM_BN * (D + FM_PAD)];   // ONE shared Q tile per WG
+  __shared__ short lds_do[FM_BN * (D + FM_PAD)];  // ONE shared dO tile per WG
   short* pbuf = lds_p[wave];
   short* dsbuf = lds_ds[wave];
-  short* qbuf = lds_q[wave];
-  short* dobuf = lds_do[wave];
+  short* qbuf = lds_q;
+  short* dobuf = lds_do;
 
   bf16x8 ak[NK], av[NK];
-  const int a_row = kv0 + (lane & 15);
+  const int a_row = min(kv0 + (lane & 15), SKV - 1);
+  if (active) {
 #pragma unroll
-  for (int c = 0; c < NK; ++c) {
-    ak[c] = *reinterpret_cast<const bf16x8*>(k_ptr + (long)a_row * ks + c * 32 + (lane >> 4) * 8);
-    av[c] = *reinterpret_cast<const bf16x8*>(v_ptr + (long)a_row * vs + c * 32 + (lane >> 4) * 8);
+    for (int c = 0; c < NK; ++c) {
+      ak[c] = *reinterpret_cast<const bf16x8*>(k_ptr + (long)a_row * ks + c * 32 + (lane >> 4) * 8);
+      av[c] = *reinterpret_cast<const bf16x8*>(v_ptr + (long)a_row * vs + c * 32 + (lane >> 4) * 8);
+    }
   }
   f32x4 dv_acc[ND], dk_acc[ND];
 #pragma unroll
@@ -355,12 +399,19 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     dk_acc[d] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
   const int my_r0 = (lane >> 4) * 4;
-  const int q_start = CAUSAL ? (kv0 / FM_BN) * FM_BN : 0;
+  // block-uniform start (first wave's tile); later waves' extra early tiles
+  // are causal-masked to p = 0 -> no contribution
+  const int q_start = CAUSAL
+      ? ((blockIdx.x * (FM_WAVES * FM_ROWS)) / FM_BN) * FM_BN : 0;
 
   for (int q0 = q_start; q0 < S; q0 += FM_BN) {
-    fm_stage_tile<D>(qbuf, q_ptr + (long)q0 * qs, qs);
-    fm_stage_tile<D>(dobuf, do_ptr + (long)q0 * os, os);
-    __builtin_amdgcn_s_waitcnt(0);  // staged tiles visible to this wave
+    fm_stage_tile_block<D>(qbuf, q_ptr + (long)q0 * qs, qs);
+    fm_stage_tile_block<D>(dobuf, do_ptr + (long)q0 * os, os);
+    __syncthreads();  // staged tiles visible block-wide
+    if (!active) {
+      __syncthreads();
+      continue;
+    }
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       f32x4 sT = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -405,8 +456,9 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
       dv_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_do, dv_acc[d], 0, 0, 0);
       dk_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_q, dk_acc[d], 0, 0, 0);
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();  // all waves done with q/do tiles before the next stage
   }
+  if (!active) return;
 #pragma unroll
   for (int qi = 0; qi < 4; ++qi) {
     const long kv_g = kv0 + my_r0 + qi;
