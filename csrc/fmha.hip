@@ -133,10 +133,13 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const int kv_end = CAUSAL
       ? min(SKV, ((q0_wg + FM_WAVES * FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
+  // a wave's own last useful tile (tiles past it are fully causal-masked)
+  const int kv_end_wave = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
+
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
     fm_stage_tile_block<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
     __syncthreads();  // staged tile visible block-wide
-    if (!active) {
+    if (!active || kv0 >= kv_end_wave) {
       __syncthreads();
       continue;
     }
@@ -293,10 +296,12 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const int kv_end = CAUSAL
       ? min(SKV, ((q0_wg + FM_WAVES * FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
+  const int kv_end_wave = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
+
   for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
     fm_stage_tile_block<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
     __syncthreads();  // staged K visible block-wide
-    if (!active) {
+    if (!active || kv0 >= kv_end_wave) {
       __syncthreads();
       continue;
     }
@@ -408,7 +413,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     fm_stage_tile_block<D>(qbuf, q_ptr + (long)q0 * qs, qs);
     fm_stage_tile_block<D>(dobuf, do_ptr + (long)q0 * os, os);
     __syncthreads();  // staged tiles visible block-wide
-    if (!active) {
+    // skip tiles whose q rows all precede this wave's kv rows (fully masked)
+    if (!active || (CAUSAL && q0 + FM_BN <= kv0)) {
       __syncthreads();
       continue;
     }
