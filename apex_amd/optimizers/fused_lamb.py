@@ -29,6 +29,7 @@ class FusedLAMB(torch.optim.Optimizer):
         set_grad_none=True,
         max_grad_norm=1.0,
         use_nvlamb=False,
+        capturable=False,
     ):
         if amsgrad:
             raise RuntimeError("FusedLAMB does not support the AMSGrad variant.")
@@ -46,6 +47,12 @@ class FusedLAMB(torch.optim.Optimizer):
         self.set_grad_none = set_grad_none
         self.use_nvlamb = use_nvlamb
         self._dummy_overflow_buf = None
+        # capturable: lr/step live on device and bias corrections are
+        # computed in-kernel (multi_tensor_lamb_capturable), so the whole
+        # step records into a hipGraph and replays with an advancing step
+        self.capturable = capturable
+        self._lr_t = None
+        self._step_t = None
 
     def zero_grad(self, set_to_none: bool = True):
         if self.set_grad_none or set_to_none:
@@ -123,16 +130,32 @@ class FusedLAMB(torch.optim.Optimizer):
             if device.type == "cuda":
                 amp_C = get_ext("amp_C")
                 noop = self._noop_buf(device)
+                if self.capturable:
+                    if self._lr_t is None:
+                        self._lr_t = torch.full((1,), float(group["lr"]),
+                                                dtype=torch.float32, device=device)
+                        self._step_t = torch.zeros(1, dtype=torch.int32, device=device)
+                    if group is self.param_groups[0]:
+                        self._step_t.add_(1)  # device op: advances per replay
                 for lists in [[g16, p16, m16, v16], [g32, p32, m32, v32]]:
                     if not lists[0]:
                         continue
-                    multi_tensor_applier(
-                        amp_C.multi_tensor_lamb, noop, lists,
-                        group["lr"], beta1, beta2, group["eps"], group["step"],
-                        bias_correction, group["weight_decay"], grad_averaging,
-                        self.adam_w_mode, global_grad_norm, group["max_grad_norm"],
-                        self.use_nvlamb,
-                    )
+                    if self.capturable:
+                        multi_tensor_applier(
+                            amp_C.multi_tensor_lamb_capturable, noop, lists,
+                            self._lr_t, beta1, beta2, group["eps"], self._step_t,
+                            bias_correction, group["weight_decay"], grad_averaging,
+                            self.adam_w_mode, global_grad_norm, group["max_grad_norm"],
+                            self.use_nvlamb,
+                        )
+                    else:
+                        multi_tensor_applier(
+                            amp_C.multi_tensor_lamb, noop, lists,
+                            group["lr"], beta1, beta2, group["eps"], group["step"],
+                            bias_correction, group["weight_decay"], grad_averaging,
+                            self.adam_w_mode, global_grad_norm, group["max_grad_norm"],
+                            self.use_nvlamb,
+                        )
             else:
                 for lists in [[g16, p16, m16, v16], [g32, p32, m32, v32]]:
                     if lists[0]:

@@ -152,7 +152,7 @@ def main():
         # FusedAdam models run hipGraph-capturable (device lr/step: bias
         # corrections stay correct across graph replays); FusedLAMB (gpt2)
         # keeps the eager step (its host-side step count would freeze)
-        capturable = (args.model in ("bert", "llama", "transformer_lg")
+        capturable = (args.model in ("bert", "llama", "transformer_lg", "gpt2")
                       and not use_cpu and not distributed
                       and os.environ.get("APEX_BENCH_GRAPH", "1") != "0")
         if args.model == "bert":
@@ -176,7 +176,8 @@ def main():
         else:
             cfg = gpt2_345m_config(seq_len=min(args.seq_len, 1024))
             model = GPTModel(cfg).to(device)
-            opt = FusedLAMB(model.parameters(), lr=1e-4, weight_decay=0.01)
+            opt = FusedLAMB(model.parameters(), lr=1e-4, weight_decay=0.01,
+                            capturable=capturable)
             config_model = "gpt2-345m"
         # overflow_check=False (static bf16 scale): scale_loss performs no
         # host sync, so the whole O2 step is hipGraph-capturable
@@ -223,7 +224,7 @@ def main():
     # overhead entirely. Fallback to eager on any capture failure. The O2
     # transformer paths keep eager (their unscale reads the overflow flag).
     graphed = False
-    if (args.model in ("resnet50", "bert", "llama", "transformer_lg")
+    if (args.model in ("resnet50", "bert", "llama", "transformer_lg", "gpt2")
             and not use_cpu and not distributed
             and os.environ.get("APEX_BENCH_GRAPH", "1") != "0"):
         try:

@@ -72,6 +72,14 @@ void multi_tensor_lamb_stage1_cuda(long chunk_size, at::Tensor noop_flag,
                                    double weight_decay, long grad_averaging, long mode,
                                    at::Tensor global_grad_norm, double max_grad_norm);
 
+void multi_tensor_lamb_capturable_cuda(long chunk_size, at::Tensor noop_flag,
+                                       TensorLists tensor_lists, at::Tensor lr, double beta1,
+                                       double beta2, double eps, at::Tensor step,
+                                       long bias_correction, double weight_decay,
+                                       long grad_averaging, long mode,
+                                       at::Tensor global_grad_norm, double max_grad_norm,
+                                       bool use_nvlamb);
+
 void multi_tensor_lamb_stage2_cuda(long chunk_size, at::Tensor noop_flag,
                                    TensorLists tensor_lists, at::Tensor param_norms,
                                    at::Tensor update_norms, double lr, double weight_decay,

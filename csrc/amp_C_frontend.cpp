@@ -31,6 +31,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused NovoGrad (per-tensor second moment vector)");
   m.def("multi_tensor_lamb", &multi_tensor_lamb_cuda,
         "Fused LAMB (global-norm clip + per-tensor trust ratio)");
+  m.def("multi_tensor_lamb_capturable", &multi_tensor_lamb_capturable_cuda,
+        "hipGraph-capturable LAMB (device lr/step/global-norm)");
   m.def("multi_tensor_lamb_stage1", &multi_tensor_lamb_stage1_cuda,
         "LAMB stage 1 only (sharded: update into g; device global-grad-norm)");
   m.def("multi_tensor_lamb_stage2", &multi_tensor_lamb_stage2_cuda,

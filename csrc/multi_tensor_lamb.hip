@@ -49,6 +49,48 @@ struct LambStage1 {
   }
 };
 
+// capturable stage 1: bias corrections from a DEVICE step pointer (powf
+// in-kernel, like AdamCapturableFunctor) so the launch replays correctly
+// inside a hipGraph with an advancing step.
+template <typename param_t, typename grad_t>
+struct LambStage1Capturable {
+  __device__ void operator()(long chunk_size, volatile int* noop,
+                             const TensorListMeta<4>& meta, int t, long chunk, float beta1,
+                             float beta2, float beta3, const int* step_ptr, int bias_correction,
+                             float eps, int mode, float decay,
+                             const float* global_grad_norm, float max_grad_norm) const {
+    const long base = chunk * chunk_size;
+    grad_t* g = reinterpret_cast<grad_t*>(meta.addrs[0][t]) + base;
+    const param_t* p = reinterpret_cast<const param_t*>(meta.addrs[1][t]) + base;
+    float* m = reinterpret_cast<float*>(meta.addrs[2][t]) + base;
+    float* v = reinterpret_cast<float*>(meta.addrs[3][t]) + base;
+    const long n = min(meta.sizes[t] - base, chunk_size);
+
+    const int step = *step_ptr;
+    float bc1_recip = 1.f, bc2_recip = 1.f;
+    if (bias_correction) {
+      bc1_recip = 1.f / (1.f - powf(beta1, (float)step));
+      bc2_recip = 1.f / (1.f - powf(beta2, (float)step));
+    }
+    const float gnorm = *global_grad_norm;
+    const float clip = (max_grad_norm > 0.f && gnorm > max_grad_norm) ? gnorm / max_grad_norm : 1.0f;
+    const float combined_scale = 1.0f / clip;
+
+    for (long i = threadIdx.x; i < n; i += blockDim.x) {
+      float gf = to_float(g[i]) * combined_scale;
+      float pf = to_float(p[i]);
+      if (mode == 0 && decay != 0.f) gf = fmaf(decay, pf, gf);
+      float mf = fmaf(beta1, m[i], beta3 * gf);
+      float vf = fmaf(beta2, v[i], (1.f - beta2) * gf * gf);
+      float update = (mf * bc1_recip) / (sqrtf(vf * bc2_recip) + eps);
+      if (mode == 1 && decay != 0.f) update = fmaf(decay, pf, update);
+      g[i] = from_float<grad_t>(update);
+      m[i] = mf;
+      v[i] = vf;
+    }
+  }
+};
+
 // ---- stage 2: p -= ratio * update, ratio = lr * pnorm/unorm (trust) ----
 template <typename param_t, typename grad_t, bool kMp>
 struct LambStage2 {
@@ -225,6 +267,42 @@ void multi_tensor_lamb_stage2_cuda(long chunk_size, at::Tensor noop_flag,
                             LambStage2<param_scalar, grad_scalar, false>(),
                             pn.data_ptr<float>(), un.data_ptr<float>(), (const float*)nullptr,
                             (float)lr, (float)weight_decay, (int)use_nvlamb,
+                            (const float*)nullptr);
+    }()));
+  }()));
+}
+
+// hipGraph-capturable LAMB: device lr/step, in-kernel bias corrections,
+// device global-grad-norm end to end.
+void multi_tensor_lamb_capturable_cuda(long chunk_size, at::Tensor noop_flag,
+                                       TensorLists tensor_lists, at::Tensor lr, double beta1,
+                                       double beta2, double eps, at::Tensor step,
+                                       long bias_correction, double weight_decay,
+                                       long grad_averaging, long mode,
+                                       at::Tensor global_grad_norm, double max_grad_norm,
+                                       bool use_nvlamb) {
+  const float beta3 = grad_averaging ? (float)(1.0 - beta1) : 1.0f;
+  auto gnorm = global_grad_norm.to(at::kFloat);
+  const auto g_t = tensor_lists[0][0].scalar_type();
+  const auto p_t = tensor_lists[1][0].scalar_type();
+  APEX_DISPATCH_FLOAT_HALF_BF(p_t, "multi_tensor_lamb_capturable", ([&] {
+    using param_scalar = scalar_t;
+    APEX_DISPATCH_FLOAT_HALF_BF(g_t, "multi_tensor_lamb_capturable", ([&] {
+      using grad_scalar = scalar_t;
+      multi_tensor_apply<4>(chunk_size, noop_flag, tensor_lists,
+                            LambStage1Capturable<param_scalar, grad_scalar>(), (float)beta1,
+                            (float)beta2, beta3, step.data_ptr<int>(), (int)bias_correction,
+                            (float)eps, (int)mode, (float)weight_decay,
+                            gnorm.data_ptr<float>(), (float)max_grad_norm);
+
+      auto pn = multi_tensor_l2norm_cuda(chunk_size, noop_flag, {tensor_lists[1]}, true)[1];
+      auto un = multi_tensor_l2norm_cuda(chunk_size, noop_flag, {tensor_lists[0]}, true)[1];
+
+      TensorLists stage2_lists = {tensor_lists[1], tensor_lists[0]};
+      multi_tensor_apply<2>(chunk_size, noop_flag, stage2_lists,
+                            LambStage2<param_scalar, grad_scalar, false>(),
+                            pn.data_ptr<float>(), un.data_ptr<float>(), lr.data_ptr<float>(),
+                            0.f, (float)weight_decay, (int)use_nvlamb,
                             (const float*)nullptr);
     }()));
   }()));

@@ -157,3 +157,50 @@ def test_loss_scaler_gpu_dynamic():
     o = [torch.empty_like(g_inf[0])]
     assert s.unscale_grads(g_inf, o)
     assert s.loss_scale() < 2.0 ** 10 or s._hysteresis_t is not None
+
+
+def test_fused_lamb_capturable_graph_replay():
+    """FusedLAMB(capturable=True): graph capture + replays track the eager
+    FusedLAMB (device step advances per replay, in-kernel bias correction)."""
+    from apex_amd.optimizers import FusedLAMB
+
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(64, 64), torch.nn.Tanh(),
+                                torch.nn.Linear(64, 32)).cuda()
+    ref = torch.nn.Sequential(torch.nn.Linear(64, 64), torch.nn.Tanh(),
+                              torch.nn.Linear(64, 32)).cuda()
+    ref.load_state_dict(model.state_dict())
+    opt = FusedLAMB(model.parameters(), lr=1e-3, weight_decay=0.01,
+                    capturable=True, set_grad_none=False)
+    ropt = FusedLAMB(ref.parameters(), lr=1e-3, weight_decay=0.01,
+                     set_grad_none=False)
+    xs = [torch.randn(8, 64, device="cuda") for _ in range(5)]
+
+    def fwd_bwd(m, x):
+        for p in m.parameters():
+            if p.grad is not None:
+                p.grad.zero_()
+        m(x).pow(2).mean().backward()
+
+    # eager warmup (momentum init, algo caches)
+    fwd_bwd(model, xs[0]); opt.step()
+    fwd_bwd(ref, xs[0]); ropt.step()
+
+    fwd_bwd(model, xs[1])
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        opt.step()
+    g.replay()
+    fwd_bwd(ref, xs[1]); ropt.step()
+    torch.cuda.synchronize()
+    for p, rp in zip(model.parameters(), ref.parameters()):
+        torch.testing.assert_close(p.detach(), rp.detach(), rtol=2e-4, atol=2e-5)
+
+    for x in xs[2:]:
+        fwd_bwd(model, x)
+        g.replay()
+        fwd_bwd(ref, x); ropt.step()
+    torch.cuda.synchronize()
+    for p, rp in zip(model.parameters(), ref.parameters()):
+        torch.testing.assert_close(p.detach(), rp.detach(), rtol=5e-4, atol=5e-5)
+    assert int(opt._step_t.item()) == 5
