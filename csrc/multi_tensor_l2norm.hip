@@ -10,6 +10,10 @@
 // reproducible across runs — required by the L1 determinism harness; fp32
 // atomics would not be).
 #include "amp_C.h"
+
+#include <mutex>
+#include <string>
+#include <unordered_map>
 #include "multi_tensor_apply.h"
 
 namespace {
@@ -118,8 +122,25 @@ __global__ void __launch_bounds__(MTA_BLOCK) norm_cleanup_kernel(
   if (threadIdx.x == 0) *dst = kMax ? r : sqrtf(r);
 }
 
-// host helper: global chunk prefix per tensor -> device int tensor
+// host helper: global chunk prefix per tensor -> device int tensor.
+// CACHED by (device, chunk_size, sizes): the prefix depends only on tensor
+// sizes, and the H2D upload must NOT happen inside a hipGraph capture — a
+// captured pageable-host copy replays against freed host memory (this was a
+// GPU memory fault in the captured FusedLAMB step). The first (eager,
+// warmup) call uploads; captured calls reuse the resident device tensor.
 at::Tensor make_prefix(const std::vector<at::Tensor>& ts, long chunk_size, at::Device dev) {
+  static std::mutex mu;
+  static std::unordered_map<std::string, at::Tensor> cache;
+  std::string key = std::to_string((int)dev.index()) + ":" + std::to_string(chunk_size);
+  for (auto& t : ts) {
+    key += ',';
+    key += std::to_string(t.numel());
+  }
+  {
+    std::lock_guard<std::mutex> g(mu);
+    auto it = cache.find(key);
+    if (it != cache.end()) return it->second;
+  }
   std::vector<int> pfx(ts.size() + 1, 0);
   long c = 0;
   for (size_t i = 0; i < ts.size(); ++i) {
@@ -127,7 +148,10 @@ at::Tensor make_prefix(const std::vector<at::Tensor>& ts, long chunk_size, at::D
     pfx[i + 1] = (int)c;
   }
   auto cpu = at::from_blob(pfx.data(), {(long)pfx.size()}, at::kInt).clone();
-  return cpu.to(dev, /*non_blocking=*/true);
+  auto devt = cpu.to(dev);
+  std::lock_guard<std::mutex> g(mu);
+  cache.emplace(key, devt);
+  return devt;
 }
 
 std::vector<at::Tensor> l2norm_impl(long chunk_size, at::Tensor noop_flag,
