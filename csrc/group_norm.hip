@@ -236,7 +236,24 @@ int gn_splits(long work) {
 // L2 provides the persistence.
 constexpr long GN_ONEPASS_SLAB_BYTES = 1 * 1024 * 1024;  // per-block L2 budget
 
-template <typename T, bool AFFINE, bool SILU>
+template <typename T, int N>
+struct VecN {
+  alignas(sizeof(T) * N) T a[N];
+};
+template <typename T, int N>
+__device__ __forceinline__ void load_vecn(VecN<T, N>& d, const T* s) {
+  d = *reinterpret_cast<const VecN<T, N>*>(s);
+}
+template <typename T, int N>
+__device__ __forceinline__ void store_vecn(T* d, const VecN<T, N>& s) {
+  *reinterpret_cast<VecN<T, N>*>(d) = s;
+}
+
+// VW = elements per vector access (8 when the per-group channel count
+// divides by 8, else 4/2/1) — the r2 scalar form measured 250-500 GB/s and
+// LOST to two-pass; the vectorized form makes the single launch pay off at
+// the launch-bound diffusion shapes.
+template <typename T, bool AFFINE, bool SILU, int VW>
 __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
     const T* __restrict__ x, T* __restrict__ y, float* __restrict__ mean_out,
     float* __restrict__ rstd_out, const float* __restrict__ w, const float* __restrict__ b,
@@ -246,17 +263,24 @@ __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
   const long cpg = C / G;
   const long c0 = g * cpg;
   const long count = HW * cpg;
+  const long vpr = cpg / VW;          // vectors per row
+  const long nvec = HW * vpr;
   const T* xp = x + n * HW * C + c0;
 
-  // pass A: stats. Consecutive threads take consecutive channels within a
-  // row (cpg-wide contiguous segments in NHWC).
+  // pass A: stats (VW-wide vector loads; consecutive threads take
+  // consecutive vectors within a row)
   float sum = 0.f, sq = 0.f;
-  for (long i = threadIdx.x; i < count; i += blockDim.x) {
-    const long r = i / cpg;
-    const long c = i % cpg;
-    const float v = to_float(xp[r * C + c]);
-    sum += v;
-    sq = fmaf(v, v, sq);
+  for (long vi = threadIdx.x; vi < nvec; vi += blockDim.x) {
+    const long r = vi / vpr;
+    const long cc = (vi % vpr) * VW;
+    VecN<T, VW> v;
+    load_vecn(v, xp + r * C + cc);
+#pragma unroll
+    for (int j = 0; j < VW; ++j) {
+      const float f = to_float(v.a[j]);
+      sum += f;
+      sq = fmaf(f, f, sq);
+    }
   }
   __shared__ float smem[GN_BLOCK / WAVE_SIZE];
   sum = block_reduce_sum(sum, smem);
@@ -273,15 +297,21 @@ __global__ void __launch_bounds__(GN_BLOCK) gn_fwd_onepass_kernel(
   __syncthreads();
   const float mu = s_mu, rs = s_rs;
 
-  // pass B: normalize from the (L2-resident) slab
+  // pass B: normalize from the (L2-resident) slab, vectorized both ways
   T* yp = y + n * HW * C + c0;
-  for (long i = threadIdx.x; i < count; i += blockDim.x) {
-    const long r = i / cpg;
-    const long c = i % cpg;
-    float v = (to_float(xp[r * C + c]) - mu) * rs;
-    if (AFFINE) v = fmaf(v, w[c0 + c], b[c0 + c]);
-    if (SILU) v = v / (1.f + __expf(-v));
-    yp[r * C + c] = from_float<T>(v);
+  for (long vi = threadIdx.x; vi < nvec; vi += blockDim.x) {
+    const long r = vi / vpr;
+    const long cc = (vi % vpr) * VW;
+    VecN<T, VW> v, o;
+    load_vecn(v, xp + r * C + cc);
+#pragma unroll
+    for (int j = 0; j < VW; ++j) {
+      float f = (to_float(v.a[j]) - mu) * rs;
+      if (AFFINE) f = fmaf(f, w[c0 + cc + j], b[c0 + cc + j]);
+      if (SILU) f = f / (1.f + __expf(-f));
+      o.a[j] = from_float<T>(f);
+    }
+    store_vecn(yp + r * C + cc, o);
   }
 }
 
@@ -320,13 +350,26 @@ std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tens
 
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "group_norm_nhwc_fwd", ([&] {
     if (one_pass) {
+      // widest vector dividing the per-group channel count, <= 16 bytes
+      const int vw_max = (int)(16 / xc.element_size());
+      int vw = 1;
+      for (int cand : {8, 4, 2}) {
+        if (cand <= vw_max && (cpg % cand) == 0) { vw = cand; break; }
+      }
       auto launch1 = [&](auto aff, auto sl) {
-        auto kfn = gn_fwd_onepass_kernel<scalar_t, decltype(aff)::value, decltype(sl)::value>;
-        hipLaunchKernelGGL(kfn, dim3((uint32_t)(N * G)), dim3(GN_BLOCK), 0, stream,
-                           (const scalar_t*)xc.data_ptr(), (scalar_t*)y.data_ptr(),
-                           mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           affine ? w32.data_ptr<float>() : nullptr,
-                           affine ? b32.data_ptr<float>() : nullptr, HW, C, G, (float)eps);
+        auto run = [&](auto vwc) {
+          auto kfn = gn_fwd_onepass_kernel<scalar_t, decltype(aff)::value,
+                                           decltype(sl)::value, decltype(vwc)::value>;
+          hipLaunchKernelGGL(kfn, dim3((uint32_t)(N * G)), dim3(GN_BLOCK), 0, stream,
+                             (const scalar_t*)xc.data_ptr(), (scalar_t*)y.data_ptr(),
+                             mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                             affine ? w32.data_ptr<float>() : nullptr,
+                             affine ? b32.data_ptr<float>() : nullptr, HW, C, G, (float)eps);
+        };
+        if (vw == 8) run(std::integral_constant<int, 8>{});
+        else if (vw == 4) run(std::integral_constant<int, 4>{});
+        else if (vw == 2) run(std::integral_constant<int, 2>{});
+        else run(std::integral_constant<int, 1>{});
       };
       using Tt = std::true_type;
       using Ff = std::false_type;
