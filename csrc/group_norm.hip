@@ -339,14 +339,21 @@ std::vector<at::Tensor> group_norm_nhwc_fwd(at::Tensor x, c10::optional<at::Tens
   auto b32 = affine ? bias->to(at::kFloat).contiguous() : at::Tensor();
   auto stream = current_stream();
 
-  // MEASURED (profiles/probe_group_norm2.log): the one-pass form is slower
-  // than two-pass at nearly every diffusion shape on this pool (0.4-1.1x) —
-  // per-element div/mod addressing and 2-byte loads dominate before the
-  // saved HBM pass can pay. Auto therefore keeps two-pass; passes=1 opts in
-  // explicitly (the reference's knob surface is preserved).
+  // MEASURED (profiles/probe_group_norm3.log): with VW-wide vector access
+  // the one-pass form beats two-pass 1.4-2.7x when the per-group channel
+  // count vectorizes 4+ wide and the slab stays L2-resident (C >= 512 at
+  // G=32); narrow-vector shapes (C=320 -> cpg=10) still lose, so auto keeps
+  // two-pass there. passes=1/2 force either form (reference knob surface).
   const long slab_bytes = HW * cpg * (long)xc.element_size();
+  const int vw_elig_max = (int)(16 / xc.element_size());
+  int vw_elig = 1;
+  for (int cand : {8, 4, 2}) {
+    if (cand <= vw_elig_max && (cpg % cand) == 0) { vw_elig = cand; break; }
+  }
   const bool eligible = slab_bytes <= GN_ONEPASS_SLAB_BYTES;
-  const bool one_pass = eligible && passes == 1;
+  const bool one_pass = eligible &&
+      (passes == 1 ||
+       (passes == 0 && vw_elig >= 4 && slab_bytes <= 256 * 1024 && N * G >= 128));
 
   APEX_DISPATCH_FLOAT_HALF_BF(xc.scalar_type(), "group_norm_nhwc_fwd", ([&] {
     if (one_pass) {
