@@ -32,6 +32,13 @@ constexpr int FM_WAVES = 4;   // waves per workgroup
 constexpr int FM_BN = 32;     // kv tile (one MFMA K step for P@V)
 constexpr int FM_PAD = 8;     // LDS tile row padding (shorts) -> 16B-aligned rows
 
+// wave-local LDS drain WITHOUT waiting outstanding vector-memory loads:
+// s_waitcnt lgkmcnt(0), vmcnt(max), expcnt(max) — gfx9-family encoding
+// (vmcnt bits [3:0]+[15:14], expcnt [6:4], lgkmcnt [13:8]). Used for the
+// per-wave P/dS fragment bounces so they do NOT serialize against the
+// double-buffered tile prefetch in flight.
+#define FM_WAIT_LDS() __builtin_amdgcn_s_waitcnt(0xC07F)
+
 // Stage one 32 x D bf16 tile from HBM into LDS (row-major, padded rows),
 // wave-cooperative and fully vectorized (bf16x8 both sides). Replaces the
 // per-fragment scalar column-strided HBM gathers that capped the round-1
@@ -101,10 +108,11 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   const short* k_ptr = K + bb * kb + hh * kh;
   const short* v_ptr = V + bb * vb + hh * vh;
 
+  constexpr int FM_TILE_ELEMS = FM_BN * (D + FM_PAD);
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_v[FM_BN * (D + FM_PAD)];  // ONE tile per workgroup
+  __shared__ short lds_v[2 * FM_TILE_ELEMS];  // double-buffered shared tile
   short* pbuf = lds_p[wave];
-  short* vbuf = lds_v;
+  short* vbuf_pair = lds_v;
 
   // Q rows for this wave, as A-fragments, resident for the whole pass
   bf16x8 aq[NK];
@@ -136,11 +144,20 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
   // a wave's own last useful tile (tiles past it are fully causal-masked)
   const int kv_end_wave = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile_block<D>(vbuf, v_ptr + (long)kv0 * vs, vs);
-    __syncthreads();  // staged tile visible block-wide
+  const int ntiles = (kv_end + FM_BN - 1) / FM_BN;
+  if (ntiles > 0) fm_stage_tile_block<D>(vbuf_pair, v_ptr, vs);
+  __syncthreads();
+  int vb_cur = 0;
+  for (int ti = 0; ti < ntiles; ++ti) {
+    const int kv0 = ti * FM_BN;
+    // prefetch the next tile into the other buffer while this one computes
+    if (ti + 1 < ntiles)
+      fm_stage_tile_block<D>(vbuf_pair + (vb_cur ^ 1) * FM_TILE_ELEMS,
+                             v_ptr + (long)(kv0 + FM_BN) * vs, vs);
+    short* vbuf = vbuf_pair + vb_cur * FM_TILE_ELEMS;
     if (!active || kv0 >= kv_end_wave) {
       __syncthreads();
+      vb_cur ^= 1;
       continue;
     }
     // ---- S = scale * Q K^T for this 16 x 32 tile (two 16x16 halves) ----
@@ -204,7 +221,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
         const __hip_bfloat16 pb = __float2bfloat16(p_val[j][q]);
         pbuf[(my_r0 + q) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&pb);
       }
-    __builtin_amdgcn_s_waitcnt(0);  // wave-local: drain LDS stores before reads
+    FM_WAIT_LDS();  // wave-local: drain LDS stores before reads
     const bf16x8 ap = *reinterpret_cast<const bf16x8*>(
         pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
 
@@ -220,7 +237,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
       }
       acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc[d], 0, 0, 0);
     }
-    __syncthreads();  // all waves done reading before the next tile's stores
+    __syncthreads();  // prefetch visible + all reads of the current tile done
+    vb_cur ^= 1;
   }
 
   if (!active) return;
@@ -272,10 +290,11 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
+  constexpr int FM_TILE_ELEMS = FM_BN * (D + FM_PAD);
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_k[FM_BN * (D + FM_PAD)];  // ONE shared K tile per WG
+  __shared__ short lds_k[2 * FM_TILE_ELEMS];  // double-buffered shared K tile
   short* pbuf = lds_p[wave];
-  short* kbuf = lds_k;
+  short* kbuf_pair = lds_k;
 
   bf16x8 aq[NK], ado[NK];
   const int a_row = min(q0 + (lane & 15), S - 1);
@@ -298,11 +317,19 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
 
   const int kv_end_wave = CAUSAL ? min(SKV, ((q0 + FM_ROWS - 1) / FM_BN + 1) * FM_BN) : SKV;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += FM_BN) {
-    fm_stage_tile_block<D>(kbuf, k_ptr + (long)kv0 * ks, ks);
-    __syncthreads();  // staged K visible block-wide
+  const int ntiles = (kv_end + FM_BN - 1) / FM_BN;
+  if (ntiles > 0) fm_stage_tile_block<D>(kbuf_pair, k_ptr, ks);
+  __syncthreads();
+  int kb_cur = 0;
+  for (int ti = 0; ti < ntiles; ++ti) {
+    const int kv0 = ti * FM_BN;
+    if (ti + 1 < ntiles)
+      fm_stage_tile_block<D>(kbuf_pair + (kb_cur ^ 1) * FM_TILE_ELEMS,
+                             k_ptr + (long)(kv0 + FM_BN) * ks, ks);
+    short* kbuf = kbuf_pair + kb_cur * FM_TILE_ELEMS;
     if (!active || kv0 >= kv_end_wave) {
       __syncthreads();
+      kb_cur ^= 1;
       continue;
     }
 #pragma unroll
@@ -330,7 +357,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
         pbuf[(my_r0 + qi) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&db);
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    FM_WAIT_LDS();
     const bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
         pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
 #pragma unroll
@@ -341,7 +368,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
         bK[jj] = kbuf[((lane >> 4) * 8 + jj) * (D + FM_PAD) + d * 16 + (lane & 15)];
       dq_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, bK, dq_acc[d], 0, 0, 0);
     }
-    __syncthreads();  // all waves done with kbuf before the next stage
+    __syncthreads();  // prefetch visible + all reads of the current tile done
+    kb_cur ^= 1;
   }
   if (!active) return;
 #pragma unroll
@@ -379,14 +407,15 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const float* lse = LSE + (long)bh * S;
   const float* delta = DELTA + (long)bh * S;
 
+  constexpr int FM_TILE_ELEMS = FM_BN * (D + FM_PAD);
   __shared__ short lds_p[FM_WAVES][FM_ROWS * FM_BN];
   __shared__ short lds_ds[FM_WAVES][FM_ROWS * FM_BN];
-  __shared__ short lds_q[FM_BN * (D + FM_PAD)];   // ONE shared Q tile per WG
-  __shared__ short lds_do[FM_BN * (D + FM_PAD)];  // ONE shared dO tile per WG
+  __shared__ short lds_q[2 * FM_TILE_ELEMS];   // double-buffered shared tiles
+  __shared__ short lds_do[2 * FM_TILE_ELEMS];
   short* pbuf = lds_p[wave];
   short* dsbuf = lds_ds[wave];
-  short* qbuf = lds_q;
-  short* dobuf = lds_do;
+  short* qbuf_pair = lds_q;
+  short* dobuf_pair = lds_do;
 
   bf16x8 ak[NK], av[NK];
   const int a_row = min(kv0 + (lane & 15), SKV - 1);
@@ -409,13 +438,27 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
   const int q_start = CAUSAL
       ? ((blockIdx.x * (FM_WAVES * FM_ROWS)) / FM_BN) * FM_BN : 0;
 
-  for (int q0 = q_start; q0 < S; q0 += FM_BN) {
-    fm_stage_tile_block<D>(qbuf, q_ptr + (long)q0 * qs, qs);
-    fm_stage_tile_block<D>(dobuf, do_ptr + (long)q0 * os, os);
-    __syncthreads();  // staged tiles visible block-wide
+  const int ntiles = (S - q_start + FM_BN - 1) / FM_BN;
+  if (ntiles > 0) {
+    fm_stage_tile_block<D>(qbuf_pair, q_ptr + (long)q_start * qs, qs);
+    fm_stage_tile_block<D>(dobuf_pair, do_ptr + (long)q_start * os, os);
+  }
+  __syncthreads();
+  int tb_cur = 0;
+  for (int ti = 0; ti < ntiles; ++ti) {
+    const int q0 = q_start + ti * FM_BN;
+    if (ti + 1 < ntiles) {
+      fm_stage_tile_block<D>(qbuf_pair + (tb_cur ^ 1) * FM_TILE_ELEMS,
+                             q_ptr + (long)(q0 + FM_BN) * qs, qs);
+      fm_stage_tile_block<D>(dobuf_pair + (tb_cur ^ 1) * FM_TILE_ELEMS,
+                             do_ptr + (long)(q0 + FM_BN) * os, os);
+    }
+    short* qbuf = qbuf_pair + tb_cur * FM_TILE_ELEMS;
+    short* dobuf = dobuf_pair + tb_cur * FM_TILE_ELEMS;
     // skip tiles whose q rows all precede this wave's kv rows (fully masked)
     if (!active || (CAUSAL && q0 + FM_BN <= kv0)) {
       __syncthreads();
+      tb_cur ^= 1;
       continue;
     }
 #pragma unroll
@@ -445,7 +488,7 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
         dsbuf[idx] = *reinterpret_cast<const short*>(&db);
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    FM_WAIT_LDS();
     const bf16x8 a_p = *reinterpret_cast<const bf16x8*>(
         pbuf + (lane & 15) * FM_BN + (lane >> 4) * 8);
     const bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
@@ -462,7 +505,8 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
       dv_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_do, dv_acc[d], 0, 0, 0);
       dk_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_q, dk_acc[d], 0, 0, 0);
     }
-    __syncthreads();  // all waves done with q/do tiles before the next stage
+    __syncthreads();  // prefetch visible + all reads of the current tiles done
+    tb_cur ^= 1;
   }
   if (!active) return;
 #pragma unroll
