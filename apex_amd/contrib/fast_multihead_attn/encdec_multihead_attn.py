@@ -61,8 +61,9 @@ class EncdecMultiheadAttn(torch.nn.Module):
         if (not need_weights and key_padding_mask is None and attn_mask is None
                 and flash_attention_supported(q, dropout=dropout_active, k=k)):
             # MFMA flash cross-attention (Sq != Skv supported): no sq x sk
-            # matrix, strided BSHD views pass copy-free
-            ctx = flash_attention(q, k, v, causal=False, scale=self.scaling)
+            # matrix, strided BSHD views pass copy-free, fused dropout
+            ctx = flash_attention(q, k, v, causal=False, scale=self.scaling,
+                                  dropout_p=dropout_active)
             ctx = ctx.permute(2, 0, 1, 3).reshape(sq * b, h)
             out = fused_dense_function(ctx, self.out_proj_weight, self.out_proj_bias)
             out = out.reshape(sq, b, h)

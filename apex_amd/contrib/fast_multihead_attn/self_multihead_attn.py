@@ -62,9 +62,10 @@ class SelfMultiheadAttn(torch.nn.Module):
         if (not need_weights and key_padding_mask is None
                 and (attn_mask is None or attn_mask == "causal")
                 and flash_attention_supported(q, dropout=dropout_active)):
-            # MFMA flash path: hardware-validated round 2, no S x S matrix
+            # MFMA flash path (fused philox attention dropout in-kernel)
             ctx = flash_attention(q, k, v,
-                                  causal=(attn_mask == "causal"), scale=self.scaling)
+                                  causal=(attn_mask == "causal"), scale=self.scaling,
+                                  dropout_p=dropout_active)
             ctx = ctx.permute(2, 0, 1, 3).reshape(s * b, h)
             out = fused_dense_function(ctx, self.out_proj_weight, self.out_proj_bias)
             out = out.reshape(s, b, h)

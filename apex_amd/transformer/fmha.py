@@ -21,15 +21,17 @@ from .._ext import get_ext
 
 def flash_attention_supported(q, dropout=0.0, k=None):
     """True when the MFMA flash kernel can take these tensors: CUDA bf16,
-    head_dim 64/128, seq lens % 32 == 0, no attention dropout. Pass ``k``
-    for cross-attention (Skv may differ from Sq)."""
+    head_dim 64/128, seq lens % 32 == 0. Attention dropout is FUSED
+    (philox mask regenerated in the backward kernels — no stored mask), so
+    any ``dropout`` in [0, 1) is supported on the GPU path. Pass ``k`` for
+    cross-attention (Skv may differ from Sq)."""
     return (
         q.is_cuda
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
         and q.shape[-2] % 32 == 0
         and (k is None or k.shape[-2] % 32 == 0)
-        and dropout == 0.0
+        and 0.0 <= dropout < 1.0
     )
 
 
@@ -46,12 +48,14 @@ def eager_attention_reference(q, k, v, causal=False, scale=None):
     return torch.matmul(p, v.float()), lse
 
 
-def flash_attention_forward(q, k, v, causal=False, scale=None):
+def flash_attention_forward(q, k, v, causal=False, scale=None, dropout_p=0.0, seed=0):
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
         ext = get_ext("mfma")
-        out, lse = ext.fmha_fwd(q, k, v, causal, float(scale))
+        out, lse = ext.fmha_fwd(q, k, v, causal, float(scale), float(dropout_p), int(seed))
         return out, lse
+    if dropout_p:
+        raise RuntimeError("flash attention dropout is GPU-only")
     out, lse = eager_attention_reference(q, k, v, causal, scale)
     return out.to(q.dtype), lse
 
@@ -80,22 +84,33 @@ class FlashAttentionFunction(torch.autograd.Function):
     use_fused_backward = True
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, dropout_p=0.0):
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        out, lse = flash_attention_forward(q, k, v, causal, scale)
+        seed = 0
+        if dropout_p:
+            # CPU-RNG seed: deterministic under torch.manual_seed; the same
+            # (seed, element-index) philox stream regenerates the mask in
+            # the backward kernels (mask-free backward)
+            seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+        out, lse = flash_attention_forward(q, k, v, causal, scale, dropout_p, seed)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.dropout = (dropout_p, seed)
         return out, lse
 
     @staticmethod
     def backward(ctx, dout, dlse_unused):
         q, k, v, out, lse = ctx.saved_tensors
         causal, scale = ctx.causal, ctx.scale
-        if FlashAttentionFunction.use_fused_backward and q.is_cuda:
+        dropout_p, seed = ctx.dropout
+        if q.is_cuda and (FlashAttentionFunction.use_fused_backward or dropout_p):
+            # dropout requires the fused backward (the mask lives in the
+            # philox stream, regenerated in-kernel)
             ext = get_ext("mfma")
-            dq, dk, dv = ext.fmha_bwd(dout, q, k, v, out, lse, causal, float(scale))
-            return dq, dk, dv, None, None
+            dq, dk, dv = ext.fmha_bwd(dout, q, k, v, out, lse, causal, float(scale),
+                                      float(dropout_p), int(seed))
+            return dq, dk, dv, None, None, None
         if q.is_cuda:
             # batched-GEMM recompute: 5 hipBLASLt GEMMs + 3 single-pass
             # fused kernels (delta / p / ds from csrc/fmha.hip) — P is
@@ -110,7 +125,7 @@ class FlashAttentionFunction(torch.autograd.Function):
             ds = ext.fmha_ds(p, dp, delta, float(scale))      # bf16, 1 pass
             dq = torch.matmul(ds, k)
             dk = torch.matmul(ds.transpose(-1, -2), q)
-            return dq, dk, dv, None, None
+            return dq, dk, dv, None, None, None
         S = q.shape[-2]
         qf, kf, vf = q.float(), k.float(), v.float()
         dof = dout.float()
@@ -133,10 +148,11 @@ class FlashAttentionFunction(torch.autograd.Function):
             ds = p * (dp - delta[..., i0:i1, :])
             dq[..., i0:i1, :] = torch.matmul(ds, kf) * scale
             dk += torch.matmul(ds.transpose(-1, -2), qi) * scale
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None
 
 
-def flash_attention(q, k, v, causal=False, scale=None):
-    """Differentiable flash attention; returns the attention output."""
-    out, _ = FlashAttentionFunction.apply(q, k, v, causal, scale)
+def flash_attention(q, k, v, causal=False, scale=None, dropout_p=0.0):
+    """Differentiable flash attention; returns the attention output.
+    ``dropout_p`` applies fused philox attention dropout (GPU only)."""
+    out, _ = FlashAttentionFunction.apply(q, k, v, causal, scale, dropout_p)
     return out

@@ -39,6 +39,51 @@ constexpr int FM_PAD = 8;     // LDS tile row padding (shorts) -> 16B-aligned ro
 // double-buffered tile prefetch in flight.
 #define FM_WAIT_LDS() __builtin_amdgcn_s_waitcnt(0xC07F)
 
+// Philox4x32-10 attention dropout (reference analogue: the philox.cuh-era
+// fused-attention dropout). Counter = flat (bh, q, kv) index of the
+// attention-probability element, so forward and BOTH backward kernels
+// regenerate the identical mask — no stored mask tensor.
+__device__ __forceinline__ void fm_philox_round(uint32_t& c0, uint32_t& c1, uint32_t& c2,
+                                                uint32_t& c3, uint32_t k0, uint32_t k1) {
+  const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  const uint32_t hi0 = __umulhi(M0, c0), lo0 = M0 * c0;
+  const uint32_t hi1 = __umulhi(M1, c2), lo1 = M1 * c2;
+  c0 = hi1 ^ c1 ^ k0;
+  c1 = lo1;
+  c2 = hi0 ^ c3 ^ k1;
+  c3 = lo0;
+}
+
+__device__ __forceinline__ float fm_philox_uniform(unsigned long long seed, long idx) {
+  uint32_t c0 = (uint32_t)((unsigned long)idx >> 2);
+  uint32_t c1 = (uint32_t)((unsigned long)idx >> 34);
+  uint32_t c2 = 0u, c3 = 0u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    fm_philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  uint32_t res;
+  switch (idx & 3) {
+    case 0: res = c0; break;
+    case 1: res = c1; break;
+    case 2: res = c2; break;
+    default: res = c3; break;
+  }
+  return (res >> 8) * (1.0f / 16777216.0f);
+}
+
+// post-softmax keep/scale: returns p * mask / (1-prob) for element
+// (bh, row=q, col=kv); the softmax denominator stays PRE-dropout.
+__device__ __forceinline__ float fm_dropout_p(float p, unsigned long long seed, long bh,
+                                              long q_g, long kv_g, long S, long SKV,
+                                              float prob, float rinv) {
+  const long idx = ((long)bh * S + q_g) * SKV + kv_g;
+  return fm_philox_uniform(seed, idx) >= prob ? p * rinv : 0.f;
+}
+
 // Stage one 32 x D bf16 tile from HBM into LDS (row-major, padded rows),
 // wave-cooperative and fully vectorized (bf16x8 both sides). Replaces the
 // per-fragment scalar column-strided HBM gathers that capped the round-1
@@ -88,11 +133,12 @@ __device__ __forceinline__ float row_reduce_sum16(float v) {
   return v;
 }
 
-template <bool CAUSAL, int D>
+template <bool CAUSAL, int D, bool DROPOUT = false>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     short* __restrict__ O, float* __restrict__ LSE, int S, int SKV, float scale, int H,
-    long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs) {
+    long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
+    float drop_p = 0.f, float drop_rinv = 1.f, unsigned long long drop_seed = 0) {
   constexpr int NK = D / 32;  // K chunks for Q@K^T
   constexpr int ND = D / 16;  // 16-col output groups for P@V
   const int lane = threadIdx.x & 63;
@@ -214,11 +260,17 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
       for (int q = 0; q < 4; ++q) acc[d][q] *= alpha[q];
 
     // ---- P: D-fragment -> A-fragment via the wave-local LDS bounce ----
+    // (dropout applies to the P used for P@V; l_run keeps the pre-dropout
+    // softmax denominator)
 #pragma unroll
     for (int j = 0; j < 2; ++j)
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
-        const __hip_bfloat16 pb = __float2bfloat16(p_val[j][q]);
+        float pv = p_val[j][q];
+        if (DROPOUT)
+          pv = fm_dropout_p(pv, drop_seed, bh, q0 + my_r0 + q,
+                            kv0 + j * 16 + (lane & 15), S, SKV, drop_p, drop_rinv);
+        const __hip_bfloat16 pb = __float2bfloat16(pv);
         pbuf[(my_r0 + q) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&pb);
       }
     FM_WAIT_LDS();  // wave-local: drain LDS stores before reads
@@ -268,13 +320,14 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_fwd_kernel(
 //        roles (S^T = K Q^T, lse/delta indexed by the q column);
 //        dV += P^T dO, dK += dS^T Q * scale.
 
-template <bool CAUSAL, int D>
+template <bool CAUSAL, int D, bool DROPOUT = false>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
     const float* __restrict__ DELTA, short* __restrict__ dQ, int S, int SKV, float scale, int H,
     long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
-    long ob, long oh, long os) {
+    long ob, long oh, long os, float drop_p = 0.f, float drop_rinv = 1.f,
+    unsigned long long drop_seed = 0) {
   constexpr int NK = D / 32;
   constexpr int ND = D / 16;
   const int lane = threadIdx.x & 63;
@@ -352,7 +405,13 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
         const int row_g = q0 + my_r0 + qi;
         const int col_g = kv0 + j * 16 + (lane & 15);
         float p = (CAUSAL && col_g > row_g) ? 0.f : __expf(s[qi] * scale - lse[row_g]);
-        const float ds = p * (dp[qi] - delta[row_g]);
+        // with dropout: dS = P * (M*rinv*dPd - delta) — the Sum term folds
+        // to the SAME delta = rowsum(dO*O) as the no-dropout case
+        float dpe = dp[qi];
+        if (DROPOUT)
+          dpe = fm_philox_uniform(drop_seed, ((long)bh * S + row_g) * SKV + col_g) >= drop_p
+                    ? dpe * drop_rinv : 0.f;
+        const float ds = p * (dpe - delta[row_g]);
         const __hip_bfloat16 db = __float2bfloat16(ds);
         pbuf[(my_r0 + qi) * FM_BN + j * 16 + (lane & 15)] = *reinterpret_cast<const short*>(&db);
       }
@@ -384,14 +443,15 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dq_kernel(
   }
 }
 
-template <bool CAUSAL, int D>
+template <bool CAUSAL, int D, bool DROPOUT = false>
 __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
     const short* __restrict__ Q, const short* __restrict__ K, const short* __restrict__ V,
     const short* __restrict__ dO, const float* __restrict__ LSE,
     const float* __restrict__ DELTA, short* __restrict__ dK, short* __restrict__ dV,
     int S, int SKV, float scale, int H,
     long qb, long qh, long qs, long kb, long kh, long ks, long vb, long vh, long vs,
-    long ob, long oh, long os) {
+    long ob, long oh, long os, float drop_p = 0.f, float drop_rinv = 1.f,
+    unsigned long long drop_seed = 0) {
   constexpr int NK = D / 32;
   constexpr int ND = D / 16;
   const int lane = threadIdx.x & 63;
@@ -480,9 +540,18 @@ __global__ void __launch_bounds__(FM_WAVES * 64, 2) fmha_bwd_dkv_kernel(
         const int kv_g = kv0 + my_r0 + qi;
         const int q_g = q0 + j * 16 + (lane & 15);
         float p = (CAUSAL && kv_g > q_g) ? 0.f : __expf(sT[qi] * scale - lse[q_g]);
-        const float ds = p * (dpT[qi] - delta[q_g]);
+        // dV consumes the POST-dropout P-hat; dS uses pre-dropout P with
+        // the masked/rescaled dPd (same identities as the dq kernel)
+        float p_hat = p, dpe = dpT[qi];
+        if (DROPOUT) {
+          const bool keep =
+              fm_philox_uniform(drop_seed, ((long)bh * S + q_g) * SKV + kv_g) >= drop_p;
+          p_hat = keep ? p * drop_rinv : 0.f;
+          dpe = keep ? dpe * drop_rinv : 0.f;
+        }
+        const float ds = p * (dpe - delta[q_g]);
         const int idx = (my_r0 + qi) * FM_BN + j * 16 + (lane & 15);
-        const __hip_bfloat16 pb = __float2bfloat16(p);
+        const __hip_bfloat16 pb = __float2bfloat16(p_hat);
         const __hip_bfloat16 db = __float2bfloat16(ds);
         pbuf[idx] = *reinterpret_cast<const short*>(&pb);
         dsbuf[idx] = *reinterpret_cast<const short*>(&db);
@@ -693,7 +762,7 @@ inline bool fm_strided_ok(const at::Tensor& t) {
 }  // namespace
 
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
-                                 double scale) {
+                                 double scale, double dropout_p, long seed) {
   TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_fwd: bf16 only");
   TORCH_CHECK(q.dim() == 4, "fmha_fwd: [B, H, S, D]");
   auto qc = fm_strided_ok(q) ? q : q.contiguous();
@@ -713,21 +782,32 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
   dim3 grid((S + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
   dim3 block(FM_WAVES * 64);
   const float sc = (float)scale;
+  const float dp = (float)dropout_p;
+  const float dr = dp > 0.f ? 1.f / (1.f - dp) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
 
-#define FMHA_LAUNCH(CAUSAL, DD)                                                            \
-  hipLaunchKernelGGL((fmha_fwd_kernel<CAUSAL, DD>), grid, block, 0, stream,                \
+#define FMHA_LAUNCH(CAUSAL, DD, DROP)                                                      \
+  hipLaunchKernelGGL((fmha_fwd_kernel<CAUSAL, DD, DROP>), grid, block, 0, stream,          \
                      (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),             \
                      (const short*)vc.data_ptr(), (short*)out.data_ptr(),                  \
                      lse.data_ptr<float>(), S, SKV, sc, H,                                 \
                      qc.stride(0), qc.stride(1), qc.stride(2),                             \
                      kc.stride(0), kc.stride(1), kc.stride(2),                             \
-                     vc.stride(0), vc.stride(1), vc.stride(2))
-  if (causal) {
-    if (D == 64) FMHA_LAUNCH(true, 64);
-    else FMHA_LAUNCH(true, 128);
+                     vc.stride(0), vc.stride(1), vc.stride(2), dp, dr, sd)
+  if (dp > 0.f) {
+    if (causal) {
+      if (D == 64) FMHA_LAUNCH(true, 64, true);
+      else FMHA_LAUNCH(true, 128, true);
+    } else {
+      if (D == 64) FMHA_LAUNCH(false, 64, true);
+      else FMHA_LAUNCH(false, 128, true);
+    }
+  } else if (causal) {
+    if (D == 64) FMHA_LAUNCH(true, 64, false);
+    else FMHA_LAUNCH(true, 128, false);
   } else {
-    if (D == 64) FMHA_LAUNCH(false, 64);
-    else FMHA_LAUNCH(false, 128);
+    if (D == 64) FMHA_LAUNCH(false, 64, false);
+    else FMHA_LAUNCH(false, 128, false);
   }
 #undef FMHA_LAUNCH
   HIP_CHECK(hipGetLastError());
@@ -735,7 +815,8 @@ std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool 
 }
 
 std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
-                                 at::Tensor out, at::Tensor lse, bool causal, double scale) {
+                                 at::Tensor out, at::Tensor lse, bool causal, double scale,
+                                 double dropout_p, long seed) {
   TORCH_CHECK(q.scalar_type() == at::ScalarType::BFloat16, "fmha_bwd: bf16 only");
   auto qc = fm_strided_ok(q) ? q : q.contiguous();
   auto kc = fm_strided_ok(k) ? k : k.contiguous();
@@ -758,10 +839,13 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
   dim3 grid_kv((SKV + FM_WAVES * FM_ROWS - 1) / (FM_WAVES * FM_ROWS), B * H);
   dim3 block(FM_WAVES * 64);
   const float sc = (float)scale;
+  const float dp = (float)dropout_p;
+  const float dr = dp > 0.f ? 1.f / (1.f - dp) : 1.f;
+  const unsigned long long sd = (unsigned long long)seed;
 
-#define FMHA_BWD_LAUNCH(CAUSAL, DD)                                                        \
+#define FMHA_BWD_LAUNCH(CAUSAL, DD, DROP)                                                  \
   do {                                                                                     \
-    hipLaunchKernelGGL((fmha_bwd_dq_kernel<CAUSAL, DD>), grid, block, 0, stream,           \
+    hipLaunchKernelGGL((fmha_bwd_dq_kernel<CAUSAL, DD, DROP>), grid, block, 0, stream,     \
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
@@ -769,8 +853,8 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
                        qc.stride(0), qc.stride(1), qc.stride(2),                           \
                        kc.stride(0), kc.stride(1), kc.stride(2),                           \
                        vc.stride(0), vc.stride(1), vc.stride(2),                           \
-                       doc.stride(0), doc.stride(1), doc.stride(2));                       \
-    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD>), grid_kv, block, 0, stream,       \
+                       doc.stride(0), doc.stride(1), doc.stride(2), dp, dr, sd);           \
+    hipLaunchKernelGGL((fmha_bwd_dkv_kernel<CAUSAL, DD, DROP>), grid_kv, block, 0, stream, \
                        (const short*)qc.data_ptr(), (const short*)kc.data_ptr(),           \
                        (const short*)vc.data_ptr(), (const short*)doc.data_ptr(),          \
                        lsec.data_ptr<float>(), delta.data_ptr<float>(),                    \
@@ -778,14 +862,22 @@ std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at
                        qc.stride(0), qc.stride(1), qc.stride(2),                           \
                        kc.stride(0), kc.stride(1), kc.stride(2),                           \
                        vc.stride(0), vc.stride(1), vc.stride(2),                           \
-                       doc.stride(0), doc.stride(1), doc.stride(2));                       \
+                       doc.stride(0), doc.stride(1), doc.stride(2), dp, dr, sd);           \
   } while (0)
-  if (causal) {
-    if (D == 64) FMHA_BWD_LAUNCH(true, 64);
-    else FMHA_BWD_LAUNCH(true, 128);
+  if (dp > 0.f) {
+    if (causal) {
+      if (D == 64) FMHA_BWD_LAUNCH(true, 64, true);
+      else FMHA_BWD_LAUNCH(true, 128, true);
+    } else {
+      if (D == 64) FMHA_BWD_LAUNCH(false, 64, true);
+      else FMHA_BWD_LAUNCH(false, 128, true);
+    }
+  } else if (causal) {
+    if (D == 64) FMHA_BWD_LAUNCH(true, 64, false);
+    else FMHA_BWD_LAUNCH(true, 128, false);
   } else {
-    if (D == 64) FMHA_BWD_LAUNCH(false, 64);
-    else FMHA_BWD_LAUNCH(false, 128);
+    if (D == 64) FMHA_BWD_LAUNCH(false, 64, false);
+    else FMHA_BWD_LAUNCH(false, 128, false);
   }
 #undef FMHA_BWD_LAUNCH
   HIP_CHECK(hipGetLastError());

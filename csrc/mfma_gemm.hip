@@ -322,18 +322,24 @@ at::Tensor gemm_bias_mfma(at::Tensor X, at::Tensor W, at::Tensor bias) {
 at::Tensor mfma_tile_probe(at::Tensor A_bf16, at::Tensor B_bf16);
 
 std::vector<at::Tensor> fmha_fwd(at::Tensor q, at::Tensor k, at::Tensor v, bool causal,
-                                 double scale);
+                                 double scale, double dropout_p, long seed);
 std::vector<at::Tensor> fmha_bwd(at::Tensor dout, at::Tensor q, at::Tensor k, at::Tensor v,
-                                 at::Tensor out, at::Tensor lse, bool causal, double scale);
+                                 at::Tensor out, at::Tensor lse, bool causal, double scale,
+                                 double dropout_p, long seed);
 at::Tensor fmha_delta(at::Tensor dout, at::Tensor out);
 at::Tensor fmha_p(at::Tensor s, at::Tensor lse, double scale, bool causal);
 at::Tensor fmha_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fmha_fwd", &fmha_fwd,
-        "flash-attention fwd (bf16, D=64/128) -> (out, lse)");
+        "flash-attention fwd (bf16, D=64/128, philox dropout) -> (out, lse)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"), py::arg("scale"),
+        py::arg("dropout_p") = 0.0, py::arg("seed") = 0);
   m.def("fmha_bwd", &fmha_bwd,
-        "flash-attention MFMA bwd -> (dq, dk, dv)");
+        "flash-attention MFMA bwd (mask-free dropout) -> (dq, dk, dv)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"),
+        py::arg("lse"), py::arg("causal"), py::arg("scale"),
+        py::arg("dropout_p") = 0.0, py::arg("seed") = 0);
   m.def("fmha_delta", &fmha_delta, "rowsum(dout*out) fp32 (one pass)");
   m.def("fmha_p", &fmha_p, "p = exp(s*scale - lse[row]) (+causal mask), bf16 one pass");
   m.def("fmha_ds", &fmha_ds, "ds = p*(dp - delta[row])*scale, bf16 one pass");
