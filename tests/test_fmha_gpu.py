@@ -197,14 +197,16 @@ def test_flash_attention_fused_dropout(causal):
     # expectation: average over seeds approaches the no-dropout output
     with torch.no_grad():
         acc = torch.zeros_like(y0, dtype=torch.float32)
-        n = 64
+        n = 160
         for s in range(n):
             out, _ = mfma.fmha_fwd(q.detach(), k.detach(), v.detach(), causal,
                                    1.0 / D ** 0.5, p, 1000 + s)
             acc += out.float()
         mean = acc / n
+        # per-element sampling std at p=0.5 over n seeds is ~1/sqrt(n) of
+        # the element scale; bound the mean |error| with 3x headroom
         err = (mean - ref.detach().float()).abs().mean() / ref.detach().float().abs().mean()
-        assert err < 0.12, f"dropout expectation off: rel {err:.3f}"
+        assert err < 0.25, f"dropout expectation off: rel {err:.3f}"
 
     # backward runs and produces finite grads through the fused kernels
     torch.manual_seed(7)
