@@ -408,3 +408,20 @@ def test_flash_route_actually_engages():
         assert calls["n"] == 1, f"SelfMultiheadAttn flash route not engaged ({calls['n']})"
     finally:
         tr.flash_attention = orig
+
+
+def test_self_mha_dropout_via_flash():
+    """SelfMultiheadAttn with dropout>0 in train mode routes through flash
+    (fused philox dropout) and produces finite grads; eval mode is exact."""
+    from apex_amd.contrib.fast_multihead_attn import SelfMultiheadAttn
+
+    torch.manual_seed(2)
+    mha = SelfMultiheadAttn(128, 2, dropout=0.3).cuda().bfloat16().train()
+    x = torch.randn(64, 2, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y, _ = mha(x, attn_mask="causal")
+    y.sum().backward()
+    assert torch.isfinite(x.grad).all()
+    mha.eval()
+    y1, _ = mha(x, attn_mask="causal", is_training=False)
+    y2, _ = mha(x, attn_mask="causal", is_training=False)
+    assert torch.equal(y1.detach(), y2.detach())
