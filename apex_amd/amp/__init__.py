@@ -506,9 +506,18 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
             fp32_grads = [p.grad for p in stash.all_fp32_from_fp32_params if p.grad is not None]
             if fp32_grads:
                 scaler.unscale_grads(fp32_grads, fp32_grads, scale_override=loss_scale)
-        else:  # O0/O1: unscale in place
+        else:  # O0/O1/O3: unscale in place — SPLIT BY DTYPE: with
+            # keep_batchnorm_fp32 the grad list mixes bf16/fp16 model grads
+            # with fp32 BN grads, and one multi_tensor_scale launch
+            # dispatches on the first tensor's dtype (a mixed list misreads
+            # the rest -> permanent overflow -> every step skipped; caught
+            # by the round-2 full L1 sweep at O3+keep_bn_fp32)
             grads = [p.grad for p in itertools.chain(*[g["params"] for g in opt.param_groups]) if p.grad is not None]
-            scaler.unscale_grads(grads, grads)
+            by_dtype = {}
+            for g in grads:
+                by_dtype.setdefault(g.dtype, []).append(g)
+            for gs in by_dtype.values():
+                scaler.unscale_grads(gs, gs)
     overflow = scaler.finish_unscale()
 
     if overflow:

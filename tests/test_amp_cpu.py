@@ -276,3 +276,36 @@ def test_overflow_check_off_static_scale():
                for p in stash.all_fp32_from_fp16_params)
     assert not getattr(opt, "_amp_skip_next_step", False)
     opt.step()
+
+
+def test_o3_keep_bn_fp32_mixed_dtype_unscale():
+    """O3 + keep_batchnorm_fp32=True + static scale: the grad list mixes
+    bf16 (conv) and fp32 (BN) grads — unscale must split by dtype (a single
+    mixed multi_tensor_scale launch misread the fp32 grads and skipped
+    every step; round-2 full L1 sweep catch)."""
+    import torch
+    from apex_amd import amp
+    from apex_amd.amp._amp_state import _amp_state
+
+    _amp_state.reset()
+    model = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3, padding=1),
+                                torch.nn.BatchNorm2d(8), torch.nn.ReLU(),
+                                torch.nn.Flatten(), torch.nn.Linear(8 * 8 * 8, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    model, opt = amp.initialize(model, opt, opt_level="O3",
+                                cast_model_type=torch.bfloat16,
+                                keep_batchnorm_fp32=True, loss_scale=128.0,
+                                verbosity=0)
+    x = torch.randn(4, 3, 8, 8)
+    y = torch.randint(0, 4, (4,))
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x).float(), y)
+        with amp.scale_loss(loss, opt) as scaled:
+            scaled.backward()
+        # grads must be unscaled (no 128x inflation) and steps must not skip
+        assert not getattr(opt, "_amp_skip_next_step", False)
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], f"no learning: {losses}"
