@@ -21,10 +21,21 @@ def clip_grad_norm_(parameters, max_norm, norm_type=2.0, error_if_nonfinite=Fals
         return torch.tensor(0.0)
     device = grads[0].device
 
+    # multi-tensor launches dispatch on the FIRST tensor's dtype — group a
+    # mixed fp32/bf16/fp16 grad list per dtype (same hazard as the amp O3 +
+    # keep_batchnorm_fp32 unscale fixed in round 2)
+    by_dtype = {}
+    for g in grads:
+        by_dtype.setdefault(g.dtype, []).append(g)
+
     if device.type == "cuda" and norm_type == 2.0 and all(g.device == device for g in grads):
         amp_C = get_ext("amp_C")
         noop = torch.zeros(1, dtype=torch.int32, device=device)
-        total_norm = multi_tensor_applier(amp_C.multi_tensor_l2norm, noop, [grads], False)[0].squeeze()
+        partial = [
+            multi_tensor_applier(amp_C.multi_tensor_l2norm, noop, [gs], False)[0].squeeze()
+            for gs in by_dtype.values()
+        ]
+        total_norm = torch.norm(torch.stack(partial)) if len(partial) > 1 else partial[0]
     else:
         if norm_type == float("inf"):
             total_norm = max(g.detach().abs().max().to(device) for g in grads)
@@ -44,7 +55,8 @@ def clip_grad_norm_(parameters, max_norm, norm_type=2.0, error_if_nonfinite=Fals
         if device.type == "cuda" and norm_type == 2.0:
             amp_C = get_ext("amp_C")
             noop = torch.zeros(1, dtype=torch.int32, device=device)
-            multi_tensor_applier(amp_C.multi_tensor_scale, noop, [grads, grads], float(clip_coef))
+            for gs in by_dtype.values():
+                multi_tensor_applier(amp_C.multi_tensor_scale, noop, [gs, gs], float(clip_coef))
         else:
             for g in grads:
                 g.detach().mul_(clip_coef.to(g.device))

@@ -167,3 +167,25 @@ def test_permutation_scores_kernel_matches_torch():
         g = wp.reshape(rows, parts.shape[0], 2, m)
         ref = g.topk(2, dim=3).values.sum(dim=(0, 2, 3))
         torch.testing.assert_close(scores[k], ref, rtol=1e-4, atol=1e-3)
+
+
+def test_clip_grad_mixed_dtypes_gpu():
+    """clip_grad_norm_ over a MIXED bf16+fp32 grad list (the conv+BN case)
+    must match torch's reference clip (per-dtype multi-tensor grouping)."""
+    from apex_amd.contrib.clip_grad import clip_grad_norm_
+
+    torch.manual_seed(0)
+    ps = [torch.randn(256, device="cuda", dtype=torch.bfloat16, requires_grad=True),
+          torch.randn(128, device="cuda", dtype=torch.float32, requires_grad=True),
+          torch.randn(64, device="cuda", dtype=torch.bfloat16, requires_grad=True)]
+    for p in ps:
+        p.grad = torch.randn_like(p) * 3.0
+    ref_grads = [p.grad.clone() for p in ps]
+
+    total = clip_grad_norm_(ps, max_norm=1.0)
+    ref_total = torch.norm(torch.stack([g.float().norm() for g in ref_grads]))
+    torch.testing.assert_close(total.float(), ref_total, rtol=1e-2, atol=1e-3)
+    coef = 1.0 / (ref_total + 1e-6)
+    for p, g0 in zip(ps, ref_grads):
+        torch.testing.assert_close(p.grad.float(), (g0.float() * coef),
+                                   rtol=1e-2, atol=1e-3)
