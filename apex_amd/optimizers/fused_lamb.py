@@ -49,7 +49,10 @@ class FusedLAMB(torch.optim.Optimizer):
         self._dummy_overflow_buf = None
         # capturable: lr/step live on device and bias corrections are
         # computed in-kernel (multi_tensor_lamb_capturable), so the whole
-        # step records into a hipGraph and replays with an advancing step
+        # step records into a hipGraph and replays with an advancing step.
+        # Note: graph REPLAYS advance only the device counter (_step_t);
+        # read it (not group["step"]) for the true step count, and update
+        # _lr_t in place for lr schedules under capture.
         self.capturable = capturable
         self._lr_t = None
         self._step_t = None
