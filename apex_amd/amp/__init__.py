@@ -518,7 +518,13 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
                 by_dtype.setdefault(g.dtype, []).append(g)
             for gs in by_dtype.values():
                 scaler.unscale_grads(gs, gs)
-    overflow = scaler.finish_unscale()
+    if _amp_state.opt_properties.overflow_check:
+        overflow = scaler.finish_unscale()
+    else:
+        # overflow_check=False (static scale): no device-to-host read at
+        # all — the iteration stays async / hipGraph-capturable
+        scaler._in_iteration = False
+        overflow = False
 
     if overflow:
         for opt in opt_list:

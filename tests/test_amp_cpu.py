@@ -309,3 +309,30 @@ def test_o3_keep_bn_fp32_mixed_dtype_unscale():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0], f"no learning: {losses}"
+
+
+def test_overflow_check_off_static_128():
+    """overflow_check=False with static scale 128: grads correctly unscaled,
+    never skipped, no finish_unscale host read."""
+    import torch
+    from apex_amd import amp
+    from apex_amd.amp._amp_state import _amp_state
+
+    _amp_state.reset()
+    model = torch.nn.Linear(8, 8)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    model, opt = amp.initialize(model, opt, opt_level="O0", loss_scale=128.0,
+                                verbosity=0, overflow_check=False)
+    x = torch.randn(4, 8)
+    ref = {id(p): None for p in model.parameters()}
+    loss = model(x).pow(2).mean()
+    loss.backward()
+    expected = [p.grad.clone() for p in model.parameters()]
+    for p in model.parameters():
+        p.grad = None
+    loss = model(x).pow(2).mean()
+    with amp.scale_loss(loss, opt) as scaled:
+        scaled.backward()
+    for p, e in zip(model.parameters(), expected):
+        torch.testing.assert_close(p.grad, e, rtol=1e-5, atol=1e-6)
+    assert not getattr(opt, "_amp_skip_next_step", False)
