@@ -79,7 +79,8 @@ class _Conv1x1BiasReLU(torch.autograd.Function):
         s = (stride, stride) if isinstance(stride, int) else tuple(stride)
         h, w = -(-H // s[0]), -(-W // s[1])
         if s != (1, 1):
-            gx = dy2d.new_zeros((n, ci, H, W)).to(memory_format=torch.channels_last)
+            gx = torch.empty(n, ci, H, W, dtype=dy2d.dtype, device=dy2d.device,
+                             memory_format=torch.channels_last).zero_()
             gx[:, :, ::s[0], ::s[1]] = _from_gemm_view(dx2d, n, h, w, ci)
         else:
             gx = _from_gemm_view(dx2d, n, h, w, ci).contiguous(
