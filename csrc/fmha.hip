@@ -1,18 +1,25 @@
-// apex_amd._mfma fmha_fwd — flash-attention forward for gfx950 (EXPERIMENTAL:
-// compile-checked + unit-tested against the eager composition; not yet wired
-// into the model hot path — see ROADMAP round-2 perf plan).
+// apex_amd._mfma — flash attention for gfx950 (hardware-validated round 2;
+// wired into the transformer models and contrib MHA modules).
 //
 // One 256-thread workgroup per (batch*head, 64 query rows); each wave64 owns
-// 16 query rows independently (no cross-wave barriers — causal waves finish
-// their shorter KV loops early). Per 32-wide KV tile:
-//   S = Q K^T via mfma_f32_16x16x32_bf16 (Q held as A-fragments in registers
-//       for the whole row block; K loaded straight from HBM in B-fragment
-//       order — both are 16-byte vector loads, no LDS staging),
+// 16 query rows. Per 32-wide KV tile:
+//   the K/V (fwd: V; dq: K; dkv: Q+dO) tile is staged ONCE per workgroup
+//       into a DOUBLE-BUFFERED shared LDS tile (block-cooperative,
+//       vectorized both sides; the next tile prefetches during the current
+//       tile's MFMA work, block barriers order the buffers),
+//   S = Q K^T via mfma_f32_16x16x32_bf16 (Q held as A-fragments in
+//       registers for the whole row block),
 //   online softmax (row max/sum via 4-step shfl_xor over the 16 column
 //       lanes; m/l replicated across those lanes),
+//   optional fused philox attention dropout (counter = flat (bh, q, kv)
+//       index; the backward kernels regenerate the identical mask),
 //   P V via one MFMA per 16 head-dim columns, with P transposed from the
 //       D-fragment to the A-fragment layout through a 1 KB per-wave LDS
-//       bounce (wave-local: s_waitcnt instead of a block barrier).
+//       bounce (wave-local: lgkm-only s_waitcnt, so the bounce does not
+//       serialize against the prefetch's vector-memory loads).
+// Inputs are read through (b, h, s) strides with D contiguous, so BSHD
+// views of a packed QKV projection pass with no .contiguous() copies;
+// cross-attention (Sq != Skv) is supported (causal requires Sq == Skv).
 //
 // Fragment layouts (verified on-device by mfma_tile_probe):
 //   A[r][k]: lane l -> r = l%16, k = (l/16)*8 + j
