@@ -502,6 +502,10 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         if self.capturable and self.device.type == "cuda":
             self._step_capturable(grad_scaler)
             return loss
+        if self.capturable:
+            # capturable ctor on a CPU model: the host path below runs; keep
+            # the device counter in sync so state_dict stays correct
+            self._step_t.fill_(self._step + 1)
 
         if grad_scaler is not None:
             inv_scale = float(grad_scaler._get_scale_async().double().reciprocal())

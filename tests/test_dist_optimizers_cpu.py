@@ -502,3 +502,20 @@ def test_dist_adam_found_inf_skips_step():
     assert opt._step == 1
     changed = any(not torch.equal(p.detach(), b) for p, b in zip(model.parameters(), before))
     assert changed
+
+
+def test_dist_adam_capturable_cpu_state_dict():
+    """capturable=True constructed on CPU: the host step path keeps the
+    device counter in sync, so state_dict reports the true step."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    model = _make_model()
+    opt = DistributedFusedAdam(model.parameters(), lr=1e-3, bucket_cap_mb=1,
+                               overlap_grad_sync=False, capturable=True)
+    for _ in range(3):
+        for p in model.parameters():
+            p.grad = torch.randn_like(p)
+            opt._grad_copy(p)
+        opt.step()
+    sd = opt.state_dict()
+    assert sd["step"] == 3
