@@ -153,6 +153,11 @@ class FlashAttentionFunction(torch.autograd.Function):
 
 def flash_attention(q, k, v, causal=False, scale=None, dropout_p=0.0):
     """Differentiable flash attention; returns the attention output.
-    ``dropout_p`` applies fused philox attention dropout (GPU only)."""
+    ``dropout_p`` applies fused philox attention dropout (GPU only).
+
+    hipGraph note: the philox seed is drawn per FORWARD on the host, so a
+    captured graph bakes one mask and replays it. Training steps that need
+    fresh attention-dropout masks per iteration must run the attention
+    eagerly (or re-capture); the bundled bench configs use dropout 0."""
     out, _ = FlashAttentionFunction.apply(q, k, v, causal, scale, dropout_p)
     return out
