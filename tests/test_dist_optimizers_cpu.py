@@ -519,3 +519,41 @@ def test_dist_adam_capturable_cpu_state_dict():
         opt.step()
     sd = opt.state_dict()
     assert sd["step"] == 3
+
+
+def _lamb_flags_worker(rank, world_size):
+    """DistributedFusedLAMB surface knobs on gloo: clip_after_ar=False
+    (pre-AR norm), set_global_scale, complete_reductions — training still
+    converges and matches the clip_after_ar=True trajectory when grads are
+    identical across ranks (pre/post-AR norms coincide then)."""
+    from apex_amd.contrib.optimizers import DistributedFusedLAMB
+
+    torch.manual_seed(77)
+    m1 = _make_model(seed=9)
+    m2 = _make_model(seed=9)
+    o1 = DistributedFusedLAMB(m1.parameters(), lr=1e-2, bucket_cap_mb=1,
+                              clip_after_ar=True, max_grad_norm=0.5,
+                              overlap_grad_sync=False)
+    o2 = DistributedFusedLAMB(m2.parameters(), lr=1e-2, bucket_cap_mb=1,
+                              clip_after_ar=False, max_grad_norm=0.5,
+                              overlap_grad_sync=False)
+    o2.set_global_scale(1.0)
+    for it in range(4):
+        # identical grads on every rank: pre-AR and post-AR norms agree
+        for i, (p1, p2) in enumerate(zip(m1.parameters(), m2.parameters())):
+            torch.manual_seed(500 + it * 13 + i)
+            g = torch.randn_like(p1)
+            p1.grad = g.clone()
+            p2.grad = g.clone()
+            o1._grad_copy(p1)
+            o2._grad_copy(p2)
+        o1.complete_reductions()
+        o2.complete_reductions()
+        o1.step()
+        o2.step()
+        for p1, p2 in zip(m1.parameters(), m2.parameters()):
+            torch.testing.assert_close(p1.detach(), p2.detach(), rtol=1e-4, atol=1e-5)
+
+
+def test_dist_lamb_clip_before_ar_flags():
+    run_distributed(_lamb_flags_worker, world_size=2)
