@@ -557,3 +557,35 @@ def _lamb_flags_worker(rank, world_size):
 
 def test_dist_lamb_clip_before_ar_flags():
     run_distributed(_lamb_flags_worker, world_size=2)
+
+
+def _gsd_worker(rank, world_size):
+    """grad_sync_dtype=bf16: gradients communicate in bf16 (halved traffic)
+    while the optimizer state and update stay fp32 — trajectory tracks the
+    fp32-comm run within bf16 rounding."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    m1 = _make_model(seed=21)
+    m2 = _make_model(seed=21)
+    o1 = DistributedFusedAdam(m1.parameters(), lr=1e-3, bucket_cap_mb=1,
+                              overlap_grad_sync=False)
+    o2 = DistributedFusedAdam(m2.parameters(), lr=1e-3, bucket_cap_mb=1,
+                              overlap_grad_sync=False,
+                              grad_sync_dtype=torch.bfloat16)
+    assert o2.buckets[0].grad_data.dtype == torch.bfloat16
+    for it in range(4):
+        for i, (p1, p2) in enumerate(zip(m1.parameters(), m2.parameters())):
+            torch.manual_seed(900 + it * 7 + i)
+            g = torch.randn_like(p1) * 0.1
+            p1.grad = g.clone()
+            p2.grad = g.clone()
+            o1._grad_copy(p1)
+            o2._grad_copy(p2)
+        o1.step()
+        o2.step()
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        torch.testing.assert_close(p1.detach(), p2.detach(), rtol=3e-2, atol=3e-3)
+
+
+def test_dist_adam_grad_sync_dtype_bf16():
+    run_distributed(_gsd_worker, world_size=2)
