@@ -336,3 +336,18 @@ def test_overflow_check_off_static_128():
     for p, e in zip(model.parameters(), expected):
         torch.testing.assert_close(p.grad, e, rtol=1e-5, atol=1e-6)
     assert not getattr(opt, "_amp_skip_next_step", False)
+
+
+def test_transformer_example_cpu_smoke():
+    """examples/transformer/train_flash.py --cpu runs end to end."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "examples", "transformer", "train_flash.py"),
+         "--cpu"],
+        cwd=repo, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "tokens/s" in out.stdout
