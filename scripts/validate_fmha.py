@@ -92,8 +92,6 @@ def main():
     print("ALL OK" if ok else "NUMERICS FAILURES — debug before wiring")
 
 
-if __name__ == "__main__":
-    main()
 
 
 def validate_round2_surface():
@@ -143,3 +141,6 @@ _base_main = main
 def main():  # noqa: F811
     _base_main()
     validate_round2_surface()
+
+if __name__ == "__main__":
+    main()
