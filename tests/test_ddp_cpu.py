@@ -73,6 +73,108 @@ def test_ddp_allreduce_always_fp32():
     run_distributed(_ddp_worker, world_size=2, args=({"allreduce_always_fp32": True, "message_size": 1},))
 
 
+def _trigger_worker(rank, world_size):
+    # allreduce_trigger_params: buckets flush ONLY when a trigger param's
+    # grad lands (reference ctor knob, ddp_race_condition_test.py:38-41)
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _model_ctor()
+    trigger = [list(model.parameters())[0]]  # last grad produced in backward
+    ddp = DDP(model, allreduce_trigger_params=trigger)
+    torch.manual_seed(7 + rank)
+    x = torch.randn(4, 8)
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x)
+    ddp(x).sum().backward()
+    expected = _expected_avg_grad(xs, _model_ctor)
+    for a, e in zip([p.grad for p in ddp.module.parameters()], expected):
+        torch.testing.assert_close(a, e, rtol=1e-5, atol=1e-6)
+
+
+def test_ddp_trigger_params():
+    run_distributed(_trigger_worker, world_size=2)
+
+
+def _retain_buffers_worker(rank, world_size):
+    # retain_allreduce_buffers keeps the reduced flat buckets on the module
+    # (reference: retained buffers become the master grads for amp)
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _model_ctor()
+    ddp = DDP(model, message_size=1, retain_allreduce_buffers=True)
+    torch.manual_seed(7 + rank)
+    x = torch.randn(4, 8)
+    ddp(x).sum().backward()
+    assert len(ddp.allreduce_buffers) >= 1
+    n_buf = sum(f.numel() for f in ddp.allreduce_buffers)
+    n_par = sum(p.numel() for p in ddp.module.parameters())
+    assert n_buf == n_par
+    # the flat buffers hold exactly the reduced grads that were copied back
+    flat_grads = torch.cat([p.grad.reshape(-1) for p in ddp.module.parameters()])
+    flat_bufs = torch.cat([f.reshape(-1) for f in ddp.allreduce_buffers])
+    torch.testing.assert_close(torch.sort(flat_bufs).values,
+                               torch.sort(flat_grads).values)
+
+
+def test_ddp_retain_allreduce_buffers():
+    run_distributed(_retain_buffers_worker, world_size=2)
+
+
+def _no_average_worker(rank, world_size):
+    # gradient_average=False: grads are SUMMED across ranks, not averaged
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _model_ctor()
+    ddp = DDP(model, message_size=1, gradient_average=False)
+    torch.manual_seed(7 + rank)
+    x = torch.randn(4, 8)
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x)
+    ddp(x).sum().backward()
+    expected = [g * world_size for g in _expected_avg_grad(xs, _model_ctor)]
+    for a, e in zip([p.grad for p in ddp.module.parameters()], expected):
+        torch.testing.assert_close(a, e, rtol=1e-5, atol=1e-6)
+
+
+def test_ddp_gradient_average_false():
+    run_distributed(_no_average_worker, world_size=2)
+
+
+class _MixedDtypeModel(torch.nn.Module):
+    # fp32 + bf16 params in one module: _flush_bucket must split buckets by
+    # grad dtype (flatten() cannot mix dtypes)
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(42)
+        self.a = torch.nn.Linear(8, 8)
+        self.b = torch.nn.Linear(8, 2).to(torch.bfloat16)
+
+    def forward(self, x):
+        return self.b(self.a(x).to(torch.bfloat16)).float()
+
+
+def _mixed_dtype_worker(rank, world_size):
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    model = _MixedDtypeModel()
+    ddp = DDP(model, message_size=1)
+    torch.manual_seed(7 + rank)
+    x = torch.randn(4, 8)
+    xs = [torch.empty_like(x) for _ in range(world_size)]
+    dist.all_gather(xs, x)
+    ddp(x).sum().backward()
+    expected = _expected_avg_grad(xs, _MixedDtypeModel)
+    for a, e in zip([p.grad for p in ddp.module.parameters()], expected):
+        assert a.dtype == e.dtype
+        tol = dict(rtol=2e-2, atol=2e-2) if a.dtype == torch.bfloat16 else \
+            dict(rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(a, e, **tol)
+
+
+def test_ddp_mixed_dtype_buckets():
+    run_distributed(_mixed_dtype_worker, world_size=2)
+
+
 def _param_sync_worker(rank, world_size):
     from apex_amd.parallel import DistributedDataParallel as DDP
 
