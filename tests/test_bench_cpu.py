@@ -32,3 +32,25 @@ def test_bench_cpu_gloo_two_ranks():
     for key in ["metric", "unit", "steps", "warmup", "ms_per_step", "higher_is_better",
                 "scaling", "vs_baseline", "dtype", "data", "config"]:
         assert key in rec
+
+
+@pytest.mark.timeout(600)
+def test_bench_cpu_gloo_two_ranks_bert():
+    # transformer path at world>1: amp O2 bf16 + DDP bucketed all-reduce +
+    # FusedAdam (eager — capture is disabled when distributed)
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node", "2",
+        "--master-addr", "127.0.0.1", "--master-port", "29612",
+        "bench.py", "--device", "cpu", "--model", "bert", "--seq-len", "64",
+        "--batch", "2", "--steps", "2", "--warmup", "1", "--skip-adam-bench",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=560)
+    assert out.returncode == 0, f"bench failed:\n{out.stdout[-2000:]}\n{out.stderr[-2000:]}"
+    json_lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["value"] > 0
+    assert rec["config"]["model"] == "bert-base"
+    assert rec["config"]["hipgraph_step"] is False
