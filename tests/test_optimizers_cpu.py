@@ -187,3 +187,56 @@ def test_optimizer_state_dict_roundtrip():
     # states should evolve identically after load (params started equal)
     for a, b in zip(ps_a, ps_b):
         torch.testing.assert_close(opt_a.state[a]["exp_avg"], opt_b.state[b]["exp_avg"])
+
+
+def test_fused_mixed_precision_lamb_masters_track_fp32():
+    # bf16 params + reduced_precision_dtype=bf16 keep fp32 masters; given
+    # identical fp32 grads the master trajectory must match a pure-fp32 run,
+    # and the model params must equal the masters cast down each step
+    # (reference apex/optimizers/fused_mixed_precision_lamb.py semantics)
+    from apex_amd.optimizers import FusedMixedPrecisionLamb
+
+    torch.manual_seed(0)
+    base = [torch.randn(7, 5), torch.randn(11)]
+    p32 = [b.clone().requires_grad_(True) for b in base]
+    pbf = [b.clone().bfloat16().requires_grad_(True) for b in base]
+
+    o32 = FusedMixedPrecisionLamb(p32, lr=1e-2, weight_decay=0.01)
+    obf = FusedMixedPrecisionLamb(pbf, lr=1e-2, weight_decay=0.01,
+                                  reduced_precision_dtype=torch.bfloat16)
+    masters = obf.param_groups_full_precision[0]["params"]
+    assert all(m is not None and m.dtype == torch.float32 for m in masters)
+
+    for i in range(5):
+        torch.manual_seed(100 + i)
+        for a, b in zip(p32, pbf):
+            g = torch.randn_like(a)
+            a.grad = g.clone()
+            b.grad = g.clone().bfloat16()
+        o32.step()
+        obf.step()
+
+    assert int(o32.param_groups[0]["step"].item()) == 5
+    for m, p in zip(masters, pbf):
+        torch.testing.assert_close(p.detach(), m.to(torch.bfloat16))
+    # masters track the fp32 run up to bf16 gradient rounding
+    for a, m in zip(p32, masters):
+        torch.testing.assert_close(a.detach(), m, rtol=3e-2, atol=3e-2)
+
+
+def test_fused_mixed_precision_lamb_converges():
+    from apex_amd.optimizers import FusedMixedPrecisionLamb
+
+    torch.manual_seed(3)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Tanh(),
+                                torch.nn.Linear(32, 1))
+    opt = FusedMixedPrecisionLamb(model.parameters(), lr=5e-2)
+    x, y = torch.randn(64, 16), torch.randn(64, 1)
+    losses = []
+    for _ in range(50):
+        opt.zero_grad()
+        loss = torch.nn.functional.mse_loss(model(x), y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0] * 0.7
