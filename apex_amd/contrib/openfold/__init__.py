@@ -9,6 +9,7 @@ the same capabilities are served by the library's HIP kernels):
 
 from .fused_adam_swa import FusedAdamSWA
 from .layer_norm import LayerNormSmallShapeOptImpl
-from .mha import AttnTri
+from .mha import AttnBiasJIT, AttnNoBiasJIT, AttnTri, CanSchTriMHA
 
-__all__ = ["FusedAdamSWA", "LayerNormSmallShapeOptImpl", "AttnTri"]
+__all__ = ["FusedAdamSWA", "LayerNormSmallShapeOptImpl", "AttnTri",
+           "AttnBiasJIT", "AttnNoBiasJIT", "CanSchTriMHA"]

@@ -1,11 +1,61 @@
-"""OpenFold multi-head attention (reference: openfold_triton's AttnTri),
-composed from the library's fused softmax and hipBLASLt GEMMs."""
+"""OpenFold multi-head attention (reference: openfold_triton's AttnTri /
+AttnBiasJIT / AttnNoBiasJIT / CanSchTriMHA, mha.py:36-469), composed from
+the library's fused softmax and hipBLASLt GEMMs."""
 
 import math
 
 import torch
 
 from ...transformer import scaled_masked_softmax
+
+_MHA_ENABLED = True
+
+
+def enable():
+    global _MHA_ENABLED
+    _MHA_ENABLED = True
+
+
+def disable():
+    global _MHA_ENABLED
+    _MHA_ENABLED = False
+
+
+def CanSchTriMHA(in_shape, has_bias=True, inf=1e9, training=True):
+    """Accelerated-path predicate (reference mha.py:36: a hard-coded list of
+    OpenFold shapes the Triton kernel was tuned for). The MI355X composition
+    is shape-generic — any [*, Q|S, H, S, D] attention runs on the fused
+    wave64 softmax kernels — so this answers only the semantic constraints:
+    the module-level enable()/disable() switch and the reference's
+    ``inf == 1e9`` contract (the fused kernel folds the mask additively)."""
+    if not _MHA_ENABLED:
+        return False
+    if inf != 1e9:
+        return False
+    return len(in_shape) >= 4
+
+
+def _attention_bias(query, key, value, mask, bias, inf):
+    """Eager reference attention, OpenFold float-mask convention
+    (mask 1.0 = keep, 0.0 = drop; additive -inf fold)."""
+    scale = 1.0 / math.sqrt(query.size(-1))
+    logits = torch.matmul(query * scale, torch.swapdims(key, -2, -1))
+    logits = logits + (mask - 1.0) * inf
+    if bias is not None:
+        logits = logits + bias
+    return torch.matmul(torch.softmax(logits, dim=-1), value)
+
+
+def _attention_no_bias(query, key, value, mask, inf):
+    return _attention_bias(query, key, value, mask, None, inf)
+
+
+try:  # compiled variants (reference: AttnBiasJIT/AttnNoBiasJIT, mha.py:468)
+    AttnBiasJIT = torch.compile(_attention_bias)
+    AttnNoBiasJIT = torch.compile(_attention_no_bias)
+except Exception:  # pragma: no cover - inductor unavailable
+    AttnBiasJIT = _attention_bias
+    AttnNoBiasJIT = _attention_no_bias
 
 
 def AttnTri(q, k, v, mask, bias=None, inf=1e9):

@@ -92,3 +92,40 @@ def test_fused_adam_swa_from_optim():
         torch.testing.assert_close(s, 0.9 * b + 0.1 * p.detach(), rtol=1e-5, atol=1e-6)
     for c, p in zip(bf16, fp32):  # compute copy refreshed
         torch.testing.assert_close(c, p.detach().to(torch.bfloat16))
+
+
+def test_openfold_attn_jit_variants_and_predicate():
+    # reference openfold_triton surface: AttnBiasJIT / AttnNoBiasJIT /
+    # CanSchTriMHA / enable-disable, and the alias module names
+    from apex_amd.contrib.openfold import (
+        AttnBiasJIT, AttnNoBiasJIT, AttnTri, CanSchTriMHA)
+    from apex_amd.contrib.openfold import mha as mha_mod
+    import apex_amd.contrib.openfold_triton as alias
+    import apex_amd.contrib.cudnn_gbn as gbn_alias
+
+    assert alias.AttnTri is AttnTri
+    assert gbn_alias.GroupBatchNorm2d is not None
+
+    torch.manual_seed(2)
+    b, h, s, d = 2, 3, 8, 16
+    q, k, v = (torch.randn(b, h, s, d) for _ in range(3))
+    mask = (torch.rand(b, 1, 1, s) > 0.2).float()  # 1.0 keep / 0.0 drop
+    bias = torch.randn(1, h, s, s)
+
+    out_b = AttnBiasJIT(q, k, v, mask, bias, 1e9)
+    out_nb = AttnNoBiasJIT(q, k, v, mask, 1e9)
+    # eager reference: standard softmax attention with additive mask fold
+    scale = 1.0 / math.sqrt(d)
+    logits = q @ k.transpose(-2, -1) * scale + (mask - 1.0) * 1e9
+    torch.testing.assert_close(
+        out_nb, torch.softmax(logits, -1) @ v, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(
+        out_b, torch.softmax(logits + bias, -1) @ v, rtol=1e-4, atol=1e-5)
+
+    assert CanSchTriMHA([1, 128, 8, 256, 32])
+    assert not CanSchTriMHA([1, 128, 8, 256, 32], inf=1e4)
+    mha_mod.disable()
+    try:
+        assert not CanSchTriMHA([1, 128, 8, 256, 32])
+    finally:
+        mha_mod.enable()
