@@ -6,14 +6,15 @@ API parity with the reference ``apex.contrib.sparsity.ASP``
     ASP.prune_trained_model(model, optimizer)
 
 computes N:M masks for whitelisted layers (Linear/Conv) and monkey-patches
-``optimizer.step`` so masks are re-applied after every step. Channel
-permutation search (the reference's accuracy-preserving reordering) is not
-implemented yet in this round; ``allow_permutation=True`` warns and proceeds
-without permutation.
+``optimizer.step`` so masks are re-applied after every step. With
+``allow_permutation=True`` the accuracy-preserving channel permutation
+search (permutation_lib: greedy channel-swap + exhaustive stripe-pair, GPU
+kernel in csrc/permutation_search.hip) reorders input channels before the
+masks are computed, with FX-graph propagation to keep the model function
+unchanged.
 """
 
 import types
-import warnings
 
 import torch
 
