@@ -68,8 +68,8 @@ class FlashAttentionFunction(torch.autograd.Function):
         dV = P^T dO,  dP = dO V^T,  dS = P * (dP - D_i),
         dQ = dS K * scale,  dK = dS^T Q * scale.
     Only q-tile-sized intermediates are live (no S x S matrix), and each
-    tile's math is plain GEMMs (hipBLASLt on GPU). A fused MFMA backward is
-    the round-2 follow-up.
+    tile's math is plain GEMMs (hipBLASLt on GPU). The fused MFMA backward
+    (``use_fused_backward``, default) replaces this recompute path on GPU.
     """
 
     CHUNK = 256  # q rows recomputed per tile (CPU fallback)
