@@ -39,6 +39,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "LAMB stage 2 only (sharded: caller-reduced per-tensor norms)");
   m.def("multi_tensor_lamb_mp", &multi_tensor_lamb_mp_cuda,
         "Graph-safe LAMB with device-tensor state and fp32 masters");
+  // reference-name aliases (amp_C_frontend.cpp exports these with the
+  // _cuda suffix; downstream callers use either)
+  m.def("multi_tensor_lamb_stage1_cuda", &multi_tensor_lamb_stage1_cuda,
+        "alias of multi_tensor_lamb_stage1");
+  m.def("multi_tensor_lamb_stage2_cuda", &multi_tensor_lamb_stage2_cuda,
+        "alias of multi_tensor_lamb_stage2");
   m.def("update_scale_hysteresis", &update_scale_hysteresis_cuda,
         "On-device dynamic loss-scale update with hysteresis");
 }
