@@ -79,6 +79,15 @@ class FusedLAMB(torch.optim.Optimizer):
 
         # Global grad norm over ALL groups (reference computes one per step).
         device = self.param_groups[0]["params"][0].device
+        if (self.capturable and len(self.param_groups) > 1
+                and device.type == "cuda"
+                and torch.cuda.is_current_stream_capturing()
+                and len({float(g["lr"]) for g in self.param_groups}) > 1):
+            # one shared _lr_t cannot carry per-group lrs inside a graph
+            # (eager steps refresh it per group, stream-ordered)
+            raise RuntimeError(
+                "capturable FusedLAMB: cannot capture with divergent "
+                "per-group lrs; use one lr or capture per group")
         g_all_16, g_all_32 = [], []
         for group in self.param_groups:
             for p in group["params"]:
@@ -138,6 +147,11 @@ class FusedLAMB(torch.optim.Optimizer):
                         self._lr_t = torch.full((1,), float(group["lr"]),
                                                 dtype=torch.float32, device=device)
                         self._step_t = torch.zeros(1, dtype=torch.int32, device=device)
+                    elif not torch.cuda.is_current_stream_capturing():
+                        # eager steps track group["lr"] (lr schedules); the
+                        # refresh is skipped DURING capture so replays read
+                        # whatever the user writes into _lr_t on device
+                        self._lr_t.fill_(float(group["lr"]))
                     if group is self.param_groups[0]:
                         self._step_t.add_(1)  # device op: advances per replay
                 for lists in [[g16, p16, m16, v16], [g32, p32, m32, v32]]:

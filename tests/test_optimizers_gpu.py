@@ -204,3 +204,38 @@ def test_fused_lamb_capturable_graph_replay():
     for p, rp in zip(model.parameters(), ref.parameters()):
         torch.testing.assert_close(p.detach(), rp.detach(), rtol=5e-4, atol=5e-5)
     assert int(opt._step_t.item()) == 5
+
+
+@pytest.mark.gpu
+def test_fused_lamb_capturable_eager_lr_schedule():
+    """Eager capturable steps must track group["lr"] changes (the device
+    _lr_t is refreshed per group outside capture); two groups with
+    different lrs each see their own lr."""
+    torch.manual_seed(0)
+    pa = torch.randn(64, device="cuda", requires_grad=True)
+    pb = torch.randn(64, device="cuda", requires_grad=True)
+    ref_a = pa.detach().clone().requires_grad_(True)
+    ref_b = pb.detach().clone().requires_grad_(True)
+
+    opt = FusedLAMB([{"params": [pa], "lr": 1e-2},
+                     {"params": [pb], "lr": 1e-3}],
+                    weight_decay=0.01, capturable=True)
+    ref = FusedLAMB([{"params": [ref_a], "lr": 1e-2},
+                     {"params": [ref_b], "lr": 1e-3}],
+                    weight_decay=0.01, capturable=False)
+
+    for i in range(3):
+        if i == 2:  # lr schedule tick
+            for o in (opt, ref):
+                o.param_groups[0]["lr"] = 5e-3
+                o.param_groups[1]["lr"] = 5e-4
+        torch.manual_seed(10 + i)
+        for p, r in ((pa, ref_a), (pb, ref_b)):
+            g = torch.randn_like(p)
+            p.grad, r.grad = g.clone(), g.clone()
+        opt.step()
+        ref.step()
+
+    torch.testing.assert_close(pa, ref_a, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(pb, ref_b, rtol=1e-5, atol=1e-6)
+    assert float(opt._lr_t) == 5e-4  # last group's eager refresh
