@@ -623,6 +623,14 @@ class DistributedFusedAdam(torch.optim.Optimizer):
         # step advances only on non-overflow iterations (device-side)
         self._step_t.add_(1 - self._noop)
 
+        if not torch.cuda.is_current_stream_capturing():
+            # eager capturable steps track group["lr"] (lr schedules);
+            # skipped DURING capture so replays read whatever the user
+            # writes into the device lr_t
+            for group in self.param_groups:
+                if "lr_t" in group:
+                    group["lr_t"].fill_(float(group["lr"]))
+
         for b in self.buckets:
             group = b.group
             beta1, beta2 = group["betas"]

@@ -162,3 +162,34 @@ def test_dist_adam_capturable_graph_replay():
         torch.testing.assert_close(p.detach(), rp.detach(), rtol=1e-4, atol=1e-5)
     # device step counter advanced once per iteration (1 eager + 1 captured + 3 replays)
     assert int(opt._step_t.item()) == 5
+
+
+@pytest.mark.gpu
+def test_dist_adam_capturable_eager_lr_schedule():
+    """Eager capturable steps must track group["lr"] (the device lr_t is
+    refreshed outside capture), matching an eager AdamW with the same
+    schedule."""
+    from apex_amd.contrib.optimizers import DistributedFusedAdam
+
+    torch.manual_seed(1)
+    p = torch.randn(256, device="cuda", requires_grad=True)
+    r = p.detach().clone().requires_grad_(True)
+    opt = DistributedFusedAdam([p], lr=1e-2, weight_decay=0.01,
+                               overlap_grad_sync=False, capturable=True)
+    ropt = torch.optim.AdamW([r], lr=1e-2, weight_decay=0.01)
+
+    for i in range(4):
+        if i == 2:  # schedule tick mid-run
+            opt.param_groups[0]["lr"] = 2.5e-3
+            for g in ropt.param_groups:
+                g["lr"] = 2.5e-3
+        torch.manual_seed(50 + i)
+        g = torch.randn_like(p)
+        p.grad = g.clone()
+        r.grad = g.clone()
+        opt._grad_copy(p)
+        opt.step()
+        ropt.step()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(p.detach(), r.detach(), rtol=1e-4, atol=1e-5)
+    assert float(opt.param_groups[0]["lr_t"]) == pytest.approx(2.5e-3, rel=1e-6)
