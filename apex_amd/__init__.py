@@ -25,6 +25,7 @@ from . import parallel  # noqa: F401
 from . import fused_dense  # noqa: F401
 from . import mlp  # noqa: F401
 from . import multi_tensor_apply  # noqa: F401
+from . import tracing  # noqa: F401
 from . import transformer  # noqa: F401
 from . import contrib  # noqa: F401
 

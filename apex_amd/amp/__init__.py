@@ -32,6 +32,8 @@ import contextlib
 import functools
 import itertools
 
+from ..tracing import trace_mark
+
 import torch
 
 from ._amp_state import _amp_state, maybe_print
@@ -491,6 +493,7 @@ def scale_loss(loss, optimizers, loss_id=0, model=None, delay_unscale=False):
     # bracket: all unscales below share one overflow flag; the dynamic scale
     # ticks exactly once per iteration in finish_unscale (regardless of how
     # many optimizers / grad sets were unscaled)
+    trace_mark("amp.unscale")
     scaler.begin_unscale()
     for opt in opt_list:
         if hasattr(opt, "_amp_stash"):  # O2: unscale fp16 grads into masters

@@ -17,6 +17,7 @@ import torch
 
 from .._ext import get_ext
 from ..multi_tensor_apply import multi_tensor_applier
+from ..tracing import traced
 
 
 class FusedAdam(torch.optim.Optimizer):
@@ -90,6 +91,7 @@ class FusedAdam(torch.optim.Optimizer):
         return self._dummy_overflow_buf
 
     @torch.no_grad()
+    @traced("FusedAdam.step")
     def step(self, closure=None, grads=None, output_params=None, scale=None, grad_norms=None, grad_scaler=None):
         if any(p is not None for p in [grads, output_params, scale, grad_norms]):
             raise RuntimeError(

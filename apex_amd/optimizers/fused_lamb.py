@@ -12,6 +12,7 @@ import torch
 
 from .._ext import get_ext
 from ..multi_tensor_apply import multi_tensor_applier
+from ..tracing import traced
 
 
 class FusedLAMB(torch.optim.Optimizer):
@@ -71,6 +72,7 @@ class FusedLAMB(torch.optim.Optimizer):
         return self._dummy_overflow_buf
 
     @torch.no_grad()
+    @traced("FusedLAMB.step")
     def step(self, closure=None):
         loss = None
         if closure is not None:

@@ -21,6 +21,7 @@ import torch
 import torch.distributed as dist
 
 from .._ext import get_ext
+from ..tracing import trace_mark
 
 
 def flatten(tensors):
@@ -176,6 +177,7 @@ class DistributedDataParallel(torch.nn.Module):
             self._allreduce_bucket(grads)
 
     def _allreduce_bucket(self, grads):
+        trace_mark(f"ddp.allreduce_bucket({len(grads)}t)")
         stream = None
         if self._use_streams:
             stream = self._streams[self._next_stream]

@@ -269,3 +269,28 @@ def test_rope_2d_cpu_invariants():
                                          ang_h.cos(), ang_h.sin(),
                                          ang_w.cos(), ang_w.sin())
     torch.testing.assert_close(out2.norm(dim=-1), t.norm(dim=-1), rtol=1e-5, atol=1e-5)
+
+
+def test_tracing_noop_and_enabled_paths(monkeypatch):
+    # APEX_TRACE unset -> pure no-op; enabled -> balanced push/pop via the
+    # torch.cuda.nvtx seam (roctx on ROCm)
+    import apex_amd.tracing as tr
+
+    monkeypatch.setattr(tr, "_enabled_cache", False)
+    with tr.trace_range("x"):
+        pass
+    tr.trace_mark("y")  # no-op, must not touch torch.cuda
+
+    calls = []
+    monkeypatch.setattr(tr, "_enabled_cache", True)
+    monkeypatch.setattr(torch.cuda.nvtx, "range_push", lambda n: calls.append(("push", n)))
+    monkeypatch.setattr(torch.cuda.nvtx, "range_pop", lambda: calls.append(("pop", None)))
+    monkeypatch.setattr(torch.cuda.nvtx, "mark", lambda n: calls.append(("mark", n)))
+    with tr.trace_range("fwd"):
+        tr.trace_mark("inner")
+    @tr.traced("step")
+    def f():
+        return 7
+    assert f() == 7
+    assert calls == [("push", "fwd"), ("mark", "inner"), ("pop", None),
+                     ("push", "step"), ("pop", None)]
