@@ -195,17 +195,21 @@ class DistributedFusedLAMB(DistributedFusedAdam):
 
         # per-tensor segment norms -> one cross-rank reduce -> trust ratios
         for ent in self._lamb_plan:
-            if not ent["seg_p"]:
-                continue
-            pn = multi_tensor_applier(amp_C.multi_tensor_l2norm, self._noop,
-                                      [ent["seg_p"]], True)[1]
-            un = multi_tensor_applier(amp_C.multi_tensor_l2norm, self._noop,
-                                      [ent["seg_u"]], True)[1]
             full = torch.zeros(self._lamb_n_tensors, 2, device=self.device)
-            full.index_put_((ent["idx_t"],),
-                            torch.stack([pn.pow(2), un.pow(2)], dim=1), accumulate=True)
+            if ent["seg_p"]:
+                pn = multi_tensor_applier(amp_C.multi_tensor_l2norm, self._noop,
+                                          [ent["seg_p"]], True)[1]
+                un = multi_tensor_applier(amp_C.multi_tensor_l2norm, self._noop,
+                                          [ent["seg_u"]], True)[1]
+                full.index_put_((ent["idx_t"],),
+                                torch.stack([pn.pow(2), un.pow(2)], dim=1), accumulate=True)
+            # every rank MUST join the reduce even with an all-padding shard
+            # (an empty seg list on one rank would desync the collective
+            # sequence and hang the group)
             if self.world_size > 1:
                 dist.all_reduce(full, group=self.process_group)
+            if not ent["seg_p"]:
+                continue
             norms = full.sqrt()
             seg_pn = norms.index_select(0, ent["idx_t"])[:, 0].contiguous()
             seg_un = norms.index_select(0, ent["idx_t"])[:, 1].contiguous()

@@ -589,3 +589,43 @@ def _gsd_worker(rank, world_size):
 
 def test_dist_adam_grad_sync_dtype_bf16():
     run_distributed(_gsd_worker, world_size=2)
+
+
+def _lamb_empty_shard_worker(rank, world_size):
+    # a 1-element param group pads its bucket to world_size, leaving rank 1's
+    # shard ALL padding: the lamb plan must stay rank-symmetric (same entry
+    # count on every rank) so the fused path's per-group all_reduce sequence
+    # matches — the empty-seg rank still joins the reduce (regression for the
+    # skip-collective hang) — and the step must still be correct end to end
+    from apex_amd.contrib.optimizers import DistributedFusedLAMB
+
+    torch.manual_seed(0)
+    w = torch.randn(6, 4, requires_grad=True)
+    scalar = torch.randn(1, requires_grad=True)
+    opt = DistributedFusedLAMB(
+        [{"params": [w], "lr": 1e-2}, {"params": [scalar], "lr": 1e-3}],
+        weight_decay=0.01, bucket_cap_mb=1)
+
+    opt._build_lamb_plan()
+    n_entries = torch.tensor([len(opt._lamb_plan)])
+    counts = [torch.empty_like(n_entries) for _ in range(world_size)]
+    dist.all_gather(counts, n_entries)
+    assert all(int(c) == int(counts[0]) for c in counts)
+    empty_here = any(not ent["seg_p"] for ent in opt._lamb_plan)
+    flags = [torch.zeros(1) for _ in range(world_size)]
+    dist.all_gather(flags, torch.tensor([1.0 if empty_here else 0.0]))
+    # the scenario is real: at least one rank holds an all-padding shard
+    assert sum(float(f) for f in flags) >= 1.0, "expected an empty-seg rank"
+
+    for it in range(3):
+        torch.manual_seed(10 + it)
+        for p in (w, scalar):
+            p.grad = torch.randn_like(p)
+        for p in (w, scalar):
+            opt._grad_copy(p)
+        opt.step()
+    assert torch.isfinite(w).all() and torch.isfinite(scalar).all()
+
+
+def test_dist_lamb_empty_shard_rank_symmetric():
+    run_distributed(_lamb_empty_shard_worker, world_size=2)
