@@ -8,8 +8,6 @@ RCCL all_gather over xGMI instead — same semantics, and latency-bound either
 way for C-length vectors.
 """
 
-import torch
-
 from ...parallel.sync_batchnorm import SyncBatchNorm, create_syncbn_process_group
 
 
