@@ -66,6 +66,20 @@ class FusedLAMB(torch.optim.Optimizer):
         else:
             super().zero_grad(set_to_none=False)
 
+    def state_dict(self):
+        if self.capturable and self._step_t is not None:
+            # graph replays advance only the device counter; sync the host
+            # per-group counts so checkpoints carry the true step
+            true_step = int(self._step_t.item())
+            for group in self.param_groups:
+                group["step"] = true_step
+        return super().state_dict()
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        if self.capturable and self._step_t is not None:
+            self._step_t.fill_(int(self.param_groups[0].get("step", 0)))
+
     def _noop_buf(self, device):
         if self._dummy_overflow_buf is None or self._dummy_overflow_buf.device != device:
             self._dummy_overflow_buf = torch.zeros(1, dtype=torch.int32, device=device)
@@ -148,7 +162,11 @@ class FusedLAMB(torch.optim.Optimizer):
                     if self._lr_t is None:
                         self._lr_t = torch.full((1,), float(group["lr"]),
                                                 dtype=torch.float32, device=device)
-                        self._step_t = torch.zeros(1, dtype=torch.int32, device=device)
+                        # resume-aware: group["step"] was already advanced
+                        # for THIS step above, so seed with step-1 and let
+                        # the add_(1) below land on the true count
+                        self._step_t = torch.full((1,), int(group["step"]) - 1,
+                                                  dtype=torch.int32, device=device)
                     elif not torch.cuda.is_current_stream_capturing():
                         # eager steps track group["lr"] (lr schedules); the
                         # refresh is skipped DURING capture so replays read

@@ -240,3 +240,29 @@ def test_fused_mixed_precision_lamb_converges():
         opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] * 0.7
+
+
+def test_fused_lamb_capturable_checkpoint_step_sync():
+    # graph replays advance only the device _step_t; state_dict must carry
+    # the true count and load must restore it (unit-level: _step_t stubbed
+    # on CPU — the cuda branch creates it device-side from group["step"]-1)
+    from apex_amd.optimizers import FusedLAMB
+
+    ps = [torch.randn(4, requires_grad=True)]
+    ps[0].grad = torch.randn(4)
+    opt = FusedLAMB(ps, lr=1e-3, capturable=True)
+    opt.step()  # CPU ref path: group["step"] = 1
+    opt._step_t = torch.tensor([7], dtype=torch.int32)  # pretend 6 replays
+    sd = opt.state_dict()
+    assert all(g["step"] == 7 for g in sd["param_groups"])
+
+    ps2 = [torch.randn(4, requires_grad=True)]
+    opt2 = FusedLAMB(ps2, lr=1e-3, capturable=True)
+    opt2._step_t = torch.tensor([0], dtype=torch.int32)
+    opt2.load_state_dict(sd)
+    assert int(opt2._step_t) == 7
+    # fresh optimizer without a live _step_t: the loaded host count is the
+    # seed for device-side creation (group["step"] round-tripped)
+    opt3 = FusedLAMB([torch.randn(4, requires_grad=True)], lr=1e-3, capturable=True)
+    opt3.load_state_dict(sd)
+    assert opt3.param_groups[0]["step"] == 7 and opt3._step_t is None
