@@ -85,6 +85,23 @@ class FusedAdam(torch.optim.Optimizer):
         else:
             super().zero_grad(set_to_none=False)
 
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        if not self.capturable:
+            return
+        # capturable kernels read lr/step from DEVICE pointers; a checkpoint
+        # loaded without map_location leaves the group tensors on CPU (and a
+        # CPU lr inside a later capture records a pageable H2D that faults
+        # on replay) — pin them to the params' device here
+        for group in self.param_groups:
+            if not group["params"]:
+                continue
+            device = group["params"][0].device
+            for key in ("lr", "step"):
+                v = group.get(key)
+                if torch.is_tensor(v) and v.device != device:
+                    group[key] = v.to(device)
+
     def _noop_buf(self, device):
         if self._dummy_overflow_buf is None or self._dummy_overflow_buf.device != device:
             self._dummy_overflow_buf = torch.zeros(1, dtype=torch.int32, device=device)

@@ -266,3 +266,24 @@ def test_fused_lamb_capturable_checkpoint_step_sync():
     opt3 = FusedLAMB([torch.randn(4, requires_grad=True)], lr=1e-3, capturable=True)
     opt3.load_state_dict(sd)
     assert opt3.param_groups[0]["step"] == 7 and opt3._step_t is None
+
+
+def test_fused_adam_capturable_load_normalizes_group_tensors():
+    # capturable FusedAdam keeps lr/step as tensors inside param_groups;
+    # after load_state_dict they must sit on the params' device (CPU here —
+    # the override is a no-op move, but must not crash and must keep values)
+    from apex_amd.optimizers import FusedAdam
+
+    ps = [torch.randn(4, requires_grad=True)]
+    ps[0].grad = torch.randn(4)
+    opt = FusedAdam(ps, lr=1e-3, capturable=True)
+    opt.step()
+    sd = opt.state_dict()
+    ps2 = [torch.randn(4, requires_grad=True)]
+    opt2 = FusedAdam(ps2, lr=1e-3, capturable=True)
+    opt2.load_state_dict(sd)
+    g = opt2.param_groups[0]
+    assert torch.is_tensor(g["lr"]) and g["lr"].device == ps2[0].device
+    assert float(g["lr"]) == pytest.approx(1e-3, rel=1e-6)
+    step = g["step"] if not torch.is_tensor(g["step"]) else int(g["step"].item())
+    assert int(step) == 1
