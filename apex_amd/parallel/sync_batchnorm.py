@@ -344,8 +344,11 @@ class SyncBatchNorm(_BatchNorm):
             else:
                 exponential_average_factor = self.momentum
 
-        if not self.training:
+        if not self.training and self.track_running_stats:
             # eval with channel_last: normalize with running stats manually
+            # (track_running_stats=False has no running stats — fall through
+            # to the Function, which normalizes by batch stats, as torch's
+            # _BatchNorm does when running stats are None)
             shape = [1] * input.dim()
             ax = input.dim() - 1 if channel_last else 1
             shape[ax] = self.num_features

@@ -196,3 +196,24 @@ def test_syncbn_channels_last_memory_format():
     torch.testing.assert_close(sbn.weight.grad, bn.weight.grad, rtol=1e-5, atol=1e-6)
     torch.testing.assert_close(sbn.running_mean, bn.running_mean, rtol=1e-5, atol=1e-6)
     torch.testing.assert_close(sbn.running_var, bn.running_var, rtol=1e-5, atol=1e-6)
+
+
+def test_syncbn_eval_no_running_stats_uses_batch_stats():
+    # track_running_stats=False in eval(): no running stats exist — torch's
+    # _BatchNorm normalizes by batch stats; SyncBatchNorm must match (it
+    # used to dereference running_var=None and crash)
+    from apex_amd.parallel import SyncBatchNorm
+
+    torch.manual_seed(0)
+    for channel_last in (False, True):
+        sbn = SyncBatchNorm(6, track_running_stats=False,
+                            channel_last=channel_last).eval()
+        ref = torch.nn.BatchNorm2d(6, track_running_stats=False).eval()
+        ref.load_state_dict({k: v for k, v in sbn.state_dict().items()})
+        x = torch.randn(4, 6, 5, 5) if not channel_last else torch.randn(4, 5, 5, 6)
+        out = sbn(x)
+        xr = x if not channel_last else x.permute(0, 3, 1, 2)
+        expected = ref(xr)
+        if channel_last:
+            expected = expected.permute(0, 2, 3, 1)
+        torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-5)
