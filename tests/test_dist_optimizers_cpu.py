@@ -20,10 +20,11 @@ def _set_rank_grads(model, rank, world, it):
     for i, p in enumerate(model.parameters()):
         torch.manual_seed(1000 * it + i)
         base = torch.randn_like(p)
-        # rank-dependent component that averages to zero across ranks
+        # rank-dependent component that averages to zero across ANY world
+        # size: coefficients rank - (world-1)/2 sum to zero
         delta = torch.randn_like(p)
-        sign = 1.0 if rank == 0 else -1.0
-        p.grad = base + sign * delta
+        coef = rank - (world - 1) / 2.0
+        p.grad = base + coef * delta
         mean_grads.append(base.clone())
     return mean_grads
 
@@ -629,3 +630,13 @@ def _lamb_empty_shard_worker(rank, world_size):
 
 def test_dist_lamb_empty_shard_rank_symmetric():
     run_distributed(_lamb_empty_shard_worker, world_size=2)
+
+
+def test_dist_adam_matches_adamw_world4():
+    # W=4 shards hit different padding/segment splits than W=2 (the driver's
+    # 8-GPU tier is the only larger-world execution — de-risk it here)
+    run_distributed(_dist_adam_worker, world_size=4, args=(True, False))
+
+
+def test_dist_lamb_matches_fused_lamb_world4():
+    run_distributed(_lamb_worker, world_size=4)
