@@ -263,3 +263,8 @@ def _no_sync_worker(rank, world_size):
 
 def test_ddp_no_sync():
     run_distributed(_no_sync_worker, world_size=2)
+
+
+def test_ddp_grad_average_world4():
+    # the driver's 8-GPU tier is the only larger-world run; cover W=4 here
+    run_distributed(_ddp_worker, world_size=4, args=({"message_size": 1},))
