@@ -217,3 +217,9 @@ def test_syncbn_eval_no_running_stats_uses_batch_stats():
         if channel_last:
             expected = expected.permute(0, 2, 3, 1)
         torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-5)
+
+
+def test_syncbn_vs_bn_world4():
+    # W=4 stat merge (Chan combine over 4 partials) vs plain BN on the
+    # concatenated batch
+    run_distributed(_syncbn_worker, world_size=4, args=(False,))
