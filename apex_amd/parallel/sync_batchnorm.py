@@ -152,7 +152,11 @@ class SyncBatchnormFunction(torch.autograd.Function):
                 mean_all = torch.stack([c_[:c] for c_ in combined_list])
                 var_all = torch.stack([c_[c:2 * c] for c_ in combined_list])
                 count_all = torch.stack([c_[2 * c:] for c_ in combined_list]).view(-1)
-                total_count = int(count * world_size)
+                # per-rank batch sizes may DIFFER (reference
+                # two_gpu_unit_test's uneven-batch rung): the true element
+                # count is the sum of gathered counts, and it feeds both the
+                # backward's mean_dy divisor and the unbiased-var EMA
+                total_count = int(count_all.sum().item())
                 if use_kernels:
                     rm = running_mean if fused_track else None
                     rv = running_var if fused_track else None

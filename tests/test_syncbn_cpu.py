@@ -223,3 +223,53 @@ def test_syncbn_vs_bn_world4():
     # W=4 stat merge (Chan combine over 4 partials) vs plain BN on the
     # concatenated batch
     run_distributed(_syncbn_worker, world_size=4, args=(False,))
+
+
+def _uneven_batch_worker(rank, world_size):
+    # per-rank batch sizes differ (reference two_gpu_unit_test's uneven-batch
+    # rung): stats must merge count-WEIGHTED, matching BN on the global batch
+    from apex_amd.parallel import SyncBatchNorm
+
+    C = 8
+    n_local = 3 + 2 * rank  # 3, 5
+    torch.manual_seed(20 + rank)
+    x = torch.randn(n_local, C, 4, 4, requires_grad=True)
+
+    sbn = SyncBatchNorm(C)
+    sbn.train()
+
+    sizes = [3 + 2 * r for r in range(world_size)]
+    pad = max(sizes)
+    xp = torch.zeros(pad, C, 4, 4)
+    xp[:n_local] = x.detach()
+    gathered = [torch.empty_like(xp) for _ in range(world_size)]
+    dist.all_gather(gathered, xp)
+    global_x = torch.cat([g[:s] for g, s in zip(gathered, sizes)], dim=0)
+    global_x = global_x.requires_grad_(True)
+
+    ref_bn = torch.nn.BatchNorm2d(C)
+    ref_bn.train()
+    with torch.no_grad():
+        ref_bn.weight.copy_(sbn.weight)
+        ref_bn.bias.copy_(sbn.bias)
+
+    out = sbn(x)
+    ref_global = ref_bn(global_x)
+    lo = sum(sizes[:rank])
+    torch.testing.assert_close(out, ref_global[lo:lo + n_local],
+                               rtol=1e-4, atol=1e-5)
+    # backward: grads must match the global-batch BN's slice
+    g = torch.ones_like(out)
+    out.backward(g)
+    ref_global.backward(torch.ones_like(ref_global))
+    torch.testing.assert_close(x.grad, global_x.grad[lo:lo + n_local],
+                               rtol=1e-4, atol=1e-5)
+    # running stats follow the global batch
+    torch.testing.assert_close(sbn.running_mean, ref_bn.running_mean,
+                               rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(sbn.running_var, ref_bn.running_var,
+                               rtol=1e-4, atol=1e-5)
+
+
+def test_syncbn_uneven_per_rank_batches():
+    run_distributed(_uneven_batch_worker, world_size=2)
