@@ -287,3 +287,23 @@ def test_fused_adam_capturable_load_normalizes_group_tensors():
     assert float(g["lr"]) == pytest.approx(1e-3, rel=1e-6)
     step = g["step"] if not torch.is_tensor(g["step"]) else int(g["step"].item())
     assert int(step) == 1
+
+
+def test_fused_optimizers_frozen_params():
+    # reference frozen-model test (L0 test_fused_optimizer.py:201): params
+    # with no grad are skipped, not zero-updated
+    from apex_amd.optimizers import FusedAdam, FusedLAMB, FusedSGD, FusedNovoGrad
+
+    for cls, kw in ((FusedAdam, {}), (FusedLAMB, {}), (FusedSGD, {"momentum": 0.9}),
+                    (FusedNovoGrad, {})):
+        torch.manual_seed(0)
+        live = torch.randn(6, requires_grad=True)
+        frozen = torch.randn(6, requires_grad=False)
+        frozen_before = frozen.detach().clone()
+        opt = cls([live, frozen], lr=1e-2, **kw)
+        for i in range(3):
+            torch.manual_seed(i)
+            live.grad = torch.randn(6)
+            opt.step()
+        assert torch.equal(frozen, frozen_before), cls.__name__
+        assert not torch.equal(live.detach(), frozen_before), cls.__name__
