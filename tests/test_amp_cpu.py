@@ -351,3 +351,46 @@ def test_transformer_example_cpu_smoke():
         cwd=repo, capture_output=True, text=True, timeout=420)
     assert out.returncode == 0, out.stderr[-1500:]
     assert "tokens/s" in out.stdout
+
+
+def _master_params_worker(rank, world_size):
+    # reference tests/distributed/amp_master_params: after O2+DDP training,
+    # fp32 masters agree across ranks and model params equal masters cast
+    import torch.distributed as dist
+    from apex_amd import amp
+    from apex_amd.parallel import DistributedDataParallel as DDP
+
+    torch.manual_seed(42)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Tanh(),
+                                torch.nn.Linear(16, 4))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    model, opt = amp.initialize(model, opt, opt_level="O2",
+                                cast_model_type=torch.bfloat16,
+                                loss_scale=128.0, verbosity=0)
+    ddp = DDP(model, message_size=1)
+    for it in range(3):
+        torch.manual_seed(100 + it * world_size + rank)
+        x = torch.randn(4, 8, dtype=torch.bfloat16)
+        ddp.zero_grad()
+        loss = ddp(x).float().pow(2).mean()
+        with amp.scale_loss(loss, opt) as scaled:
+            scaled.backward()
+        opt.step()
+
+    masters = [p.detach().clone() for p in amp.master_params(opt)]
+    flat = torch.cat([m.reshape(-1).float() for m in masters])
+    flats = [torch.empty_like(flat) for _ in range(world_size)]
+    dist.all_gather(flats, flat)
+    for f in flats[1:]:
+        torch.testing.assert_close(flats[0], f)  # masters identical across ranks
+    model_params = [p for p in ddp.module.parameters()]
+    n_bf16 = sum(1 for p in model_params if p.dtype == torch.bfloat16)
+    assert n_bf16 == len(model_params)
+    for p, m in zip(model_params, masters):
+        torch.testing.assert_close(p.detach(), m.to(torch.bfloat16))
+
+
+def test_amp_o2_master_params_consistent_across_ranks():
+    from utils import run_distributed
+
+    run_distributed(_master_params_worker, world_size=2)
