@@ -54,3 +54,14 @@ def test_bench_cpu_gloo_two_ranks_bert():
     assert rec["value"] > 0
     assert rec["config"]["model"] == "bert-base"
     assert rec["config"]["hipgraph_step"] is False
+
+
+@pytest.mark.timeout(300)
+def test_dcgan_example_cpu_smoke():
+    # the dcgan example (reference examples/dcgan/main_amp.py twin: two
+    # optimizers sharing one amp.initialize, multi-loss scale_loss) must run
+    # end to end on CPU
+    cmd = [sys.executable, "examples/dcgan/main_amp.py",
+           "--iters", "2", "--batch-size", "4", "--opt-level", "O1"]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, f"dcgan failed:\n{out.stdout[-1500:]}\n{out.stderr[-1500:]}"
