@@ -65,3 +65,17 @@ def test_dcgan_example_cpu_smoke():
            "--iters", "2", "--batch-size", "4", "--opt-level", "O1"]
     out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=280)
     assert out.returncode == 0, f"dcgan failed:\n{out.stdout[-1500:]}\n{out.stderr[-1500:]}"
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("script", [
+    "examples/simple/distributed/distributed_data_parallel.py",
+    "examples/simple/distributed/zero2_distributed_fused_adam.py",
+])
+def test_simple_distributed_examples_two_ranks(script):
+    cmd = [sys.executable, "-m", "torch.distributed.run",
+           "--nnodes=1", "--nproc-per-node", "2",
+           "--master-addr", "127.0.0.1", "--master-port", "29613",
+           script]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, f"{script} failed:\n{out.stdout[-1500:]}\n{out.stderr[-1500:]}"
