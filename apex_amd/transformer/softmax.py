@@ -9,8 +9,11 @@ loads and 64-lane shuffle reductions (the reference's 32-wide warp tiling is
 exactly what we do NOT port).
 
 Shapes follow the reference contracts:
-- scaled_softmax / scaled_masked_softmax: input [b, np, sq, sk], sk <= 16384;
-  mask [b, 1, sq, sk] broadcast over heads, additive-bool (masked = -10000).
+- scaled_softmax / scaled_masked_softmax: input [b, np, sq, sk]; mask
+  [b, 1, sq, sk] broadcast over heads, additive-bool (masked = -10000).
+  The reference caps sk at 16384 (scaled_masked_softmax_cuda.cu:43); the
+  wave64 online-softmax kernels here have NO row ceiling, so "generic"
+  is the same code path rather than a separate extension.
 - scaled_upper_triang_masked_softmax: input [attn_batches, sq, sq] (causal).
 - generic: arbitrary sk via block-level reduction.
 """
