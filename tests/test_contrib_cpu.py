@@ -286,3 +286,63 @@ def test_exhaustive_search_monotone_random():
         perm = exhaustive_search(w)
         assert sorted(perm.tolist()) == list(range(cols))
         assert efficacy(w[:, perm]) >= base - 1e-5
+
+
+def test_xentropy_padding_idx_masks_loss_and_grad():
+    # every padding_idx target contributes zero loss AND zero grad
+    from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
+
+    torch.manual_seed(6)
+    logits = torch.randn(12, 20, requires_grad=True)
+    labels = torch.randint(0, 20, (12,))
+    labels[::3] = 7  # padding rows
+    losses = SoftmaxCrossEntropyLoss.apply(logits, labels, 0.1, 7, False)
+    assert torch.all(losses[::3] == 0)
+    losses.sum().backward()
+    assert torch.all(logits.grad[::3] == 0)
+    # non-padding rows match torch CE
+    keep = torch.ones(12, dtype=torch.bool)
+    keep[::3] = False
+    ref = torch.nn.functional.cross_entropy(
+        logits.detach()[keep], labels[keep], reduction="none", label_smoothing=0.1)
+    torch.testing.assert_close(losses[keep], ref, rtol=1e-5, atol=1e-6)
+
+
+def test_xentropy_full_smoothing_uniform_target():
+    # smoothing=1.0: the target is uniform over classes — loss equals the
+    # mean negative log prob; grads are softmax(p) - 1/K
+    from apex_amd.contrib.xentropy import SoftmaxCrossEntropyLoss
+
+    torch.manual_seed(7)
+    K = 10
+    logits = torch.randn(6, K, requires_grad=True)
+    labels = torch.randint(1, K, (6,))
+    losses = SoftmaxCrossEntropyLoss.apply(logits, labels, 1.0, 0, False)
+    ref = torch.nn.functional.cross_entropy(logits.detach(), labels,
+                                            reduction="none", label_smoothing=1.0)
+    torch.testing.assert_close(losses, ref, rtol=1e-5, atol=1e-6)
+    losses.sum().backward()
+    expected = torch.softmax(logits.detach(), -1) - 1.0 / K
+    torch.testing.assert_close(logits.grad, expected, rtol=1e-5, atol=1e-6)
+
+
+def test_focal_loss_gamma0_matches_weighted_bce():
+    # gamma=0, alpha=0.5 reduces focal loss to (0.5x) sigmoid BCE summed over
+    # real classes / num_positives
+    from apex_amd.contrib.focal_loss import focal_loss
+
+    torch.manual_seed(8)
+    N, K = 24, 12
+    x = torch.randn(N, K, requires_grad=True)
+    y = torch.randint(-1, K, (N,))  # -1 = negative sample rows
+    nps = torch.tensor([max(float((y >= 0).sum()), 1.0)])
+    loss = focal_loss(x, y, nps, K, 0.5, 0.0, 0.0)
+
+    onehot = torch.zeros(N, K)
+    valid = y >= 0
+    onehot[valid] = torch.nn.functional.one_hot(y[valid], K).float()
+    bce = torch.nn.functional.binary_cross_entropy_with_logits(
+        x.detach(), onehot, reduction="none")
+    expected = 0.5 * bce.sum() / nps
+    torch.testing.assert_close(loss.reshape(()), expected.reshape(()),
+                               rtol=1e-4, atol=1e-5)
